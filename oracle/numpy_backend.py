@@ -1,0 +1,163 @@
+"""CPU oracle backend — TEST INFRASTRUCTURE (see oracle/__init__.py).
+
+Implements the ramba_amd backend interface with NumPy shard containers and
+(for world_size > 1) torch.distributed `gloo` transport, so the full SPMD
+planner / halo-exchange / reduction logic runs on CPU ranks in CI.  It is
+injected explicitly by tests via `ramba_amd.init(backend=NumpyBackend())`;
+the product never selects it.
+"""
+
+import numpy as np
+
+from ramba_amd import ir
+from ramba_amd.shardview import box_shape
+from . import ir_interp
+
+
+class NumpyBackend:
+    name = "numpy-oracle"
+
+    def __init__(self):
+        self.containers = {}   # gid -> np.ndarray
+        self.temps = {}        # name -> np.ndarray
+        self.rt = None
+        self._dist = None
+
+    def attach(self, rt):
+        self.rt = rt
+
+    # -- process group ------------------------------------------------------
+
+    def init_process_group(self, rank, world):
+        import torch.distributed as dist
+        self._dist = dist
+        if not dist.is_initialized():
+            dist.init_process_group("gloo", rank=rank, world_size=world)
+
+    def _d(self):
+        if self._dist is None:
+            import torch.distributed as dist
+            self._dist = dist
+        return self._dist
+
+    # -- memory -------------------------------------------------------------
+
+    def alloc_container(self, bd, rt):
+        _, cshape, _, _ = rt.shard_geometry(bd)
+        if cshape is None:
+            self.containers[bd.gid] = None
+            return
+        self.containers[bd.gid] = np.empty(cshape, dtype=bd.dtype)
+
+    def free_container(self, bd):
+        self.containers.pop(bd.gid, None)
+
+    def alloc_temp(self, name, shape, dtype):
+        self.temps[name] = np.empty(shape, dtype=dtype)
+
+    def free_temps(self):
+        self.temps.clear()
+
+    # -- box copies ----------------------------------------------------------
+
+    def _cont(self, bd):
+        c = self.containers.get(bd.gid)
+        assert c is not None, f"shard for gid {bd.gid} not allocated"
+        return c
+
+    def copy_container_to_temp(self, bd, rt, part_box, vname, need_box):
+        src = self._cont(bd)[rt.container_slice(bd, part_box)]
+        lo = part_box[0] - need_box[0]
+        sl = tuple(slice(int(lo[i]), int(lo[i]) + src.shape[i])
+                   for i in range(src.ndim))
+        self.temps[vname][sl] = src
+
+    def pack_box(self, bd, rt, box):
+        import torch
+        arr = np.ascontiguousarray(self._cont(bd)[rt.container_slice(bd, box)])
+        return torch.from_numpy(arr)
+
+    def new_message_buffer(self, shape, dtype):
+        import torch
+        return torch.from_numpy(np.empty(shape, dtype=dtype))
+
+    def exchange(self, sends, recvs):
+        if not sends and not recvs:
+            return
+        dist = self._d()
+        ops = []
+        for (dst, buf) in sends:
+            ops.append(dist.P2POp(dist.isend, buf, dst))
+        for (src, buf) in recvs:
+            ops.append(dist.P2POp(dist.irecv, buf, src))
+        for req in dist.batch_isend_irecv(ops):
+            req.wait()
+
+    def unpack_box_to_container(self, bd, rt, box, buf):
+        self._cont(bd)[rt.container_slice(bd, box)] = buf.numpy()
+
+    def unpack_box_to_temp(self, vname, need_box, box, buf):
+        lo = box[0] - need_box[0]
+        a = buf.numpy()
+        sl = tuple(slice(int(lo[i]), int(lo[i]) + a.shape[i])
+                   for i in range(a.ndim))
+        self.temps[vname][sl] = a
+
+    def box_to_numpy(self, bd, rt, box):
+        return np.ascontiguousarray(
+            self._cont(bd)[rt.container_slice(bd, box)])
+
+    def write_core_from_numpy(self, bd, rt, nparr):
+        core = rt.core_box(bd, rt.rank)
+        self._cont(bd)[rt.container_slice(bd, core)] = nparr
+
+    # -- collectives ----------------------------------------------------------
+
+    def bcast_numpy(self, obj, root):
+        if self.rt.world == 1:
+            return obj
+        dist = self._d()
+        lst = [obj]
+        dist.broadcast_object_list(lst, src=root)
+        return lst[0]
+
+    _RED_MAP = {"sum": "SUM", "prod": "PRODUCT", "min": "MIN", "max": "MAX",
+                "all": "MIN", "any": "MAX"}
+
+    def allreduce(self, value, kind):
+        import torch
+        dist = self._d()
+        v = np.asarray(value)
+        dt = v.dtype
+        if dt == np.bool_:
+            v = v.astype(np.uint8)
+        t = torch.from_numpy(v.reshape(1).copy())
+        dist.all_reduce(t, op=getattr(dist.ReduceOp, self._RED_MAP[kind]))
+        out = t.numpy()[0]
+        if dt == np.bool_:
+            out = bool(out)
+        return np.asarray(out, dtype=dt)[()]
+
+    # -- kernel execution ------------------------------------------------------
+
+    def launch(self, plan):
+        env = {"__scalars__": plan.scalars}
+        for op in plan.operands:
+            if op.kind == "temp":
+                base = self.temps[op.temp_key]
+            else:
+                base = self.containers.get(op.bd.gid)
+                if base is None:
+                    continue
+            flat = base.reshape(-1)
+            isz = base.itemsize
+            strides = tuple(s * isz for s in op.strides)
+            env[op.name] = np.lib.stride_tricks.as_strided(
+                flat[op.offset0:] if op.offset0 else flat,
+                shape=plan.itershape, strides=strides)
+        for name, dt in plan.dead_vars.items():
+            env[name] = np.empty(plan.itershape, dtype=dt)
+        return ir_interp.run_statements(plan, env)
+
+    def sync(self):
+        pass

@@ -1,0 +1,642 @@
+"""NumPy-compatible API surface over the fused-deferred engine.
+
+The in-scope slice of the reference's `ramba.ndarray`
+(ramba/ramba.py:5409-5591 object model; op tables 7842-7993; creation
+routines 8552-8991; slice getitem/setitem 6429-6620/6143-6296).  Every
+elementwise / reduction op records IR into the current fused group via
+`deferred.add_op`; nothing executes until `sync()` / a reduction / `asarray`.
+"""
+
+import numbers
+
+import numpy as np
+
+from . import deferred, ir
+from .common import default_border, default_divisions, dprint
+from .shardview import View
+
+
+# ---------------------------------------------------------------------------
+# dtype helpers
+# ---------------------------------------------------------------------------
+
+def _dtype_of(x):
+    if isinstance(x, ndarray):
+        return x.dtype
+    if isinstance(x, ir.Expr):
+        return x.dtype
+    return x  # python scalar: weak promotion via np.result_type
+
+
+def _result_dtype(op, a, b):
+    if op in ir.BOOL_RESULT_BINOPS:
+        return ir.BOOL
+    da, db = _dtype_of(a), _dtype_of(b)
+    rt = np.result_type(da, db)
+    if op == "div":
+        if rt.kind in "bui":
+            rt = np.dtype(np.float64)
+    return rt
+
+
+def _shape_of(x):
+    if isinstance(x, ndarray):
+        return x.shape
+    if isinstance(x, ir.Expr):
+        return ()
+    return ()
+
+
+def _bcast(x, shape):
+    if isinstance(x, ndarray) and x.shape != shape:
+        return x.broadcast_to(shape)
+    return x
+
+
+# ---------------------------------------------------------------------------
+# ndarray
+# ---------------------------------------------------------------------------
+
+class ndarray:
+    __slots__ = ("bdarray", "view", "readonly", "__weakref__")
+
+    def __init__(self, bd, view, readonly=False):
+        self.bdarray = bd
+        self.view = view
+        self.readonly = readonly
+        bd.nviews += 1
+
+    def __del__(self):
+        try:
+            self.bdarray.decref(deferred._state.get("runtime"))
+        except Exception:
+            pass
+
+    # -- basic properties ---------------------------------------------------
+
+    @property
+    def shape(self):
+        return self.view.shape
+
+    @property
+    def dtype(self):
+        return self.bdarray.dtype
+
+    @property
+    def ndim(self):
+        return len(self.view.shape)
+
+    @property
+    def size(self):
+        n = 1
+        for s in self.view.shape:
+            n *= s
+        return n
+
+    @property
+    def T(self):
+        return self.transpose()
+
+    def __len__(self):
+        if self.ndim == 0:
+            raise TypeError("len() of unsized object")
+        return self.shape[0]
+
+    def __repr__(self):
+        return (f"ramba_amd.ndarray(shape={self.shape}, dtype={self.dtype}, "
+                f"gid={self.bdarray.gid})")
+
+    # -- views ---------------------------------------------------------------
+
+    def _with_view(self, view, readonly=None):
+        return ndarray(self.bdarray, view,
+                       self.readonly if readonly is None else readonly)
+
+    def broadcast_to(self, shape):
+        return self._with_view(self.view.broadcast_to(shape), readonly=True)
+
+    def broadcastable_to(self, shape):
+        try:
+            self.view.broadcast_to(shape)
+            return True
+        except ValueError:
+            return False
+
+    def transpose(self, *axes):
+        if len(axes) == 1 and isinstance(axes[0], (tuple, list)):
+            axes = tuple(axes[0])
+        return self._with_view(self.view.transpose(axes if axes else None))
+
+    def astype(self, dtype):
+        dtype = np.dtype(dtype)
+        out = _new_result(self.shape, dtype)
+        deferred.add_op(out, "=", ir.Cast(self, dtype), empty_like=empty_like)
+        return out
+
+    def copy(self):
+        return self.astype(self.dtype)
+
+    # -- indexing ------------------------------------------------------------
+
+    def __getitem__(self, index):
+        if not isinstance(index, tuple):
+            index = (index,)
+        if all(isinstance(i, (int, np.integer)) for i in index) \
+                and len(index) == self.ndim:
+            # full scalar read -- a sync point
+            deferred.flush()
+            coords = []
+            v = self.view
+            base = [int(o) for o in v.offset]
+            for d, i in enumerate(index):
+                i = int(i)
+                if i < 0:
+                    i += self.shape[d]
+                b, st = v.axis_map[d], v.steps[d]
+                if b >= 0:
+                    base[b] += i * st
+            rt = deferred.get_runtime()
+            return rt.read_element(self.bdarray, base)
+        view = self.view.apply_index(index)
+        return self._with_view(view)
+
+    def __setitem__(self, index, value):
+        # slice branch of setitem_array_executor (ramba.py:6273-6296)
+        if self.readonly:
+            raise ValueError("assignment destination is read-only")
+        view = self[index] if not (isinstance(index, tuple) and index == ()) \
+            else self
+        if isinstance(view, (np.generic, numbers.Number)):
+            raise IndexError("scalar element assignment is out of scope")
+        if isinstance(value, ndarray):
+            if value.shape != view.shape:
+                value = value.broadcast_to(view.shape)
+            if (value.bdarray.gid == view.bdarray.gid
+                    and value.view == view.view):
+                return  # self-assignment no-op (ramba.py:6287-6291)
+            deferred.add_op(view, "=", value, empty_like=empty_like)
+        elif isinstance(value, (numbers.Number, np.generic)):
+            deferred.add_op(view, "=", value, empty_like=empty_like)
+        elif isinstance(value, np.ndarray):
+            deferred.add_op(view, "=", fromarray(value), empty_like=empty_like)
+        else:
+            raise TypeError(f"cannot assign {type(value)}")
+
+    # -- conversion ----------------------------------------------------------
+
+    def asarray(self):
+        deferred.flush()
+        rt = deferred.get_runtime()
+        return rt.gather_view(self.bdarray, self.view)
+
+    def __array__(self, dtype=None, copy=None):
+        a = self.asarray()
+        if dtype is not None:
+            a = a.astype(dtype)
+        return a
+
+    # -- elementwise ops ------------------------------------------------------
+
+    def _binop(self, rhs, op, reverse=False, out_dtype=None):
+        if isinstance(rhs, (list, tuple)):
+            rhs = np.asarray(rhs)
+        if isinstance(rhs, np.ndarray) and rhs.shape == ():
+            rhs = rhs[()]
+        if isinstance(rhs, np.ndarray):
+            rhs = fromarray(rhs)
+        if not isinstance(rhs, (ndarray, numbers.Number, np.generic)):
+            return NotImplemented
+        a, b = (rhs, self) if reverse else (self, rhs)
+        shape = np.broadcast_shapes(_shape_of(a), _shape_of(b))
+        dt = _result_dtype(op, a, b) if out_dtype is None else out_dtype
+        a, b = _bcast(a, shape), _bcast(b, shape)
+        out = _new_result(shape, dt)
+        deferred.add_op(out, "=", ir.Bin(op, a, b, dt), empty_like=empty_like)
+        return out
+
+    def _unop(self, op):
+        if op in ir.FLOAT_UNOPS:
+            dt = np.dtype(np.float32) if self.dtype == np.float32 \
+                else np.dtype(np.float64)
+        elif op in ir.BOOL_RESULT_UNOPS:
+            dt = ir.BOOL
+        else:
+            dt = self.dtype
+        out = _new_result(self.shape, dt)
+        deferred.add_op(out, "=", ir.Un(op, self, dt), empty_like=empty_like)
+        return out
+
+    def _iop(self, rhs, aop):
+        if self.readonly:
+            raise ValueError("output array is read-only")
+        if isinstance(rhs, np.ndarray):
+            rhs = fromarray(rhs)
+        if isinstance(rhs, ndarray) and rhs.shape != self.shape:
+            rhs = rhs.broadcast_to(self.shape)
+        deferred.add_op(self, aop, rhs, empty_like=empty_like)
+        return self
+
+    # division -> multiplication-by-reciprocal rewrite (ramba.py:6121-6126);
+    # it changes numerics, so the oracle and kernels share it by construction.
+    def __truediv__(self, rhs):
+        if isinstance(rhs, (numbers.Number, np.generic)) \
+                and not isinstance(rhs, (bool, np.bool_)):
+            return self._binop(1.0 / rhs, "mul",
+                               out_dtype=_result_dtype("div", self, rhs))
+        if isinstance(rhs, np.ndarray):
+            rhs = fromarray(rhs)
+        if isinstance(rhs, ndarray):
+            recip = rhs.__rtruediv__(1.0)
+            return self._binop(recip, "mul",
+                               out_dtype=_result_dtype("div", self, recip))
+        return NotImplemented
+
+    def __rtruediv__(self, lhs):
+        return self._binop(lhs, "div", reverse=True)
+
+    def __itruediv__(self, rhs):
+        return self._iop(rhs, "/=")
+
+    # -- reductions (axis=None; reference array_unaryop reduction path,
+    #    ramba.py:5890-5920 + internal_reduction2b 5852-5863) ---------------
+
+    def _reduce(self, kind, dtype=None):
+        dt = self.dtype if dtype is None else np.dtype(dtype)
+        if kind in ("all", "any"):
+            dt = ir.BOOL
+        src = self
+        if kind in ("all", "any") and self.dtype != ir.BOOL:
+            src = ir.Un("logical_not", ir.Un("logical_not", self, ir.BOOL),
+                        ir.BOOL)
+        pend = deferred.add_reduction(src, kind, dt)
+        deferred.flush()
+        rt = deferred.get_runtime()
+        return rt.finish_reduction(pend)
+
+    def sum(self, axis=None, dtype=None, **kw):
+        assert axis is None, "axis reductions are queued (SURVEY §8f n1)"
+        return self._reduce("sum", dtype)
+
+    def prod(self, axis=None, dtype=None, **kw):
+        assert axis is None
+        return self._reduce("prod", dtype)
+
+    def min(self, axis=None, **kw):
+        assert axis is None
+        return self._reduce("min")
+
+    def max(self, axis=None, **kw):
+        assert axis is None
+        return self._reduce("max")
+
+    def all(self, axis=None, **kw):
+        assert axis is None
+        return self._reduce("all")
+
+    def any(self, axis=None, **kw):
+        assert axis is None
+        return self._reduce("any")
+
+    def mean(self, axis=None, **kw):
+        assert axis is None
+        dt = np.dtype(np.float32) if self.dtype == np.float32 \
+            else np.dtype(np.float64)
+        return self._reduce("sum", dt) / self.size
+
+
+# -- attach the op tables (reference make_method loops, ramba.py:7893-7973) --
+
+_BINOP_METHODS = {
+    "__add__": "add", "__sub__": "sub", "__mul__": "mul",
+    "__floordiv__": "floordiv", "__mod__": "mod", "__pow__": "pow",
+    "__gt__": "gt", "__lt__": "lt", "__ge__": "ge", "__le__": "le",
+    "__eq__": "eq", "__ne__": "ne",
+    "__and__": "bitand", "__or__": "bitor", "__xor__": "bitxor",
+    "__lshift__": "lshift", "__rshift__": "rshift",
+}
+_RBINOP_METHODS = {
+    "__radd__": "add", "__rsub__": "sub", "__rmul__": "mul",
+    "__rfloordiv__": "floordiv", "__rmod__": "mod", "__rpow__": "pow",
+}
+_IBINOP_METHODS = {
+    "__iadd__": "+=", "__isub__": "-=", "__imul__": "*=",
+    "__ifloordiv__": "//=", "__imod__": "%=", "__ipow__": "**=",
+}
+_UNOP_METHODS = {
+    "__abs__": "abs", "abs": "abs", "square": "square", "sqrt": "sqrt",
+    "sin": "sin", "cos": "cos", "tan": "tan", "sinh": "sinh",
+    "cosh": "cosh", "tanh": "tanh", "arcsin": "arcsin",
+    "arccos": "arccos", "arctan": "arctan", "__neg__": "neg",
+    "exp": "exp", "log": "log", "isnan": "isnan", "isinf": "isinf",
+    "isfinite": "isfinite", "logical_not": "logical_not",
+    "__invert__": "invert",
+}
+
+for _m, _op in _BINOP_METHODS.items():
+    setattr(ndarray, _m,
+            (lambda op: lambda self, rhs: self._binop(rhs, op))(_op))
+for _m, _op in _RBINOP_METHODS.items():
+    setattr(ndarray, _m,
+            (lambda op: lambda self, lhs: self._binop(lhs, op, reverse=True))(_op))
+for _m, _op in _IBINOP_METHODS.items():
+    setattr(ndarray, _m,
+            (lambda aop: lambda self, rhs: self._iop(rhs, aop))(_op))
+for _m, _op in _UNOP_METHODS.items():
+    setattr(ndarray, _m,
+            (lambda op: lambda self: self._unop(op))(_op))
+
+for _m in ("minimum", "maximum"):
+    setattr(ndarray, _m,
+            (lambda op: lambda self, rhs: self._binop(rhs, op))(_m))
+for _m in ("logical_and", "logical_or", "logical_xor"):
+    setattr(ndarray, _m,
+            (lambda op: lambda self, rhs: self._binop(rhs, op))(_m))
+
+
+# ---------------------------------------------------------------------------
+# creation (reference ramba.py:8552-8991)
+# ---------------------------------------------------------------------------
+
+def _world():
+    return deferred.get_runtime().world
+
+
+def _new_bd(shape, dtype, divisions=None, border=None, flex=True):
+    shape = (shape,) if isinstance(shape, (int, np.integer)) else tuple(shape)
+    dtype = np.dtype(np.float64 if dtype is None else dtype)
+    if divisions is None:
+        divisions = default_divisions(_world(), shape)
+    if border is None:
+        border = default_border
+    return deferred.bdarray(shape, dtype, divisions, border, flex=flex)
+
+
+def _new_result(shape, dtype, divisions=None):
+    bd = _new_bd(shape, dtype, divisions=divisions)
+    return ndarray(bd, View.identity(bd.shape))
+
+
+def empty(shape, dtype=None, local_border=None, distribution=None, **kw):
+    bd = _new_bd(shape, dtype, divisions=distribution, border=local_border)
+    arr = ndarray(bd, View.identity(bd.shape))
+    deferred.register_empty(arr)
+    return arr
+
+
+def empty_like(a, dtype=None, **kw):
+    return empty(a.shape, a.dtype if dtype is None else dtype)
+
+
+def full(shape, value, dtype=None, **kw):
+    if dtype is None:
+        if isinstance(value, (bool, np.bool_)):
+            dtype = np.bool_
+        elif isinstance(value, (int, np.integer)):
+            dtype = np.int64
+        else:
+            dtype = np.float64
+    arr = _new_result((shape,) if isinstance(shape, (int, np.integer))
+                      else tuple(shape), np.dtype(dtype))
+    deferred.add_op(arr, "=", value, empty_like=empty_like)
+    return arr
+
+
+def zeros(shape, dtype=None, **kw):
+    return full(shape, 0, dtype=np.float64 if dtype is None else dtype)
+
+
+def ones(shape, dtype=None, **kw):
+    return full(shape, 1, dtype=np.float64 if dtype is None else dtype)
+
+
+def zeros_like(a, dtype=None, **kw):
+    return zeros(a.shape, a.dtype if dtype is None else dtype)
+
+
+def ones_like(a, dtype=None, **kw):
+    return ones(a.shape, a.dtype if dtype is None else dtype)
+
+
+def full_like(a, value, dtype=None, **kw):
+    return full(a.shape, value, dtype=a.dtype if dtype is None else dtype)
+
+
+def arange(start, stop=None, step=None, dtype=None, **kw):
+    """reference arange_executor (ramba.py:8952-8960):
+    `res = start + step * (index[0] + global_start[0])` -- exact int64."""
+    if stop is None:
+        size = int(start)
+        expr = ir.Iota(0)
+    elif step is None:
+        size = int(stop - start)
+        expr = ir.Bin("add", start, ir.Iota(0), None)
+    else:
+        size = int((stop - start + step - 1) // step) if step > 0 else \
+            int((stop - start + step + 1) // step)
+        expr = ir.Bin("add", start, ir.Bin("mul", step, ir.Iota(0), None),
+                      None)
+    size = max(0, size)
+    if dtype is None:
+        if any(isinstance(x, (float, np.floating))
+               for x in (start, stop, step) if x is not None):
+            dtype = np.float64
+        else:
+            dtype = np.int64
+    arr = _new_result((size,), np.dtype(dtype))
+    deferred.add_op(arr, "=", expr, empty_like=empty_like)
+    return arr
+
+
+def linspace(start, stop, num=50, endpoint=True, retstep=False, dtype=None):
+    # reference ramba.py:8977-8991
+    assert num > 0
+    length = stop - start
+    step = length / (num - 1) if endpoint else length / num
+    res = arange(num) * step + start
+    if dtype is not None:
+        res = res.astype(dtype)
+    if retstep:
+        return res, step
+    return res
+
+
+def fromfunction(function, shape, dtype=None, **kw):
+    """Iota-expression fill (analog of the reference's string fillers,
+    create_array_executor ramba.py:8563).  `function` receives symbolic
+    index expressions supporting the ndarray op vocabulary."""
+    shape = tuple(int(s) for s in shape)
+    dt = np.dtype(np.float64 if dtype is None else dtype)
+    syms = [_IndexSym(ir.Iota(d)) for d in range(len(shape))]
+    expr = function(*syms)
+    if isinstance(expr, _IndexSym):
+        expr = expr.e
+    arr = _new_result(shape, dt)
+    deferred.add_op(arr, "=", expr, empty_like=empty_like)
+    return arr
+
+
+class _IndexSym:
+    """Symbolic index used by fromfunction."""
+
+    def __init__(self, e):
+        self.e = e
+
+    def _mk(self, op, other, reverse=False):
+        o = other.e if isinstance(other, _IndexSym) else other
+        a, b = (o, self.e) if reverse else (self.e, o)
+        dt = ir.BOOL if op in ir.BOOL_RESULT_BINOPS else \
+            np.result_type(_dtype_of(a), _dtype_of(b))
+        return _IndexSym(ir.Bin(op, a, b, dt))
+
+    def __add__(self, o):
+        return self._mk("add", o)
+
+    def __radd__(self, o):
+        return self._mk("add", o, True)
+
+    def __sub__(self, o):
+        return self._mk("sub", o)
+
+    def __rsub__(self, o):
+        return self._mk("sub", o, True)
+
+    def __mul__(self, o):
+        return self._mk("mul", o)
+
+    def __rmul__(self, o):
+        return self._mk("mul", o, True)
+
+    def __mod__(self, o):
+        return self._mk("mod", o)
+
+
+def fromarray(a):
+    """Distribute a host numpy array (tests / small inputs only)."""
+    a = np.asarray(a)
+    if a.shape == ():
+        return a[()]
+    arr = empty(a.shape, a.dtype)
+    deferred.flush()
+    rt = deferred.get_runtime()
+    rt.scatter_numpy(arr.bdarray, a)
+    return arr
+
+
+def array(a, dtype=None):
+    a = np.asarray(a, dtype=dtype)
+    return fromarray(a)
+
+
+# ---------------------------------------------------------------------------
+# module-level functions
+# ---------------------------------------------------------------------------
+
+def sync():
+    deferred.flush()
+    rt = deferred.get_runtime()
+    rt.backend.sync()
+
+
+def asarray(a):
+    if isinstance(a, ndarray):
+        return a.asarray()
+    return np.asarray(a)
+
+
+def where(cond, a, b):
+    """Elementwise select (ramba.py:9755-9831 `where`, the in-scope part)."""
+    shape = np.broadcast_shapes(_shape_of(cond), _shape_of(a), _shape_of(b))
+    da, db = _dtype_of(a), _dtype_of(b)
+    dt = np.result_type(da, db)
+    cond, a, b = _bcast(cond, shape), _bcast(a, shape), _bcast(b, shape)
+    out = _new_result(shape, dt)
+    deferred.add_op(out, "=", ir.Where(cond, a, b, dt), empty_like=empty_like)
+    return out
+
+
+def _module_unop(name):
+    def f(x, **kw):
+        if isinstance(x, ndarray):
+            return getattr(x, name)()
+        return getattr(np, name)(x, **kw)
+    f.__name__ = name
+    return f
+
+
+sin = _module_unop("sin")
+cos = _module_unop("cos")
+tan = _module_unop("tan")
+sinh = _module_unop("sinh")
+cosh = _module_unop("cosh")
+tanh = _module_unop("tanh")
+arcsin = _module_unop("arcsin")
+arccos = _module_unop("arccos")
+arctan = _module_unop("arctan")
+exp = _module_unop("exp")
+log = _module_unop("log")
+sqrt = _module_unop("sqrt")
+square = _module_unop("square")
+absolute = _module_unop("abs")
+isnan = _module_unop("isnan")
+isinf = _module_unop("isinf")
+isfinite = _module_unop("isfinite")
+logical_not = _module_unop("logical_not")
+
+
+def _module_reduction(name):
+    def f(x, axis=None, **kw):
+        if isinstance(x, ndarray):
+            return getattr(x, name)(axis=axis, **kw)
+        return getattr(np, name)(x, axis=axis, **kw)
+    f.__name__ = name
+    return f
+
+
+def _module_binop(name):
+    def f(a, b, **kw):
+        if isinstance(a, ndarray):
+            return getattr(a, name)(b)
+        if isinstance(b, ndarray):
+            return getattr(b, name)(a)
+        return getattr(np, name)(a, b, **kw)
+    f.__name__ = name
+    return f
+
+
+sum = _module_reduction("sum")
+prod = _module_reduction("prod")
+amin = _module_reduction("min")
+amax = _module_reduction("max")
+minimum = _module_binop("minimum")
+maximum = _module_binop("maximum")
+logical_and = _module_binop("logical_and")
+logical_or = _module_binop("logical_or")
+logical_xor = _module_binop("logical_xor")
+power = _module_binop("__pow__")
+
+
+def broadcast_to(a, shape):
+    if isinstance(a, ndarray):
+        return a.broadcast_to(shape)
+    return np.broadcast_to(a, shape)
+
+
+def transpose(a, axes=None):
+    if isinstance(a, ndarray):
+        return a.transpose(axes) if axes else a.transpose()
+    return np.transpose(a, axes)
+
+
+# numpy dtype aliases on the module (reference exports these)
+float64 = np.float64
+float32 = np.float32
+int64 = np.int64
+int32 = np.int32
+int16 = np.int16
+int8 = np.int8
+uint8 = np.uint8
+bool_ = np.bool_
+pi = np.pi
+e = np.e
+inf = np.inf
+nan = np.nan

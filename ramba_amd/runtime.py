@@ -1,0 +1,366 @@
+"""SPMD runtime — the replacement for the reference's worker side
+(`RemoteState.run_deferred_ops`, ramba/ramba.py:3493-3819).
+
+One process per GPU (the analog of the reference's MPI-SPMD mode,
+ramba.py:3986-3993).  Rank r owns shard r of every backing array, stored in a
+container that covers its division box plus a border ring
+(`LocalNdarray.bcontainer`, ramba.py:1193).  For each fused group the runtime:
+
+1. partitions the iteration space (owner-computes over the output view's
+   partition -- the reference's exec_dist subspace, ramba.py:3497),
+2. plans, deterministically and identically on every rank, which sub-boxes
+   must move between ranks (the overlap scan + comm_queues put/get sites,
+   ramba.py:3547-3693, and getborder, ramba.py:1260) and moves them with
+   torch.distributed P2P (RCCL over xGMI on GPUs, gloo in CPU tests),
+3. asks the backend to run the fused kernel over the local box,
+4. finishes reductions with one allreduce (replacing the reference's
+   gather-to-driver, ramba.py:5745-5764/5852-5863).
+"""
+
+import os
+
+import numpy as np
+
+from . import deferred, ir
+from .common import dprint, default_border, add_time
+from .shardview import (box_contains, box_empty, box_eq, box_intersect,
+                        box_shape, box_subtract, View)
+
+
+class OperandPlan:
+    """How one live array var is addressed by the kernel on this rank."""
+    __slots__ = ("name", "kind", "bd", "temp_key", "offset0", "strides",
+                 "dtype")
+
+    def __init__(self, name, kind, bd, temp_key, offset0, strides, dtype):
+        self.name = name
+        self.kind = kind            # "container" | "temp"
+        self.bd = bd
+        self.temp_key = temp_key
+        self.offset0 = offset0
+        self.strides = strides
+        self.dtype = dtype
+
+
+class KernelPlan:
+    __slots__ = ("itershape", "global_start", "operands", "scalars",
+                 "statements", "reductions", "dead_vars", "temps")
+
+    def __init__(self):
+        self.itershape = None
+        self.global_start = None
+        self.operands = []          # list[OperandPlan]
+        self.scalars = {}           # name -> (value, dtype)
+        self.statements = []
+        self.reductions = []        # list[ir.ReductionSpec]
+        self.dead_vars = {}         # name -> dtype (register temps)
+        self.temps = {}             # temp_key -> (shape, dtype) buffers
+
+
+class Runtime:
+    def __init__(self, backend, rank=0, world=1):
+        self.backend = backend
+        self.rank = rank
+        self.world = world
+
+    # ------------------------------------------------------------------
+    def shard_geometry(self, bd, r=None):
+        """(div box, container shape, container strides, border per axis)."""
+        r = self.rank if r is None else r
+        d = bd.divisions[r]
+        nd = len(bd.shape)
+        border = tuple(bd.border for _ in range(nd))
+        if np.any(d[1] < d[0]):
+            return None, None, None, border
+        cshape = tuple(int(d[1, i] - d[0, i] + 1) + 2 * bd.border
+                       for i in range(nd))
+        cstrides = [1] * nd
+        for i in range(nd - 2, -1, -1):
+            cstrides[i] = cstrides[i + 1] * cshape[i + 1]
+        return d, cshape, tuple(cstrides), border
+
+    def core_box(self, bd, r):
+        d = bd.divisions[r]
+        if np.any(d[1] < d[0]):
+            return None
+        return d.astype(np.int64)
+
+    def border_box(self, bd, r):
+        """core expanded by the border ring, clipped to the global array."""
+        d = self.core_box(bd, r)
+        if d is None:
+            return None
+        lo = np.maximum(d[0] - bd.border, 0)
+        hi = np.minimum(d[1] + bd.border,
+                        np.array(bd.shape, dtype=np.int64) - 1)
+        return np.array([lo, hi])
+
+    # ------------------------------------------------------------------
+    def execute_group(self, group):
+        live, dead = deferred.compute_live_vars(group)
+        eboxes = group.exec_boxes()
+
+        # flex arrays adopt the group's partition (ramba.py:8093-8101)
+        nd = len(group.shape)
+        for oi in live.values():
+            if oi.bd.is_flex and oi.bd.shape == group.shape:
+                divs = np.zeros((self.world, 2, nd), dtype=np.int64)
+                for r, b in enumerate(eboxes):
+                    if b is None:
+                        divs[r, 1, :] = -1
+                    else:
+                        divs[r] = b
+                oi.bd.divisions = divs
+                oi.bd.flex = False
+
+        # allocate + mark constructed (creation on first use, ramba.py:3506)
+        for oi in live.values():
+            if not oi.bd.constructed:
+                self.backend.alloc_container(oi.bd, self)
+                oi.bd.constructed = True
+
+        plan = KernelPlan()
+        plan.scalars = dict(group.scalars)
+        plan.statements = group.statements
+        plan.reductions = [s for (s, _) in group.reductions]
+        plan.dead_vars = {n: oi.dtype for n, oi in dead.items()}
+
+        # ---- per-gid needed boxes on every rank (deterministic everywhere)
+        # needed[r] : gid -> (container_fill_box or None, {var: temp box})
+        comm_msgs = []   # (dst, src, bd, box, target) target: ("border",) or ("temp", var)
+        my_plans = {}
+        for r in range(self.world):
+            ib = eboxes[r]
+            per_gid = {}
+            for gid, names in group.vars_by_gid.items():
+                vars_here = [live[n] for n in names if n in live]
+                if not vars_here:
+                    continue
+                bd = vars_here[0].bd
+                core = self.core_box(bd, r)
+                bbox = self.border_box(bd, r)
+                fit_boxes = []
+                temp_vars = {}
+                for oi in vars_here:
+                    if ib is None:
+                        need = None
+                    else:
+                        need = oi.view.image_box(ib)
+                    if oi.written and need is not None:
+                        assert core is not None and box_contains(core, need), \
+                            "write outside owned shard (incompatible dists)"
+                    if need is None:
+                        continue
+                    if core is not None and box_contains(core, need):
+                        continue  # fully local
+                    if bbox is not None and box_contains(bbox, need):
+                        fit_boxes.append(need)
+                    else:
+                        temp_vars[oi.name] = need
+                fill = None
+                if fit_boxes:
+                    lo = np.min([b[0] for b in fit_boxes], axis=0)
+                    hi = np.max([b[1] for b in fit_boxes], axis=0)
+                    fill = np.array([lo, hi])
+                per_gid[gid] = (bd, fill, temp_vars)
+                # build transfer list
+                if fill is not None:
+                    missing = box_subtract(fill, core) if core is not None \
+                        else [fill]
+                    for mbox in missing:
+                        for s in range(self.world):
+                            if s == r:
+                                continue
+                            part = box_intersect(mbox, self.core_box(bd, s))
+                            if part is not None:
+                                comm_msgs.append((r, s, bd, part, ("border", gid)))
+                for vname, need in temp_vars.items():
+                    for s in range(self.world):
+                        part = box_intersect(need, self.core_box(bd, s))
+                        if part is None:
+                            continue
+                        if s == r:
+                            continue  # local part copied below
+                        comm_msgs.append((r, s, bd, part, ("temp", vname)))
+            if r == self.rank:
+                my_plans = per_gid
+
+        # ---- allocate temp operand buffers + copy local parts
+        temp_geom = {}   # var -> (need box, strides)
+        for gid, (bd, fill, temp_vars) in my_plans.items():
+            for vname, need in temp_vars.items():
+                shape = box_shape(need)
+                self.backend.alloc_temp(vname, shape, bd.dtype)
+                strides = [1] * len(shape)
+                for i in range(len(shape) - 2, -1, -1):
+                    strides[i] = strides[i + 1] * shape[i + 1]
+                temp_geom[vname] = (need, tuple(strides))
+                core = self.core_box(bd, self.rank)
+                part = box_intersect(need, core)
+                if part is not None and bd.constructed:
+                    self.backend.copy_container_to_temp(bd, self, part,
+                                                        vname, need)
+
+        # ---- execute transfers
+        if comm_msgs:
+            self._do_comms(comm_msgs, temp_geom)
+
+        # ---- operand descriptors for the kernel
+        ib = eboxes[self.rank]
+        if ib is not None:
+            plan.itershape = box_shape(ib)
+            plan.global_start = tuple(int(x) for x in ib[0])
+            for name, oi in live.items():
+                if name in temp_geom:
+                    need, strides = temp_geom[name]
+                    off0, s = oi.view.operand_addressing(
+                        ib[0], strides, need[0], (0,) * len(strides))
+                    plan.operands.append(OperandPlan(
+                        name, "temp", oi.bd, name, off0, s, oi.dtype))
+                else:
+                    d, cshape, cstrides, border = self.shard_geometry(oi.bd)
+                    if d is None:
+                        # no local shard; var must be unused here
+                        need = oi.view.image_box(ib)
+                        assert need is None, "operand needed but unowned"
+                        plan.operands.append(OperandPlan(
+                            name, "container", oi.bd, None, 0,
+                            (0,) * len(plan.itershape), oi.dtype))
+                        continue
+                    off0, s = oi.view.operand_addressing(
+                        ib[0], cstrides, d[0], border)
+                    plan.operands.append(OperandPlan(
+                        name, "container", oi.bd, None, off0, s, oi.dtype))
+            partials = self.backend.launch(plan)
+        else:
+            partials = [np.asarray(ir.reduction_init(spec.kind, spec.dtype),
+                                   dtype=spec.dtype)[()]
+                        for spec in plan.reductions]
+
+        for (spec, pend), val in zip(group.reductions, partials):
+            pend.partial = np.asarray(val, dtype=spec.dtype)[()]
+
+        self.backend.free_temps()
+
+        # ---- deferred frees (ramba.py:8321-8328)
+        for bd in group.delete_bds:
+            if bd.constructed:
+                self.backend.free_container(bd)
+                bd.constructed = False
+        add_time("execute_group", 0.0)
+
+    # ------------------------------------------------------------------
+    def _do_comms(self, msgs, temp_geom):
+        """Move sub-boxes between ranks.  The message list is identical on
+        every rank; pairwise ordering comes from a deterministic sort."""
+        def key(m):
+            dst, src, bd, bx, tgt = m
+            return (dst, src, bd.gid, tuple(bx[0]), tuple(bx[1]), tgt[0],
+                    str(tgt[1]))
+        msgs = sorted(msgs, key=key)
+        sends, recvs = [], []
+        for m in msgs:
+            dst, src, bd, bx, tgt = m
+            if src == self.rank:
+                buf = self.backend.pack_box(bd, self, bx)
+                sends.append((dst, buf))
+            if dst == self.rank:
+                buf = self.backend.new_message_buffer(box_shape(bx), bd.dtype)
+                recvs.append((src, buf, bd, bx, tgt))
+        self.backend.exchange(sends, [(s, b) for (s, b, _, _, _) in recvs])
+        for (src, buf, bd, bx, tgt) in recvs:
+            if tgt[0] == "border":
+                self.backend.unpack_box_to_container(bd, self, bx, buf)
+            else:
+                vname = tgt[1]
+                need, _ = temp_geom[vname]
+                self.backend.unpack_box_to_temp(vname, need, bx, buf)
+
+    # ------------------------------------------------------------------
+    def finish_reduction(self, pend):
+        val = pend.partial
+        if self.world > 1:
+            val = self.backend.allreduce(val, pend.kind)
+        return np.asarray(val, dtype=pend.dtype)[()]
+
+    # ------------------------------------------------------------------
+    def gather_view(self, bd, view):
+        """Assemble the full (global) content of `view` as numpy on every
+        rank — the exit/parity boundary (`asarray`, ramba.py:5735-5764)."""
+        base = np.empty(bd.shape, dtype=bd.dtype)
+        for r in range(self.world):
+            core = self.core_box(bd, r)
+            if core is None:
+                continue
+            if r == self.rank:
+                part = self.backend.box_to_numpy(bd, self, core)
+            else:
+                part = None
+            part = self.backend.bcast_numpy(part, root=r)
+            sl = tuple(slice(int(core[0, i]), int(core[1, i]) + 1)
+                       for i in range(len(bd.shape)))
+            base[sl] = part
+        return numpy_view(base, view)
+
+    def container_slice(self, bd, box, r=None):
+        """Container index slices covering a global base box."""
+        d = bd.divisions[self.rank if r is None else r]
+        nd = len(bd.shape)
+        return tuple(slice(int(box[0, i] - d[0, i] + bd.border),
+                           int(box[1, i] - d[0, i] + bd.border) + 1)
+                     for i in range(nd))
+
+    def free_shard(self, bd):
+        if bd.constructed:
+            self.backend.free_container(bd)
+            bd.constructed = False
+
+    def scatter_numpy(self, bd, nparr):
+        """Distribute a host array: each rank copies its core slice."""
+        assert tuple(nparr.shape) == bd.shape
+        if not bd.constructed:
+            self.backend.alloc_container(bd, self)
+            bd.constructed = True
+            bd.flex = False
+        core = self.core_box(bd, self.rank)
+        if core is not None:
+            sl = tuple(slice(int(core[0, i]), int(core[1, i]) + 1)
+                       for i in range(len(bd.shape)))
+            self.backend.write_core_from_numpy(
+                bd, self, np.ascontiguousarray(nparr[sl], dtype=bd.dtype))
+
+    def read_element(self, bd, coords):
+        box = np.array([coords, coords], dtype=np.int64)
+        owner = None
+        for r in range(self.world):
+            core = self.core_box(bd, r)
+            if core is not None and box_contains(core, box):
+                owner = r
+                break
+        assert owner is not None
+        if owner == self.rank:
+            val = self.backend.box_to_numpy(bd, self, box)
+        else:
+            val = None
+        val = self.backend.bcast_numpy(val, root=owner)
+        return val.reshape(())[()]
+
+
+def numpy_view(base, view):
+    """Apply a View to a full numpy base array (host-side, exit boundary)."""
+    itemsize = base.itemsize
+    offset = 0
+    strides = []
+    bstr = base.strides
+    for b in range(len(base.shape)):
+        offset += view.offset[b] * bstr[b]
+    for v in range(view.ndim):
+        b, st = view.axis_map[v], view.steps[v]
+        strides.append(st * bstr[b] if b >= 0 else 0)
+    flat = np.lib.stride_tricks.as_strided(
+        base, shape=view.shape, strides=tuple(strides),
+        # numpy as_strided has no offset arg; slice the base buffer instead
+    ) if offset == 0 else np.lib.stride_tricks.as_strided(
+        base.reshape(-1)[offset // itemsize:], shape=view.shape,
+        strides=tuple(strides))
+    return flat.copy()
