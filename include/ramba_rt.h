@@ -1,0 +1,84 @@
+/* ramba_rt — C-ABI runtime for the MI355X-native Ramba hot-path backend.
+ *
+ * This is the drop-in boundary of SURVEY.md §8(b): the Python host lowers
+ * each fused deferred-op group (the payload of the reference's
+ * remote_exec_all("run_deferred_ops", ...) call, ramba/ramba.py:8286-8298)
+ * to generated gfx950 HIP source + a packed argument buffer, and drives it
+ * through these entry points.  No torch types cross this boundary — only
+ * plain pointers and sizes.
+ *
+ * Reference interfaces each entry point replaces (file:line in the
+ * reference repo Python-for-HPC/ramba):
+ *   rt_init           — worker startup / device binding (ramba.py:10646-10725)
+ *   rt_kernel_get     — FunctionMetadata JIT + sha-keyed code cache
+ *                       (ramba.py:249-438, fname hash at 8260)
+ *   rt_launch         — the fused-loop invocation func(global_start,
+ *                       itershape, ...) (ramba.py:3768/3779)
+ *   rt_copy_box       — shard sub-box pack/unpack for the part/halo
+ *                       exchange (comm_queues puts at ramba.py:3656,
+ *                       getborder ramba.py:1260) and gather (get_view 2160)
+ *   rt_stream_sync /
+ *   rt_device_sync    — worker-side completion (sync, ramba.py:9843)
+ *   rt_event_*        — hot-loop timing (add_time, ramba.py:945-1022)
+ *
+ * Python-side binding a maintainer would write: see INTEGRATION.md
+ * (ctypes stub).
+ */
+
+#ifndef RAMBA_RT_H
+#define RAMBA_RT_H
+
+#include <stddef.h>
+#include <stdint.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+/* All functions return 0 on success, nonzero on failure; rt_last_error()
+ * returns a static string describing the last failure (fail-fast contract —
+ * the analog of the reference's ("ERROR", worker, traceback) reply,
+ * ramba.py:3878-3881). */
+
+int         rt_init(int device);
+int         rt_device_count(void);
+const char *rt_last_error(void);
+
+/* Compile (hiprtc, gfx950) and cache a kernel. `key` is the host-computed
+ * structural hash of the fused group (the analog of the sha256 code hash at
+ * ramba.py:8260); `source` is full HIP source; `kname` the extern-C kernel
+ * symbol.  Returns a handle valid for the process lifetime. */
+int rt_kernel_get(const char *key, const char *source, const char *kname,
+                  void **out_kernel);
+
+/* Compile-only validation of generated source (no GPU needed). */
+int rt_compile_check(const char *source);
+
+/* Launch a compiled kernel with a packed argument buffer (passed via
+ * HIP_LAUNCH_PARAM_BUFFER_POINTER).  block is (bx,1,1). */
+int rt_launch(void *kernel, unsigned gx, unsigned gy, unsigned gz,
+              unsigned bx, uintptr_t stream, const void *args,
+              size_t argsize);
+
+/* Strided box copy between device buffers (pack/unpack/halo/gather).
+ * shape/strides are in elements, nd <= 4, elemsize in {1,2,4,8}. */
+int rt_copy_box(uintptr_t stream, void *dst, const void *src, int nd,
+                const int64_t *shape, const int64_t *dst_strides,
+                const int64_t *src_strides, int64_t dst_off, int64_t src_off,
+                int elemsize);
+
+int rt_stream_sync(uintptr_t stream);
+int rt_device_sync(void);
+
+/* HIP-event timing for the bench's roofline leg. */
+int   rt_event_create(void **ev);
+int   rt_event_destroy(void *ev);
+int   rt_event_record(void *ev, uintptr_t stream);
+/* elapsed ms between two recorded events (synchronises on `end`) */
+int   rt_event_elapsed(void *start, void *end, float *ms);
+
+#ifdef __cplusplus
+}
+#endif
+
+#endif /* RAMBA_RT_H */

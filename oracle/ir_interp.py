@@ -22,6 +22,8 @@ def eval_expr(e, env, itershape, global_start):
     if isinstance(e, ir.ScalarArg):
         val, dt = env["__scalars__"][e.name]
         return np.asarray(val, dtype=dt)[()]
+    if isinstance(e, ir.Const):
+        return np.asarray(e.value, dtype=e.dtype)[()]
     if isinstance(e, ir.Iota):
         n = itershape[e.axis]
         idx = np.arange(global_start[e.axis], global_start[e.axis] + n,

@@ -1,0 +1,837 @@
+"""Fused-group IR -> gfx950 HIP source.
+
+The analog of the reference's function-string emission
+(`deferred_op.execute`, ramba/ramba.py:8115-8265: codelines -> a
+`numba.pndindex` loop, sha-hashed for the JIT cache) — but emitting HIP for
+hand-tuned CDNA4 execution instead of Python for Numba:
+
+- 256-thread blocks (4 waves of 64), grid-stride over the innermost axis,
+  outer axes on blockIdx.y/z with grid-stride loops;
+- vectorised loads/stores (`ext_vector_type`, 16 B per lane) for unit-stride
+  aligned operands — the coalescing rule of the CDNA4 guide (G2/G13);
+- dead arrays live in registers (the live_gids rule, ramba.py:8123);
+- `sin`+`cos` of one operand fuse into a single `sincos` range reduction;
+- axis-less reductions: per-thread accumulator -> 64-wide `__shfl_down`
+  wave tree -> LDS across the 4 waves -> one partial per workgroup -> a
+  one-block finish kernel (replaces internal_reduction1/2b,
+  ramba.py:5789-5863; the cross-rank step is RCCL allreduce in the runtime).
+
+The kernel cache key is the structural hash of (statements, operand stride
+classes, dtypes, vec width) — runtime values (pointers, extents, strides,
+scalars) are kernel arguments, so iterating workloads reuse one kernel.
+"""
+
+import hashlib
+import struct
+
+import numpy as np
+
+from . import ir
+
+CTYPE = {
+    np.dtype(np.bool_): "unsigned char",
+    np.dtype(np.uint8): "unsigned char",
+    np.dtype(np.int8): "signed char",
+    np.dtype(np.int16): "short",
+    np.dtype(np.int32): "int",
+    np.dtype(np.int64): "long long",
+    np.dtype(np.float32): "float",
+    np.dtype(np.float64): "double",
+}
+
+VECN = {"double": "d2", "float": "f4", "long long": "l2"}
+
+
+def ctype(dt):
+    return CTYPE[np.dtype(dt)]
+
+
+def is_float(dt):
+    return np.dtype(dt).kind == "f"
+
+
+# ---------------------------------------------------------------------------
+# operand structural classes
+# ---------------------------------------------------------------------------
+
+class OpndClass:
+    __slots__ = ("name", "dtype", "inner", "outer_g", "loaded", "stored")
+    # inner: 'v' vector (unit stride, aligned), 'u' unit stride scalar,
+    #        'z' zero stride, 'g' general stride
+
+    def __init__(self, name, dtype, inner, outer_g):
+        self.name = name
+        self.dtype = np.dtype(dtype)
+        self.inner = inner
+        self.outer_g = outer_g   # tuple of bools: outer axis stride passed?
+        self.loaded = False
+        self.stored = False
+
+
+def classify_plan(plan, vec):
+    """Structural classification of every live operand."""
+    nd = len(plan.itershape)
+    out = []
+    for op in plan.operands:
+        s = op.strides
+        inner = s[nd - 1] if nd else 0
+        if inner == 1 and (op.offset0 % vec) == 0 and vec > 1:
+            icl = "v"
+        elif inner == 1:
+            icl = "u"
+        elif inner == 0:
+            icl = "z"
+        else:
+            icl = "g"
+        outer_g = tuple(s[d] != 0 for d in range(nd - 1))
+        out.append(OpndClass(op.name, op.dtype, icl, outer_g))
+    return out
+
+
+# ---------------------------------------------------------------------------
+# SSA renaming + use analysis
+# ---------------------------------------------------------------------------
+
+def _rename(e, version):
+    if isinstance(e, ir.Ref):
+        v = version.get(e.name, 0)
+        return ir.Ref(f"{e.name}@{v}", e.dtype)
+    if isinstance(e, ir.Bin):
+        return ir.Bin(e.op, _rename(e.a, version), _rename(e.b, version),
+                      e.dtype)
+    if isinstance(e, ir.Un):
+        return ir.Un(e.op, _rename(e.a, version), e.dtype)
+    if isinstance(e, ir.Cast):
+        return ir.Cast(_rename(e.a, version), e.dtype)
+    if isinstance(e, ir.Where):
+        return ir.Where(_rename(e.c, version), _rename(e.a, version),
+                        _rename(e.b, version), e.dtype)
+    return e
+
+
+def ssa_statements(statements, acc_names):
+    """Returns list of (versioned_target|None-for-acc, acc_name|None, expr)"""
+    version = {}
+    out = []
+    for st in statements:
+        if st.target in acc_names:
+            # acc = comb(acc, src): rename only the src side
+            assert isinstance(st.expr, ir.Bin)
+            src = _rename(st.expr.b, version)
+            out.append((None, (st.target, st.expr.op), src))
+        else:
+            e = _rename(st.expr, version)
+            v = version.get(st.target, 0) + 1
+            version[st.target] = v
+            out.append((f"{st.target}@{v}", None, e))
+    return out, version
+
+
+def collect_nodes(e, bag):
+    bag[e] = bag.get(e, 0) + 1
+    if isinstance(e, ir.Bin):
+        collect_nodes(e.a, bag)
+        collect_nodes(e.b, bag)
+    elif isinstance(e, (ir.Un, ir.Cast)):
+        collect_nodes(e.a, bag)
+    elif isinstance(e, ir.Where):
+        collect_nodes(e.c, bag)
+        collect_nodes(e.a, bag)
+        collect_nodes(e.b, bag)
+
+
+# ---------------------------------------------------------------------------
+# expression emission (per lane)
+# ---------------------------------------------------------------------------
+
+class LaneEmitter:
+    def __init__(self, gen, lane_tag, idx_exprs):
+        self.gen = gen
+        self.tag = lane_tag          # suffix for locals
+        self.idx = idx_exprs         # C index expr per axis (strings)
+        self.memo = {}               # node -> C var / expr string
+        self.lines = []
+        self.n = 0
+
+    def fresh(self, ct, init):
+        self.n += 1
+        v = f"t{self.n}{self.tag}"
+        self.lines.append(f"      {ct} {v} = {init};")
+        return v
+
+    def emit(self, e):
+        if e in self.memo:
+            return self.memo[e]
+        r = self._emit(e)
+        self.memo[e] = r
+        return r
+
+    def _emit(self, e):
+        g = self.gen
+        if isinstance(e, ir.Ref):
+            return g.lane_value(self, e.name, e.dtype)
+        if isinstance(e, ir.ScalarArg):
+            ct = ctype(e.dtype)
+            if np.dtype(e.dtype).kind == "f":
+                return f"(({ct})a.{e.name})"
+            return f"(({ct})a.{e.name})"
+        if isinstance(e, ir.Const):
+            ct = ctype(e.dtype)
+            v = e.value
+            if isinstance(v, float):
+                return f"(({ct}){v!r})"
+            return f"(({ct}){int(v)})"
+        if isinstance(e, ir.Iota):
+            return f"((long long)({self.idx[e.axis]}))"
+        if isinstance(e, ir.Cast):
+            a = self.emit(e.a)
+            src_dt = _dt(e.a)
+            ct = ctype(e.dtype)
+            if np.dtype(e.dtype) == np.dtype(np.bool_) and \
+                    np.dtype(src_dt) != np.dtype(np.bool_):
+                return self.fresh("unsigned char", f"({a}) != 0")
+            return self.fresh(ct, f"({ct})({a})")
+        if isinstance(e, ir.Where):
+            c = self.emit(e.c)
+            a = self.emit(e.a)
+            b = self.emit(e.b)
+            ct = ctype(e.dtype)
+            return self.fresh(ct, f"({c}) ? ({ct})({a}) : ({ct})({b})")
+        if isinstance(e, ir.Un):
+            return self._emit_un(e)
+        if isinstance(e, ir.Bin):
+            return self._emit_bin(e)
+        raise TypeError(f"bad node {e!r}")
+
+    def _emit_un(self, e):
+        # sincos pairing: if both sin(X) and cos(X) are wanted, emit one call
+        g = self.gen
+        if e.op in ("sin", "cos"):
+            other = ir.Un("cos" if e.op == "sin" else "sin", e.a, e.dtype)
+            if other in g.wanted and other not in self.memo:
+                x = self.emit(e.a)
+                ct = ctype(e.dtype)
+                fn = "sincosf" if ct == "float" else "sincos"
+                self.n += 1
+                s, c = f"sc_s{self.n}{self.tag}", f"sc_c{self.n}{self.tag}"
+                self.lines.append(
+                    f"      {ct} {s}, {c}; {fn}(({ct})({x}), &{s}, &{c});")
+                sin_node = e if e.op == "sin" else other
+                cos_node = e if e.op == "cos" else other
+                self.memo[sin_node] = s
+                self.memo[cos_node] = c
+                return self.memo[e]
+        a = self.emit(e.a)
+        src_dt = np.dtype(_dt(e.a))
+        ct = ctype(e.dtype)
+        f32 = ct == "float"
+        sfx = "f" if f32 else ""
+        op = e.op
+        simple = {"sqrt": "sqrt", "sin": "sin", "cos": "cos", "tan": "tan",
+                  "sinh": "sinh", "cosh": "cosh", "tanh": "tanh",
+                  "arcsin": "asin", "arccos": "acos", "arctan": "atan",
+                  "exp": "exp", "log": "log"}
+        if op in simple:
+            return self.fresh(ct, f"{simple[op]}{sfx}(({ct})({a}))")
+        if op == "neg":
+            return self.fresh(ct, f"-({a})")
+        if op == "abs":
+            if src_dt.kind == "f":
+                return self.fresh(ct, f"fabs{sfx}({a})")
+            if src_dt.kind == "u" or src_dt == np.dtype(np.bool_):
+                return self.fresh(ct, f"({a})")
+            return self.fresh(ct, f"(({a}) < 0 ? -({a}) : ({a}))")
+        if op == "square":
+            return self.fresh(ct, f"({a}) * ({a})")
+        if op == "isnan":
+            return self.fresh("unsigned char",
+                              f"__builtin_isnan((double)({a})) ? 1 : 0"
+                              if src_dt.kind == "f" else "0")
+        if op == "isinf":
+            return self.fresh("unsigned char",
+                              f"__builtin_isinf((double)({a})) ? 1 : 0"
+                              if src_dt.kind == "f" else "0")
+        if op == "isfinite":
+            return self.fresh("unsigned char",
+                              f"__builtin_isfinite((double)({a})) ? 1 : 0"
+                              if src_dt.kind == "f" else "1")
+        if op == "logical_not":
+            return self.fresh("unsigned char", f"(({a}) != 0) ? 0 : 1")
+        if op == "invert":
+            if src_dt == np.dtype(np.bool_):
+                return self.fresh("unsigned char", f"(({a}) != 0) ? 0 : 1")
+            return self.fresh(ct, f"~({a})")
+        raise NotImplementedError(f"unop {op}")
+
+    def _emit_bin(self, e):
+        ct = ctype(e.dtype)
+        op = e.op
+        # pow with small constant integer exponent -> multiply chain
+        # (matches NumPy's small-int fast path)
+        if op == "pow" and isinstance(e.b, ir.Const) \
+                and float(e.b.value) == int(e.b.value) \
+                and 0 <= int(e.b.value) <= 8:
+            n = int(e.b.value)
+            a = self.emit(e.a)
+            av = self.fresh(ctype(_dt(e.a)), f"{a}")
+            if n == 0:
+                return self.fresh(ct, "1")
+            cur = av
+            for _ in range(n - 1):
+                cur2 = self.fresh(ctype(_dt(e.a)), f"({cur}) * ({av})")
+                cur = cur2
+            return self.fresh(ct, f"({ct})({cur})")
+        a = self.emit(e.a)
+        b = self.emit(e.b)
+        adt, bdt = np.dtype(_dt(e.a)), np.dtype(_dt(e.b))
+        f32 = ct == "float"
+        sfx = "f" if f32 else ""
+        infix = {"add": "+", "sub": "-", "mul": "*",
+                 "bitand": "&", "bitor": "|", "bitxor": "^",
+                 "lshift": "<<", "rshift": ">>"}
+        if op in infix:
+            return self.fresh(ct, f"(({ct})({a})) {infix[op]} (({ct})({b}))")
+        cmp = {"gt": ">", "lt": "<", "ge": ">=", "le": "<=", "eq": "==",
+               "ne": "!="}
+        if op in cmp:
+            wt = ctype(np.result_type(adt, bdt))
+            return self.fresh("unsigned char",
+                              f"((({wt})({a})) {cmp[op]} (({wt})({b}))) ? 1 : 0")
+        if op == "div":
+            return self.fresh(ct, f"(({ct})({a})) / (({ct})({b}))")
+        if op == "floordiv":
+            if np.dtype(e.dtype).kind == "f":
+                return self.fresh(
+                    ct, f"floor{sfx}((({ct})({a})) / (({ct})({b})))")
+            return self.fresh(ct, f"rt_fdiv((long long)({a}), "
+                                  f"(long long)({b}))")
+        if op == "mod":
+            if np.dtype(e.dtype).kind == "f":
+                return self.fresh(ct, f"rt_fmod{sfx}(({ct})({a}), "
+                                      f"({ct})({b}))")
+            return self.fresh(ct, f"rt_imod((long long)({a}), "
+                                  f"(long long)({b}))")
+        if op == "pow":
+            if np.dtype(e.dtype).kind in "iu":
+                return self.fresh(ct, f"rt_ipow((long long)({a}), "
+                                      f"(long long)({b}))")
+            return self.fresh(ct, f"pow{sfx}(({ct})({a}), ({ct})({b}))")
+        if op in ("minimum", "maximum"):
+            lt = "<" if op == "minimum" else ">"
+            if np.dtype(e.dtype).kind == "f":
+                # NaN-propagating (NumPy minimum/maximum semantics)
+                return self.fresh(
+                    ct, f"(__builtin_isnan((double)({a})) || (({ct})({a}))"
+                        f" {lt} (({ct})({b}))) ? ({ct})({a}) : ({ct})({b})")
+            return self.fresh(
+                ct, f"((({ct})({a})) {lt} (({ct})({b}))) ? "
+                    f"({ct})({a}) : ({ct})({b})")
+        if op in ("logical_and", "logical_or", "logical_xor"):
+            cop = {"logical_and": "&&", "logical_or": "||"}.get(op)
+            if cop:
+                return self.fresh("unsigned char",
+                                  f"((({a}) != 0) {cop} (({b}) != 0)) ? 1 : 0")
+            return self.fresh("unsigned char",
+                              f"((({a}) != 0) != (({b}) != 0)) ? 1 : 0")
+        raise NotImplementedError(f"binop {op}")
+
+
+def _dt(e):
+    return e.dtype
+
+
+# ---------------------------------------------------------------------------
+# the kernel generator
+# ---------------------------------------------------------------------------
+
+PREAMBLE = r"""
+typedef long long i64;
+typedef unsigned long long u64;
+typedef __attribute__((ext_vector_type(2))) double d2_t;
+typedef __attribute__((ext_vector_type(4))) float f4_t;
+typedef __attribute__((ext_vector_type(2))) long long l2_t;
+typedef __attribute__((ext_vector_type(2))) float f2_t;
+typedef __attribute__((ext_vector_type(2))) int i2_t;
+typedef __attribute__((ext_vector_type(4))) int i4_t;
+typedef __attribute__((ext_vector_type(2))) short s2_t;
+typedef __attribute__((ext_vector_type(4))) short s4_t;
+typedef __attribute__((ext_vector_type(2))) unsigned char b2_t;
+typedef __attribute__((ext_vector_type(4))) unsigned char b4_t;
+typedef __attribute__((ext_vector_type(2))) signed char c2_t;
+typedef __attribute__((ext_vector_type(4))) signed char c4_t;
+
+__device__ __forceinline__ i64 rt_fdiv(i64 a, i64 b) {
+    i64 q = a / b, r = a % b;
+    return (r != 0 && ((r < 0) != (b < 0))) ? q - 1 : q;
+}
+__device__ __forceinline__ i64 rt_imod(i64 a, i64 b) {
+    i64 r = a % b;
+    return (r != 0 && ((r < 0) != (b < 0))) ? r + b : r;
+}
+__device__ __forceinline__ double rt_fmod(double a, double b) {
+    double r = fmod(a, b);
+    return (r != 0.0 && ((r < 0.0) != (b < 0.0))) ? r + b : r;
+}
+__device__ __forceinline__ float rt_fmodf(float a, float b) {
+    float r = fmodf(a, b);
+    return (r != 0.0f && ((r < 0.0f) != (b < 0.0f))) ? r + b : r;
+}
+__device__ __forceinline__ i64 rt_ipow(i64 b, i64 e) {
+    i64 r = 1;
+    while (e > 0) { if (e & 1) r *= b; b *= b; e >>= 1; }
+    return r;
+}
+"""
+
+
+class GeneratedKernel:
+    """Source + arg-packing recipe for one fused-group structure."""
+
+    def __init__(self, key, source, kname_main, kname_finish, finish_source,
+                 fields, vec, nd, nred):
+        self.key = key
+        self.source = source
+        self.kname_main = kname_main
+        self.kname_finish = kname_finish
+        self.finish_source = finish_source
+        self.fields = fields     # list of (kind, payload) for packing
+        self.vec = vec
+        self.nd = nd
+        self.nred = nred
+        self.handle = None
+        self.finish_handle = None
+
+
+def structural_key(plan, classes, vec):
+    h = hashlib.sha256()
+    h.update(f"nd={len(plan.itershape)};vec={vec};".encode())
+    for c in classes:
+        h.update(f"op:{c.name}:{c.dtype}:{c.inner}:{c.outer_g};".encode())
+    for n, (v, dt) in sorted(plan.scalars.items()):
+        h.update(f"sc:{n}:{dt};".encode())
+    for n, dt in sorted(plan.dead_vars.items()):
+        h.update(f"dv:{n}:{dt};".encode())
+    for st in plan.statements:
+        h.update(repr((st.target, st.expr)).encode())
+    for r in plan.reductions:
+        h.update(f"red:{r.acc}:{r.kind}:{r.dtype};".encode())
+    return h.hexdigest()[:24]
+
+
+class KernelGen:
+    def __init__(self, plan, classes, vec):
+        self.plan = plan
+        self.classes = {c.name: c for c in classes}
+        self.class_list = classes
+        self.vec = vec
+        self.nd = len(plan.itershape)
+        self.acc_specs = {r.acc: r for r in plan.reductions}
+        # SSA
+        self.ssa, self.final_version = ssa_statements(
+            plan.statements, set(self.acc_specs))
+        self.wanted = {}
+        for (_, _, e) in self.ssa:
+            collect_nodes(e, self.wanted)
+        # which operands are read (version-0 refs used) / written
+        self.read_ops = set()
+        self.written_ops = {}
+        for node in self.wanted:
+            if isinstance(node, ir.Ref):
+                base, v = node.name.split("@")
+                if base in self.classes and int(v) == 0:
+                    self.read_ops.add(base)
+        for (tgt, accn, _) in self.ssa:
+            if tgt is not None:
+                base, v = tgt.split("@")
+                if base in self.classes:
+                    self.written_ops[base] = max(
+                        self.written_ops.get(base, 0), int(v))
+
+    # -- per-lane value resolution -----------------------------------------
+
+    def lane_value(self, em, versioned, dtype):
+        base, v = versioned.split("@")
+        if int(v) == 0:
+            if base in self.classes:
+                return em.gen._load_name(em, base)
+            if base in self.plan.dead_vars:
+                raise AssertionError(f"dead var {base} read before write")
+            raise AssertionError(f"unknown var {versioned}")
+        return f"r_{base}_{v}{em.tag}"
+
+    def _load_name(self, em, name):
+        return f"ld_{name}{em.tag}"
+
+    # -- addressing ---------------------------------------------------------
+
+    def addr_expr(self, name, inner_expr):
+        """element index expression into operand `name`'s buffer."""
+        c = self.classes[name]
+        parts = [f"a.{name}_off"]
+        for d in range(self.nd - 1):
+            if c.outer_g[d]:
+                parts.append(f"i{d} * a.{name}_s{d}")
+        op = next(o for o in self.plan.operands if o.name == name)
+        inner = "v" if c.inner in ("v", "u") else (
+            None if c.inner == "z" else "g")
+        if c.inner in ("v", "u"):
+            parts.append(f"({inner_expr})")
+        elif c.inner == "g":
+            parts.append(f"({inner_expr}) * a.{name}_sx")
+        return " + ".join(parts)
+
+    # -- body generation ------------------------------------------------------
+
+    def gen_lane_body(self, tag, inner_expr, indent, vec_lane=None):
+        """Emit loads + statements + stores for ONE element.
+
+        vec_lane: (vecvar-suffix, lane index) when inside the vectorised
+        body -- loads come from preloaded vectors."""
+        idx = []
+        for d in range(self.nd - 1):
+            idx.append(f"a.gs{d} + i{d}")
+        idx.append(f"a.gs{self.nd-1} + ({inner_expr})" if self.nd else "0")
+        em = LaneEmitter(self, tag, idx)
+        # loads
+        for name in sorted(self.read_ops):
+            c = self.classes[name]
+            ct = ctype(c.dtype)
+            ldv = self._load_name(em, name)
+            if vec_lane is not None and c.inner == "v":
+                em.lines.append(
+                    f"      {ct} {ldv} = vv_{name}[{vec_lane}];")
+            else:
+                em.lines.append(
+                    f"      {ct} {ldv} = a.{name}_p["
+                    f"{self.addr_expr(name, inner_expr)}];")
+        # statements
+        last_val = {}
+        for (tgt, accinfo, expr) in self.ssa:
+            if accinfo is not None:
+                accn, comb = accinfo
+                spec = self.acc_specs[accn]
+                src = em.emit(expr)
+                ct = ctype(spec.dtype)
+                em.lines.append(
+                    f"      acc_{accn} = "
+                    f"{self.comb_expr(comb, spec.dtype, f'acc_{accn}', src)};")
+            else:
+                base, v = tgt.split("@")
+                val = em.emit(expr)
+                dt = self.classes[base].dtype if base in self.classes \
+                    else self.plan.dead_vars[base]
+                ct = ctype(dt)
+                em.lines.append(f"      {ct} r_{base}_{v}{tag} = "
+                                f"({ct})({val});")
+                last_val[base] = f"r_{base}_{v}{tag}"
+        # stores
+        for name, lastv in sorted(self.written_ops.items()):
+            c = self.classes[name]
+            final = f"r_{name}_{lastv}{tag}"
+            if vec_lane is not None and c.inner == "v":
+                em.lines.append(f"      sv_{name}[{vec_lane}] = {final};")
+            else:
+                em.lines.append(
+                    f"      a.{name}_p[{self.addr_expr(name, inner_expr)}]"
+                    f" = {final};")
+        pad = " " * (indent - 6)
+        return "\n".join(pad + ln.lstrip() if False else
+                         (" " * indent) + ln.strip() for ln in em.lines)
+
+    def comb_expr(self, comb, dtype, a, b):
+        ct = ctype(dtype)
+        if comb == "add":
+            return f"({a}) + ({b})"
+        if comb == "mul":
+            return f"({a}) * ({b})"
+        if comb in ("minimum", "maximum"):
+            lt = "<" if comb == "minimum" else ">"
+            if np.dtype(dtype).kind == "f":
+                return (f"(__builtin_isnan((double)({a})) || (({a}) {lt} "
+                        f"({b}))) ? ({ct})({a}) : ({ct})({b})")
+            return f"(({a}) {lt} ({b})) ? ({ct})({a}) : ({ct})({b})"
+        if comb == "logical_and":
+            return f"((({a}) != 0) && (({b}) != 0)) ? 1 : 0"
+        if comb == "logical_or":
+            return f"((({a}) != 0) || (({b}) != 0)) ? 1 : 0"
+        raise NotImplementedError(comb)
+
+    # -- full source -----------------------------------------------------------
+
+    def generate(self, key):
+        nd, V = self.nd, self.vec
+        L = []
+        L.append(PREAMBLE)
+        # args struct
+        L.append("struct Args {")
+        fields = []
+        for d in range(nd):
+            L.append(f"  i64 n{d};")
+            fields.append(("iter_n", d))
+        for d in range(nd):
+            L.append(f"  i64 gs{d};")
+            fields.append(("iter_gs", d))
+        for c in self.class_list:
+            ct = ctype(c.dtype)
+            L.append(f"  {ct}* {c.name}_p;")
+            fields.append(("ptr", c.name))
+            L.append(f"  i64 {c.name}_off;")
+            fields.append(("off", c.name))
+            for d in range(nd - 1):
+                if c.outer_g[d]:
+                    L.append(f"  i64 {c.name}_s{d};")
+                    fields.append(("stride", (c.name, d)))
+            if c.inner == "g":
+                L.append(f"  i64 {c.name}_sx;")
+                fields.append(("stride", (c.name, nd - 1)))
+        for n in sorted(self.plan.scalars):
+            _, dt = self.plan.scalars[n]
+            if np.dtype(dt).kind == "f":
+                L.append(f"  double {n};")
+                fields.append(("scalar_f", n))
+            else:
+                L.append(f"  i64 {n};")
+                fields.append(("scalar_i", n))
+        nred = len(self.plan.reductions)
+        if nred:
+            L.append("  char* partials;")
+            fields.append(("partials", None))
+            L.append("  i64 npartials;")
+            fields.append(("npartials", None))
+        L.append("};")
+
+        kname = f"k_{key}"
+        L.append(f'extern "C" __global__ void __launch_bounds__(256) '
+                 f"{kname}(Args a) {{")
+        # accumulators
+        for spec in self.plan.reductions:
+            ct = ctype(spec.dtype)
+            init = self.init_literal(spec)
+            L.append(f"  {ct} acc_{spec.acc} = {init};")
+        # outer loops
+        ind = 2
+        if nd >= 3:
+            L.append(f"{' '*ind}for (i64 i0 = blockIdx.z; i0 < a.n0; "
+                     f"i0 += gridDim.z) {{")
+            ind += 2
+        if nd >= 2:
+            oy = nd - 2
+            L.append(f"{' '*ind}for (i64 i{oy} = blockIdx.y; i{oy} < a.n{oy};"
+                     f" i{oy} += gridDim.y) {{")
+            ind += 2
+        x = nd - 1
+        L.append(f"{' '*ind}i64 vb = ((i64)blockIdx.x * 256 + threadIdx.x) "
+                 f"* {V};")
+        L.append(f"{' '*ind}const i64 xs = (i64)gridDim.x * 256 * {V};")
+        L.append(f"{' '*ind}for (; vb + {V} <= a.n{x}; vb += xs) {{")
+        # vector preloads
+        body_ind = ind + 2
+        for name in sorted(self.read_ops):
+            c = self.classes[name]
+            if c.inner == "v":
+                vt = self.vec_type(c.dtype)
+                L.append(f"{' '*body_ind}const {vt} vv_{name} = "
+                         f"*(const {vt}*)&a.{name}_p["
+                         f"{self.addr_expr(name, 'vb')}];")
+        for name in sorted(self.written_ops):
+            c = self.classes[name]
+            if c.inner == "v":
+                vt = self.vec_type(c.dtype)
+                L.append(f"{' '*body_ind}{vt} sv_{name};")
+        for lane in range(V):
+            L.append(f"{' '*body_ind}{{ // lane {lane}")
+            L.append(self.gen_lane_body(f"_L{lane}", f"vb + {lane}",
+                                        body_ind + 2, vec_lane=lane))
+            L.append(f"{' '*body_ind}}}")
+        for name, _ in sorted(self.written_ops.items()):
+            c = self.classes[name]
+            if c.inner == "v":
+                vt = self.vec_type(c.dtype)
+                L.append(f"{' '*body_ind}*({vt}*)&a.{name}_p["
+                         f"{self.addr_expr(name, 'vb')}] = sv_{name};")
+        L.append(f"{' '*ind}}}")
+        # tail
+        if V > 1:
+            L.append(f"{' '*ind}const i64 tstart = a.n{x} / {V} * {V};")
+            L.append(f"{' '*ind}if (vb == tstart && tstart < a.n{x}) {{")
+            L.append(f"{' '*ind}  for (i64 te = tstart; te < a.n{x}; ++te) {{")
+            L.append(self.gen_lane_body("_T", "te", ind + 4))
+            L.append(f"{' '*ind}  }}")
+            L.append(f"{' '*ind}}}")
+        if nd >= 2:
+            ind -= 2
+            L.append(f"{' '*ind}}}")
+        if nd >= 3:
+            ind -= 2
+            L.append(f"{' '*ind}}}")
+        # reduction epilogue: wave shfl tree + LDS across 4 waves
+        if nred:
+            L.append("  {")
+            L.append("    const int lane = threadIdx.x & 63;")
+            L.append("    const int wid = threadIdx.x >> 6;")
+            for spec in self.plan.reductions:
+                ct = ctype(spec.dtype)
+                an = f"acc_{spec.acc}"
+                comb, _ = ir.REDUCTIONS[spec.kind]
+                L.append(f"    for (int o = 32; o > 0; o >>= 1) {{")
+                L.append(f"      {ct} other = {self.shfl_down(spec.dtype, an, 'o')};")
+                L.append(f"      {an} = "
+                         f"{self.comb_expr(comb, spec.dtype, an, 'other')};")
+                L.append("    }")
+                L.append(f"    __shared__ {ct} lds_{spec.acc}[4];")
+                L.append(f"    if (lane == 0) lds_{spec.acc}[wid] = {an};")
+            L.append("    __syncthreads();")
+            L.append("    if (threadIdx.x == 0) {")
+            L.append("      i64 pidx = (i64)blockIdx.x + (i64)gridDim.x * "
+                     "((i64)blockIdx.y + (i64)gridDim.y * blockIdx.z);")
+            for ri, spec in enumerate(self.plan.reductions):
+                ct = ctype(spec.dtype)
+                comb, _ = ir.REDUCTIONS[spec.kind]
+                L.append(f"      {ct} v{ri} = lds_{spec.acc}[0];")
+                for w in range(1, 4):
+                    L.append(f"      v{ri} = "
+                             f"{self.comb_expr(comb, spec.dtype, f'v{ri}', f'lds_{spec.acc}[{w}]')};")
+                L.append(f"      *({ct}*)(a.partials + (i64){ri} * "
+                         f"a.npartials * 8 + pidx * 8) = v{ri};")
+            L.append("    }")
+            L.append("  }")
+        L.append("}")
+
+        kname_finish = None
+        finish_source = None
+        if nred:
+            kname_finish = f"kf_{key}"
+            finish_source = PREAMBLE + "\n" + self.gen_finish(kname_finish)
+        return "\n".join(L), kname, kname_finish, finish_source, fields
+
+    def vec_type(self, dt):
+        ct = ctype(dt)
+        V = self.vec
+        m = {("double", 2): "d2_t", ("float", 4): "f4_t",
+             ("long long", 2): "l2_t", ("float", 2): "f2_t",
+             ("int", 2): "i2_t", ("int", 4): "i4_t",
+             ("short", 2): "s2_t", ("short", 4): "s4_t",
+             ("unsigned char", 2): "b2_t", ("unsigned char", 4): "b4_t",
+             ("signed char", 2): "c2_t", ("signed char", 4): "c4_t"}
+        vt = m.get((ct, V))
+        assert vt is not None, f"no vector type for {ct} x{V}"
+        return vt
+
+    def shfl_down(self, dtype, var, off):
+        if np.dtype(dtype).itemsize < 4:
+            ct = ctype(dtype)
+            return f"({ct})__shfl_down((int)({var}), {off}, 64)"
+        return f"__shfl_down({var}, {off}, 64)"
+
+    def init_literal(self, spec):
+        v = ir.reduction_init(spec.kind, spec.dtype)
+        ct = ctype(spec.dtype)
+        if isinstance(v, float) and np.isinf(v):
+            return ("-__builtin_inf()" if v < 0 else "__builtin_inf()") \
+                if ct == "double" else \
+                ("-__builtin_inff()" if v < 0 else "__builtin_inff()")
+        if isinstance(v, bool):
+            return "1" if v else "0"
+        return f"({ct})({v!r})"
+
+    def gen_finish(self, kname):
+        """One-block finish kernel folding the per-workgroup partials into
+        per-rank outputs (device scalars)."""
+        L = []
+        L.append("struct FArgs { char* partials; i64 npartials; "
+                 + " ".join(f"char* out{i};" for i in
+                            range(len(self.plan.reductions)))
+                 + " };")
+        L.append(f'extern "C" __global__ void __launch_bounds__(256) '
+                 f"{kname}(FArgs f) {{")
+        L.append("  const int lane = threadIdx.x & 63;")
+        L.append("  const int wid = threadIdx.x >> 6;")
+        for ri, spec in enumerate(self.plan.reductions):
+            ct = ctype(spec.dtype)
+            comb, _ = ir.REDUCTIONS[spec.kind]
+            init = self.init_literal(spec)
+            L.append(f"  {{")
+            L.append(f"    {ct} acc = {init};")
+            L.append(f"    for (i64 i = threadIdx.x; i < f.npartials; "
+                     f"i += 256) {{")
+            L.append(f"      {ct} pv = *({ct}*)(f.partials + (i64){ri} * "
+                     f"f.npartials * 8 + i * 8);")
+            L.append(f"      acc = {self.comb_expr(comb, spec.dtype, 'acc', 'pv')};")
+            L.append("    }")
+            L.append("    for (int o = 32; o > 0; o >>= 1) {")
+            L.append(f"      {ct} other = {self.shfl_down(spec.dtype, 'acc', 'o')};")
+            L.append(f"      acc = {self.comb_expr(comb, spec.dtype, 'acc', 'other')};")
+            L.append("    }")
+            L.append(f"    __shared__ {ct} lds[4];")
+            L.append("    if (lane == 0) lds[wid] = acc;")
+            L.append("    __syncthreads();")
+            L.append("    if (threadIdx.x == 0) {")
+            L.append(f"      {ct} v = lds[0];")
+            for w in range(1, 4):
+                L.append(f"      v = {self.comb_expr(comb, spec.dtype, 'v', f'lds[{w}]')};")
+            L.append(f"      *({ct}*)f.out{ri} = v;")
+            L.append("    }")
+            L.append("    __syncthreads();")
+            L.append("  }")
+        L.append("}")
+        return "\n".join(L)
+
+
+def generate(plan):
+    """plan -> GeneratedKernel (source + packing recipe)."""
+    nd = len(plan.itershape)
+    if nd < 1 or nd > 3:
+        raise NotImplementedError(f"{nd}-d iteration spaces")
+    # vector width from the widest dtype among operands (16B per lane target)
+    max_es = 8
+    sizes = [np.dtype(o.dtype).itemsize for o in plan.operands] or [8]
+    max_es = max(sizes)
+    vec = 2 if max_es == 8 else (4 if max_es == 4 else 1)
+    classes = classify_plan(plan, vec)
+    # if nothing vectorises, a vec build only costs code size; keep vec
+    key = structural_key(plan, classes, vec)
+    gen = KernelGen(plan, classes, vec)
+    source, kmain, kfinish, finish_source, fields = gen.generate(key)
+    return GeneratedKernel(key, source, kmain, kfinish, finish_source,
+                           fields, vec, nd, len(plan.reductions))
+
+
+def pack_args(gk, plan, ptr_of):
+    """Pack the Args struct per gk.fields.  ptr_of(name) -> device address
+    of an operand buffer; special names '__partials__'."""
+    out = bytearray()
+    opmap = {o.name: o for o in plan.operands}
+    for kind, payload in gk.fields:
+        if kind == "iter_n":
+            out += struct.pack("<q", plan.itershape[payload])
+        elif kind == "iter_gs":
+            out += struct.pack("<q", plan.global_start[payload])
+        elif kind == "ptr":
+            out += struct.pack("<Q", ptr_of(payload))
+        elif kind == "off":
+            out += struct.pack("<q", opmap[payload].offset0)
+        elif kind == "stride":
+            name, d = payload
+            out += struct.pack("<q", opmap[name].strides[d])
+        elif kind == "scalar_f":
+            v, dt = plan.scalars[payload]
+            out += struct.pack("<d", float(v))
+        elif kind == "scalar_i":
+            v, dt = plan.scalars[payload]
+            out += struct.pack("<q", int(v))
+        elif kind == "partials":
+            out += struct.pack("<Q", ptr_of("__partials__"))
+        elif kind == "npartials":
+            out += struct.pack("<q", ptr_of("__npartials__"))
+        else:
+            raise AssertionError(kind)
+    return bytes(out)
+
+
+def pack_finish_args(partials_ptr, npartials, out_ptrs):
+    out = bytearray()
+    out += struct.pack("<Q", partials_ptr)
+    out += struct.pack("<q", npartials)
+    for p in out_ptrs:
+        out += struct.pack("<Q", p)
+    return bytes(out)

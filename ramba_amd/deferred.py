@@ -214,7 +214,7 @@ def _subst(group, tree, reads_out):
     if isinstance(tree, (numbers.Number, np.bool_, np.number)):
         name, dt = group.scalar_var(tree)
         return ir.ScalarArg(name, dt)
-    if isinstance(tree, ir.Iota):
+    if isinstance(tree, (ir.Iota, ir.Const)):
         return tree
     if isinstance(tree, ir.Bin):
         a = _subst(group, tree.a, reads_out)

@@ -1,0 +1,339 @@
+"""HIP product backend: gfx950 kernels via the C-ABI runtime.
+
+Torch is plumbing only — device allocation (shard containers as raw device
+tensors), the HIP stream, and the RCCL process group.  Every compute kernel
+is our own HIP: fused groups JIT-compiled by `codegen` + `rt_kernel_get`
+(hiprtc, gfx950), box copies by `rt_copy_box`.  FAILS LOUDLY if the
+extension or a GPU is missing — no CPU fallback exists in the product path.
+"""
+
+import ctypes
+import os
+
+import numpy as np
+
+from . import codegen
+from .common import dprint
+from .shardview import box_shape
+
+LIBPATH = os.path.join(os.path.dirname(os.path.abspath(__file__)), "_lib",
+                       "libramba_rt.so")
+
+TORCH_DTYPE = None  # filled at init
+
+
+def _load_lib():
+    if not os.path.exists(LIBPATH):
+        raise RuntimeError(
+            f"ramba_amd HIP runtime not built: {LIBPATH} missing. "
+            "Run `python __graft_entry__.py build` (hipcc, gfx950). "
+            "There is no CPU fallback in the product path.")
+    lib = ctypes.CDLL(LIBPATH)
+    lib.rt_init.argtypes = [ctypes.c_int]
+    lib.rt_device_count.restype = ctypes.c_int
+    lib.rt_last_error.restype = ctypes.c_char_p
+    lib.rt_kernel_get.argtypes = [ctypes.c_char_p, ctypes.c_char_p,
+                                  ctypes.c_char_p,
+                                  ctypes.POINTER(ctypes.c_void_p)]
+    lib.rt_launch.argtypes = [ctypes.c_void_p, ctypes.c_uint, ctypes.c_uint,
+                              ctypes.c_uint, ctypes.c_uint, ctypes.c_size_t,
+                              ctypes.c_char_p, ctypes.c_size_t]
+    lib.rt_copy_box.argtypes = [ctypes.c_size_t, ctypes.c_void_p,
+                                ctypes.c_void_p, ctypes.c_int,
+                                ctypes.POINTER(ctypes.c_int64),
+                                ctypes.POINTER(ctypes.c_int64),
+                                ctypes.POINTER(ctypes.c_int64),
+                                ctypes.c_int64, ctypes.c_int64, ctypes.c_int]
+    lib.rt_stream_sync.argtypes = [ctypes.c_size_t]
+    return lib
+
+
+def _i64arr(vals):
+    return (ctypes.c_int64 * len(vals))(*[int(v) for v in vals])
+
+
+class HipBackend:
+    name = "hip"
+
+    def __init__(self, device=None):
+        global TORCH_DTYPE
+        import torch
+        self.torch = torch
+        if not torch.cuda.is_available():
+            raise RuntimeError(
+                "ramba_amd HIP backend requires an AMD GPU "
+                "(torch.cuda.is_available() is False); no CPU fallback.")
+        self.lib = _load_lib()
+        self.device = int(os.environ.get("LOCAL_RANK", "0")) \
+            if device is None else device
+        torch.cuda.set_device(self.device)
+        self._check(self.lib.rt_init(self.device), "rt_init")
+        TORCH_DTYPE = {
+            np.dtype(np.float64): torch.float64,
+            np.dtype(np.float32): torch.float32,
+            np.dtype(np.int64): torch.int64,
+            np.dtype(np.int32): torch.int32,
+            np.dtype(np.int16): torch.int16,
+            np.dtype(np.int8): torch.int8,
+            np.dtype(np.uint8): torch.uint8,
+            np.dtype(np.bool_): torch.uint8,
+        }
+        self.containers = {}       # gid -> torch tensor
+        self.temps = {}            # name -> torch tensor
+        self.kernels = {}          # structural key -> GeneratedKernel
+        self.rt = None
+        self._dist = None
+        # optional HIP-event kernel timing (bench roofline leg)
+        self.time_kernels = False
+        self.kernel_times_ms = []
+
+    # ------------------------------------------------------------------
+    def _check(self, rc, what):
+        if rc != 0:
+            raise RuntimeError(
+                f"{what} failed: {self.lib.rt_last_error().decode()}")
+
+    def attach(self, rt):
+        self.rt = rt
+
+    def _stream(self):
+        return self.torch.cuda.current_stream().cuda_stream
+
+    def init_process_group(self, rank, world):
+        import torch.distributed as dist
+        self._dist = dist
+        if not dist.is_initialized():
+            dist.init_process_group("nccl", rank=rank, world_size=world)
+
+    def _d(self):
+        if self._dist is None:
+            import torch.distributed as dist
+            self._dist = dist
+        return self._dist
+
+    # -- memory -------------------------------------------------------------
+
+    def _tdt(self, dtype):
+        return TORCH_DTYPE[np.dtype(dtype)]
+
+    def alloc_container(self, bd, rt):
+        _, cshape, _, _ = rt.shard_geometry(bd)
+        if cshape is None:
+            self.containers[bd.gid] = None
+            return
+        self.containers[bd.gid] = self.torch.empty(
+            cshape, dtype=self._tdt(bd.dtype), device="cuda")
+
+    def free_container(self, bd):
+        self.containers.pop(bd.gid, None)
+
+    def alloc_temp(self, name, shape, dtype):
+        self.temps[name] = self.torch.empty(
+            shape, dtype=self._tdt(dtype), device="cuda")
+
+    def free_temps(self):
+        self.temps.clear()
+
+    # -- box copies -----------------------------------------------------------
+
+    def _cont(self, bd):
+        c = self.containers.get(bd.gid)
+        assert c is not None, f"shard for gid {bd.gid} not allocated"
+        return c
+
+    def _strides_of(self, t):
+        return tuple(t.stride())
+
+    def _copy(self, dst_t, dst_strides, dst_off, src_t, src_strides, src_off,
+              shape, elemsize):
+        nd = len(shape)
+        self._check(self.lib.rt_copy_box(
+            self._stream(), ctypes.c_void_p(dst_t.data_ptr()),
+            ctypes.c_void_p(src_t.data_ptr()), nd, _i64arr(shape),
+            _i64arr(dst_strides), _i64arr(src_strides),
+            int(dst_off), int(src_off), elemsize), "rt_copy_box")
+
+    def _box_off(self, bd, rt, box):
+        d = bd.divisions[rt.rank]
+        cont = self._cont(bd)
+        cs = cont.stride()
+        off = 0
+        for i in range(len(bd.shape)):
+            off += (int(box[0, i] - d[0, i]) + bd.border) * cs[i]
+        return off, cs
+
+    def copy_container_to_temp(self, bd, rt, part_box, vname, need_box):
+        cont = self._cont(bd)
+        tmp = self.temps[vname]
+        shape = box_shape(part_box)
+        src_off, cs = self._box_off(bd, rt, part_box)
+        lo = part_box[0] - need_box[0]
+        ts = tmp.stride()
+        dst_off = sum(int(lo[i]) * ts[i] for i in range(len(shape)))
+        self._copy(tmp, ts, dst_off, cont, cs, src_off, shape,
+                   cont.element_size())
+
+    def pack_box(self, bd, rt, box):
+        cont = self._cont(bd)
+        shape = box_shape(box)
+        msg = self.torch.empty(shape, dtype=cont.dtype, device="cuda")
+        src_off, cs = self._box_off(bd, rt, box)
+        self._copy(msg, msg.stride(), 0, cont, cs, src_off, shape,
+                   cont.element_size())
+        return msg
+
+    def new_message_buffer(self, shape, dtype):
+        return self.torch.empty(shape, dtype=self._tdt(dtype), device="cuda")
+
+    def exchange(self, sends, recvs):
+        if not sends and not recvs:
+            return
+        dist = self._d()
+        self.torch.cuda.synchronize()  # pack kernels complete before NCCL
+        ops = []
+        for (dst, buf) in sends:
+            ops.append(dist.P2POp(dist.isend, buf, dst))
+        for (src, buf) in recvs:
+            ops.append(dist.P2POp(dist.irecv, buf, src))
+        for req in dist.batch_isend_irecv(ops):
+            req.wait()
+
+    def unpack_box_to_container(self, bd, rt, box, buf):
+        cont = self._cont(bd)
+        shape = box_shape(box)
+        dst_off, cs = self._box_off(bd, rt, box)
+        self._copy(cont, cs, dst_off, buf, buf.stride(), 0, shape,
+                   cont.element_size())
+
+    def unpack_box_to_temp(self, vname, need_box, box, buf):
+        tmp = self.temps[vname]
+        shape = box_shape(box)
+        lo = box[0] - need_box[0]
+        ts = tmp.stride()
+        dst_off = sum(int(lo[i]) * ts[i] for i in range(len(shape)))
+        self._copy(tmp, ts, dst_off, buf, buf.stride(), 0, shape,
+                   tmp.element_size())
+
+    def box_to_numpy(self, bd, rt, box):
+        return self.pack_box(bd, rt, box).cpu().numpy().astype(
+            bd.dtype, copy=False).reshape(box_shape(box))
+
+    def write_core_from_numpy(self, bd, rt, nparr):
+        cont = self._cont(bd)
+        src = self.torch.from_numpy(
+            np.ascontiguousarray(nparr).view(
+                np.uint8 if bd.dtype == np.bool_ else bd.dtype)).to("cuda")
+        core = rt.core_box(bd, rt.rank)
+        dst_off, cs = self._box_off(bd, rt, core)
+        self._copy(cont, cs, dst_off, src, src.stride(), 0,
+                   tuple(src.shape), cont.element_size())
+
+    # -- collectives -----------------------------------------------------------
+
+    def bcast_numpy(self, obj, root):
+        if self.rt.world == 1:
+            return obj
+        dist = self._d()
+        lst = [obj]
+        dist.broadcast_object_list(lst, src=root)
+        return lst[0]
+
+    _RED_MAP = {"sum": "SUM", "prod": "PRODUCT", "min": "MIN", "max": "MAX",
+                "all": "MIN", "any": "MAX"}
+
+    def allreduce(self, value, kind):
+        dist = self._d()
+        v = np.asarray(value)
+        dt = v.dtype
+        if dt == np.bool_:
+            v = v.astype(np.uint8)
+        t = self.torch.from_numpy(v.reshape(1).copy()).to("cuda")
+        dist.all_reduce(t, op=getattr(dist.ReduceOp, self._RED_MAP[kind]))
+        out = t.cpu().numpy()[0]
+        if dt == np.bool_:
+            out = bool(out)
+        return np.asarray(out, dtype=dt)[()]
+
+    # -- kernel execution -------------------------------------------------------
+
+    def _get_kernel(self, plan):
+        gk = codegen.generate(plan)
+        cached = self.kernels.get(gk.key)
+        if cached is not None:
+            return cached
+        if int(os.environ.get("RAMBA_SHOW_CODE", "0")):
+            print(f"=== kernel {gk.key} ===\n{gk.source}\n", flush=True)
+        h = ctypes.c_void_p()
+        self._check(self.lib.rt_kernel_get(
+            gk.key.encode(), gk.source.encode(), gk.kname_main.encode(),
+            ctypes.byref(h)), "rt_kernel_get")
+        gk.handle = h.value
+        if gk.kname_finish:
+            hf = ctypes.c_void_p()
+            self._check(self.lib.rt_kernel_get(
+                (gk.key + "_f").encode(), gk.finish_source.encode(),
+                gk.kname_finish.encode(), ctypes.byref(hf)),
+                "rt_kernel_get(finish)")
+            gk.finish_handle = hf.value
+        self.kernels[gk.key] = gk
+        return gk
+
+    def launch(self, plan):
+        gk = self._get_kernel(plan)
+        nd = gk.nd
+        shape = plan.itershape
+        V = gk.vec
+        nx = shape[nd - 1]
+        gx = max(1, min(2048, (nx + 256 * V - 1) // (256 * V)))
+        gy = max(1, min(8192, shape[nd - 2])) if nd >= 2 else 1
+        gz = max(1, min(64, shape[0])) if nd >= 3 else 1
+
+        partials = None
+        outs = []
+        np_partials = gx * gy * gz
+        if gk.nred:
+            partials = self.torch.empty(gk.nred * np_partials,
+                                        dtype=self.torch.float64,
+                                        device="cuda")
+            for spec in plan.reductions:
+                outs.append(self.torch.empty(
+                    1, dtype=self._tdt(spec.dtype), device="cuda"))
+
+        def ptr_of(name):
+            if name == "__partials__":
+                return partials.data_ptr()
+            if name == "__npartials__":
+                return np_partials
+            if name in self.temps:
+                return self.temps[name].data_ptr()
+            op = next(o for o in plan.operands if o.name == name)
+            return self._cont(op.bd).data_ptr()
+
+        args = codegen.pack_args(gk, plan, ptr_of)
+        stream = self._stream()
+        if self.time_kernels:
+            ev0 = self.torch.cuda.Event(enable_timing=True)
+            ev1 = self.torch.cuda.Event(enable_timing=True)
+            ev0.record()
+        self._check(self.lib.rt_launch(
+            ctypes.c_void_p(gk.handle), gx, gy, gz, 256, stream, args,
+            len(args)), "rt_launch")
+        if self.time_kernels:
+            ev1.record()
+            ev1.synchronize()
+            self.kernel_times_ms.append(ev0.elapsed_time(ev1))
+        results = []
+        if gk.nred:
+            fargs = codegen.pack_finish_args(
+                partials.data_ptr(), np_partials,
+                [o.data_ptr() for o in outs])
+            self._check(self.lib.rt_launch(
+                ctypes.c_void_p(gk.finish_handle), 1, 1, 1, 256, stream,
+                fargs, len(fargs)), "rt_launch(finish)")
+            for spec, o in zip(plan.reductions, outs):
+                v = o.cpu().numpy()[0]
+                results.append(np.asarray(v, dtype=spec.dtype)[()])
+        return results
+
+    def sync(self):
+        self.torch.cuda.synchronize()

@@ -64,6 +64,15 @@ class ScalarArg(Expr):
 
 
 @dataclass(frozen=True)
+class Const(Expr):
+    """A compile-time constant baked into the kernel source (used for small
+    integer pow exponents so `x**2` lowers to `x*x`, matching NumPy's
+    small-int fast path)."""
+    value: Any
+    dtype: Any
+
+
+@dataclass(frozen=True)
 class Iota(Expr):
     """index[axis] + global_start[axis] (int64) — the arange codeline
     (ramba/ramba.py:8952-8960), the bit-exactness anchor."""

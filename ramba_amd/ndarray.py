@@ -210,6 +210,11 @@ class ndarray:
         shape = np.broadcast_shapes(_shape_of(a), _shape_of(b))
         dt = _result_dtype(op, a, b) if out_dtype is None else out_dtype
         a, b = _bcast(a, shape), _bcast(b, shape)
+        if op == "pow" and isinstance(b, (int, np.integer)) \
+                and not isinstance(b, (bool, np.bool_)) and 0 <= b <= 8:
+            # small-int exponent baked as a compile-time constant so the
+            # kernel emits a multiply chain (x**2 == x*x, NumPy's fast path)
+            b = ir.Const(int(b), ir.I64)
         out = _new_result(shape, dt)
         deferred.add_op(out, "=", ir.Bin(op, a, b, dt), empty_like=empty_like)
         return out
