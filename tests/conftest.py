@@ -25,7 +25,18 @@ def ra():
 def ra_gpu():
     """ramba_amd initialised with the HIP product backend."""
     import ramba_amd
-    ramba_amd.init()  # product default: HIP, fails loudly without GPU
+    from ramba_amd import deferred
+    if not ramba_amd._initialized["done"]:
+        ramba_amd.init()  # product default: HIP, fails loudly without GPU
+    else:
+        rt = deferred.get_runtime()
+        if getattr(rt.backend, "name", "") != "hip":
+            # a CPU-backend test ran first in this process; swap in HIP
+            from ramba_amd.hip_backend import HipBackend
+            deferred.flush()
+            be = HipBackend()
+            rt.backend = be
+            be.attach(rt)
     return ramba_amd
 
 
