@@ -777,16 +777,19 @@ class KernelGen:
         return "\n".join(L)
 
 
+def decide_vec(plan):
+    """Vector width from the widest operand dtype (16 B per lane target)."""
+    sizes = [np.dtype(o.dtype).itemsize for o in plan.operands] or [8]
+    max_es = max(sizes)
+    return 2 if max_es == 8 else (4 if max_es == 4 else 1)
+
+
 def generate(plan):
     """plan -> GeneratedKernel (source + packing recipe)."""
     nd = len(plan.itershape)
     if nd < 1 or nd > 3:
         raise NotImplementedError(f"{nd}-d iteration spaces")
-    # vector width from the widest dtype among operands (16B per lane target)
-    max_es = 8
-    sizes = [np.dtype(o.dtype).itemsize for o in plan.operands] or [8]
-    max_es = max(sizes)
-    vec = 2 if max_es == 8 else (4 if max_es == 4 else 1)
+    vec = decide_vec(plan)
     classes = classify_plan(plan, vec)
     # if nothing vectorises, a vec build only costs code size; keep vec
     key = structural_key(plan, classes, vec)

@@ -257,10 +257,15 @@ class HipBackend:
     # -- kernel execution -------------------------------------------------------
 
     def _get_kernel(self, plan):
-        gk = codegen.generate(plan)
-        cached = self.kernels.get(gk.key)
+        # fast path: structural key without source generation
+        vec = codegen.decide_vec(plan)
+        classes = codegen.classify_plan(plan, vec)
+        key = codegen.structural_key(plan, classes, vec)
+        cached = self.kernels.get(key)
         if cached is not None:
             return cached
+        gk = codegen.generate(plan)
+        assert gk.key == key
         if int(os.environ.get("RAMBA_SHOW_CODE", "0")):
             print(f"=== kernel {gk.key} ===\n{gk.source}\n", flush=True)
         h = ctypes.c_void_p()
@@ -284,7 +289,8 @@ class HipBackend:
         shape = plan.itershape
         V = gk.vec
         nx = shape[nd - 1]
-        gx = max(1, min(2048, (nx + 256 * V - 1) // (256 * V)))
+        cap = int(os.environ.get("RAMBA_GRID_CAP", "2048"))
+        gx = max(1, min(cap, (nx + 256 * V - 1) // (256 * V)))
         gy = max(1, min(8192, shape[nd - 2])) if nd >= 2 else 1
         gz = max(1, min(64, shape[0])) if nd >= 3 else 1
 

@@ -96,7 +96,13 @@ class TestSlicing:
     def test_strided_negative(self, ra_gpu):
         def impl(np_):
             a = np_.arange(65_536)
-            return a[::-1] + a[::2][:16384] * 0  + a[0:32768]
+            return a[::-1] + a * 3
+        run_both(impl, ra_gpu)
+
+    def test_strided_step2(self, ra_gpu):
+        def impl(np_):
+            a = np_.arange(65_536)
+            return a[::2] + a[1::2] * 2
         run_both(impl, ra_gpu)
 
     def test_setitem_views(self, ra_gpu):
