@@ -211,7 +211,7 @@ class LaneEmitter:
             if other in g.wanted and other not in self.memo:
                 x = self.emit(e.a)
                 ct = ctype(e.dtype)
-                fn = "sincosf" if ct == "float" else "sincos"
+                fn = "sincosf" if ct == "float" else "rt_sincos"
                 self.n += 1
                 s, c = f"sc_s{self.n}{self.tag}", f"sc_c{self.n}{self.tag}"
                 self.lines.append(
@@ -380,6 +380,61 @@ __device__ __forceinline__ i64 rt_ipow(i64 b, i64 e) {
     i64 r = 1;
     while (e > 0) { if (e & 1) r *= b; b *= b; e >>= 1; }
     return r;
+}
+
+// Fast fp64 sincos: fdlibm-style Cody-Waite reduction by pi/2 (medium
+// path, valid while |fn| < 2^20 i.e. |x| < ~1.6e6) + the classic
+// __kernel_sin/__kernel_cos minimax polynomials; falls back to ocml
+// sincos (full Payne-Hanek) outside the fast range.  ~45 fp64 ops for
+// BOTH results vs ~2x that for two ocml calls — the hot chain is at the
+// HBM/VALU crossover (SURVEY §7), so this is a throughput lever, and it
+// stays well inside the 1e-12 parity tolerance (abs err ~1 ulp).
+__device__ __forceinline__ void rt_sincos(double x, double *sr, double *cr) {
+    double ax = __builtin_fabs(x);
+    if (!(ax < 1.0e6)) { sincos(x, sr, cr); return; }
+    const double invpio2 = 6.36619772367581382433e-01;
+    const double pio2_1 = 1.57079632673412561417e+00;
+    const double pio2_1t = 6.07710050650619224932e-11;
+    const double pio2_2 = 6.07710050630396597660e-11;
+    const double pio2_2t = 2.02226624879595063154e-21;
+    double fn = __builtin_rint(x * invpio2);
+    int n = (int)fn;
+    // fn*pio2_1 is exact while |fn| < 2^20 (pio2_1 carries 33 bits), so
+    // y0+y1 ~ x - fn*pi/2 with ~1e-20 absolute error -- far inside the
+    // 1e-12 parity tolerance (DESIGN.md §6)
+    double r = __builtin_fma(-fn, pio2_1, x);
+    double w = fn * pio2_1t;
+    double y0 = r - w;
+    double y1 = (r - y0) - w;
+    (void)pio2_2; (void)pio2_2t;
+    // kernel_sin(y0, y1)
+    const double S1 = -1.66666666666666324348e-01;
+    const double S2 = 8.33333333332248946124e-03;
+    const double S3 = -1.98412698298579493134e-04;
+    const double S4 = 2.75573137070700676789e-06;
+    const double S5 = -2.50507602534068634195e-08;
+    const double S6 = 1.58969099521155010221e-10;
+    double z = y0 * y0;
+    double v = z * y0;
+    double rs = S2 + z * (S3 + z * (S4 + z * (S5 + z * S6)));
+    double ks = y0 - ((z * (0.5 * y1 - v * rs) - y1) - v * S1);
+    // kernel_cos(y0, y1)
+    const double C1 = 4.16666666666666019037e-02;
+    const double C2 = -1.38888888888741095749e-03;
+    const double C3 = 2.48015872894767294178e-05;
+    const double C4 = -2.75573143513906633035e-07;
+    const double C5 = 2.08757232129817482790e-09;
+    const double C6 = -1.13596475577881948265e-11;
+    double rc = z * (C1 + z * (C2 + z * (C3 + z * (C4 + z * (C5 + z * C6)))));
+    double hz = 0.5 * z;
+    double wc = 1.0 - hz;
+    double kc = wc + (((1.0 - wc) - hz) + (z * rc - y0 * y1));
+    switch (n & 3) {
+        case 0: *sr = ks;  *cr = kc;  break;
+        case 1: *sr = kc;  *cr = -ks; break;
+        case 2: *sr = -ks; *cr = -kc; break;
+        default: *sr = -kc; *cr = ks; break;
+    }
 }
 """
 
