@@ -100,23 +100,25 @@ __global__ __launch_bounds__(256) void k_flagship(const double *A, double *B,
                                                   double *C, double *D,
                                                   i64 n) {
     LOOP(d2_t a = *(const d2_t *)&A[vb]; d2_t b, c, d;
-         rt_sincos(a[0], &b[0], &c[0]); rt_sincos(a[1], &b[1], &c[1]);
-         d[0] = b[0] * b[0] + c[0] * c[0]; d[1] = b[1] * b[1] + c[1] * c[1];
+         double s0, c0, s1, c1;
+         rt_sincos(a[0], &s0, &c0); rt_sincos(a[1], &s1, &c1);
+         b[0] = s0; b[1] = s1; c[0] = c0; c[1] = c1;
+         d[0] = s0 * s0 + c0 * c0; d[1] = s1 * s1 + c1 * c1;
          *(d2_t *)&B[vb] = b; *(d2_t *)&C[vb] = c; *(d2_t *)&D[vb] = d;)
 }
 
 __global__ __launch_bounds__(256) void k_flagship_nt(const double *A,
                                                      double *B, double *C,
                                                      double *D, i64 n) {
-    LOOP(d2_t a = *(const d2_t *)&A[vb]; d2_t b, c, d;
-         rt_sincos(a[0], &b[0], &c[0]); rt_sincos(a[1], &b[1], &c[1]);
-         d[0] = b[0] * b[0] + c[0] * c[0]; d[1] = b[1] * b[1] + c[1] * c[1];
-         __builtin_nontemporal_store(b[0], &B[vb]);
-         __builtin_nontemporal_store(b[1], &B[vb + 1]);
-         __builtin_nontemporal_store(c[0], &C[vb]);
-         __builtin_nontemporal_store(c[1], &C[vb + 1]);
-         __builtin_nontemporal_store(d[0], &D[vb]);
-         __builtin_nontemporal_store(d[1], &D[vb + 1]);)
+    LOOP(d2_t a = *(const d2_t *)&A[vb];
+         double s0, c0, s1, c1;
+         rt_sincos(a[0], &s0, &c0); rt_sincos(a[1], &s1, &c1);
+         d2_t b, c, d;
+         b[0] = s0; b[1] = s1; c[0] = c0; c[1] = c1;
+         d[0] = s0 * s0 + c0 * c0; d[1] = s1 * s1 + c1 * c1;
+         __builtin_nontemporal_store(b, (d2_t*)&B[vb]);
+         __builtin_nontemporal_store(c, (d2_t*)&C[vb]);
+         __builtin_nontemporal_store(d, (d2_t*)&D[vb]);)
 }
 
 // V=4 variant (32 B per lane per stream)
