@@ -628,7 +628,7 @@ class KernelGen:
             fields.append(("iter_gs", d))
         for c in self.class_list:
             ct = ctype(c.dtype)
-            L.append(f"  {ct}* {c.name}_p;")
+            L.append(f"  {ct}* __restrict__ {c.name}_p;")
             fields.append(("ptr", c.name))
             L.append(f"  i64 {c.name}_off;")
             fields.append(("off", c.name))

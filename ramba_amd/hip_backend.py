@@ -157,9 +157,10 @@ class HipBackend:
         d = bd.divisions[rt.rank]
         cont = self._cont(bd)
         cs = cont.stride()
+        pads = rt.lo_pads(bd)
         off = 0
         for i in range(len(bd.shape)):
-            off += (int(box[0, i] - d[0, i]) + bd.border) * cs[i]
+            off += (int(box[0, i] - d[0, i]) + pads[i]) * cs[i]
         return off, cs
 
     def copy_container_to_temp(self, bd, rt, part_box, vname, need_box):

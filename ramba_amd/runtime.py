@@ -64,20 +64,42 @@ class Runtime:
         self.world = world
 
     # ------------------------------------------------------------------
+    def lo_pads(self, bd):
+        """Leading pad per axis: the border ring, with the innermost axis
+        padded up to a 128-byte boundary so every core row starts
+        cacheline-aligned (misaligned store streams cost partial-line
+        read-modify-write traffic on HBM — measured with
+        tools/probe_streams)."""
+        nd = len(bd.shape)
+        align = max(1, 128 // bd.dtype.itemsize)
+        pads = [bd.border] * nd
+        pads[nd - 1] = -(-bd.border // align) * align if bd.border else 0
+        return tuple(pads)
+
     def shard_geometry(self, bd, r=None):
-        """(div box, container shape, container strides, border per axis)."""
+        """(div box, container shape, container strides, lo pad per axis).
+
+        Container per axis = lo_pad + core + hi_pad; the innermost axis's
+        total padded extent is rounded up to the 128-B granule so OUTER
+        axis strides stay line-aligned too."""
         r = self.rank if r is None else r
         d = bd.divisions[r]
         nd = len(bd.shape)
-        border = tuple(bd.border for _ in range(nd))
+        pads = self.lo_pads(bd)
         if np.any(d[1] < d[0]):
-            return None, None, None, border
-        cshape = tuple(int(d[1, i] - d[0, i] + 1) + 2 * bd.border
-                       for i in range(nd))
+            return None, None, None, pads
+        align = max(1, 128 // bd.dtype.itemsize)
+        cshape = []
+        for i in range(nd):
+            sz = int(d[1, i] - d[0, i] + 1) + pads[i] + bd.border
+            if i == nd - 1 and nd > 1:
+                sz = -(-sz // align) * align
+            cshape.append(sz)
+        cshape = tuple(cshape)
         cstrides = [1] * nd
         for i in range(nd - 2, -1, -1):
             cstrides[i] = cstrides[i + 1] * cshape[i + 1]
-        return d, cshape, tuple(cstrides), border
+        return d, cshape, tuple(cstrides), pads
 
     def core_box(self, bd, r):
         d = bd.divisions[r]
@@ -306,8 +328,9 @@ class Runtime:
         """Container index slices covering a global base box."""
         d = bd.divisions[self.rank if r is None else r]
         nd = len(bd.shape)
-        return tuple(slice(int(box[0, i] - d[0, i] + bd.border),
-                           int(box[1, i] - d[0, i] + bd.border) + 1)
+        pads = self.lo_pads(bd)
+        return tuple(slice(int(box[0, i] - d[0, i]) + pads[i],
+                           int(box[1, i] - d[0, i]) + pads[i] + 1)
                      for i in range(nd))
 
     def free_shard(self, bd):

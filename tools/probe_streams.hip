@@ -64,19 +64,19 @@ __global__ __launch_bounds__(256) void k_init(double *A, i64 n) {
     for (; i < n; i += s) A[i] = (double)i * 0.001;
 }
 
-__global__ __launch_bounds__(256) void k_copy(const double *A, double *B,
+__global__ __launch_bounds__(256) void k_copy(const double *__restrict__ A, double *__restrict__ B,
                                               i64 n) {
     LOOP(*(d2_t *)&B[vb] = *(const d2_t *)&A[vb];)
 }
 
-__global__ __launch_bounds__(256) void k_store3(double *B, double *C,
-                                                double *D, i64 n) {
+__global__ __launch_bounds__(256) void k_store3(double *__restrict__ B, double *__restrict__ C,
+                                                double *__restrict__ D, i64 n) {
     d2_t one = {1.0, 2.0};
     LOOP(*(d2_t *)&B[vb] = one; *(d2_t *)&C[vb] = one; *(d2_t *)&D[vb] = one;)
 }
 
-__global__ __launch_bounds__(256) void k_1r3w(const double *A, double *B,
-                                              double *C, double *D, i64 n) {
+__global__ __launch_bounds__(256) void k_1r3w(const double *__restrict__ A, double *__restrict__ B,
+                                              double *__restrict__ C, double *__restrict__ D, i64 n) {
     LOOP(d2_t a = *(const d2_t *)&A[vb]; d2_t b, c, d;
          b[0] = a[0] + 1.0; b[1] = a[1] + 1.0;
          c[0] = a[0] * 2.0; c[1] = a[1] * 2.0;
@@ -84,8 +84,8 @@ __global__ __launch_bounds__(256) void k_1r3w(const double *A, double *B,
          *(d2_t *)&B[vb] = b; *(d2_t *)&C[vb] = c; *(d2_t *)&D[vb] = d;)
 }
 
-__global__ __launch_bounds__(256) void k_1r3w_nt(const double *A, double *B,
-                                                 double *C, double *D,
+__global__ __launch_bounds__(256) void k_1r3w_nt(const double *__restrict__ A, double *__restrict__ B,
+                                                 double *__restrict__ C, double *__restrict__ D,
                                                  i64 n) {
     LOOP(d2_t a = *(const d2_t *)&A[vb];
          __builtin_nontemporal_store(a[0] + 1.0, &B[vb]);
@@ -96,8 +96,8 @@ __global__ __launch_bounds__(256) void k_1r3w_nt(const double *A, double *B,
          __builtin_nontemporal_store(a[1] + 3.0, &D[vb + 1]);)
 }
 
-__global__ __launch_bounds__(256) void k_flagship(const double *A, double *B,
-                                                  double *C, double *D,
+__global__ __launch_bounds__(256) void k_flagship(const double *__restrict__ A, double *__restrict__ B,
+                                                  double *__restrict__ C, double *__restrict__ D,
                                                   i64 n) {
     LOOP(d2_t a = *(const d2_t *)&A[vb]; d2_t b, c, d;
          double s0, c0, s1, c1;
@@ -107,9 +107,9 @@ __global__ __launch_bounds__(256) void k_flagship(const double *A, double *B,
          *(d2_t *)&B[vb] = b; *(d2_t *)&C[vb] = c; *(d2_t *)&D[vb] = d;)
 }
 
-__global__ __launch_bounds__(256) void k_flagship_nt(const double *A,
-                                                     double *B, double *C,
-                                                     double *D, i64 n) {
+__global__ __launch_bounds__(256) void k_flagship_nt(const double *__restrict__ A,
+                                                     double *__restrict__ B, double *__restrict__ C,
+                                                     double *__restrict__ D, i64 n) {
     LOOP(d2_t a = *(const d2_t *)&A[vb];
          double s0, c0, s1, c1;
          rt_sincos(a[0], &s0, &c0); rt_sincos(a[1], &s1, &c1);
@@ -122,9 +122,9 @@ __global__ __launch_bounds__(256) void k_flagship_nt(const double *A,
 }
 
 // V=4 variant (32 B per lane per stream)
-__global__ __launch_bounds__(256) void k_flagship_v4(const double *A,
-                                                     double *B, double *C,
-                                                     double *D, i64 n) {
+__global__ __launch_bounds__(256) void k_flagship_v4(const double *__restrict__ A,
+                                                     double *__restrict__ B, double *__restrict__ C,
+                                                     double *__restrict__ D, i64 n) {
     typedef __attribute__((ext_vector_type(4))) double d4_t;
     i64 vb = ((i64)blockIdx.x * 256 + threadIdx.x) * 4;
     const i64 xs = (i64)gridDim.x * 256 * 4;
