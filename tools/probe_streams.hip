@@ -142,6 +142,55 @@ __global__ __launch_bounds__(256) void k_flagship_v4(const double *__restrict__ 
     }
 }
 
+
+// unroll-4: four independent 16B chunks in flight per thread
+__global__ __launch_bounds__(256) void k_copy_u4(const double *__restrict__ A,
+                                                 double *__restrict__ B, i64 n) {
+    i64 i = ((i64)blockIdx.x * 256 + threadIdx.x) * 2;
+    const i64 G = (i64)gridDim.x * 256 * 2;
+    for (; i + 6 * G + 2 <= n; i += 8 * G) {
+        d2_t a0 = *(const d2_t *)&A[i];
+        d2_t a1 = *(const d2_t *)&A[i + 2 * G];
+        d2_t a2 = *(const d2_t *)&A[i + 4 * G];
+        d2_t a3 = *(const d2_t *)&A[i + 6 * G];
+        *(d2_t *)&B[i] = a0;
+        *(d2_t *)&B[i + 2 * G] = a1;
+        *(d2_t *)&B[i + 4 * G] = a2;
+        *(d2_t *)&B[i + 6 * G] = a3;
+    }
+    for (; i + 2 <= n; i += 2 * G) *(d2_t *)&B[i] = *(const d2_t *)&A[i];
+}
+
+__global__ __launch_bounds__(256) void k_1r3w_u2(const double *__restrict__ A,
+                                                 double *__restrict__ B,
+                                                 double *__restrict__ C,
+                                                 double *__restrict__ D, i64 n) {
+    i64 i = ((i64)blockIdx.x * 256 + threadIdx.x) * 2;
+    const i64 G = (i64)gridDim.x * 256 * 2;
+    for (; i + 2 * G + 2 <= n; i += 4 * G) {
+        d2_t a0 = *(const d2_t *)&A[i];
+        d2_t a1 = *(const d2_t *)&A[i + 2 * G];
+        d2_t b0, c0, d0, b1, c1, d1;
+        b0[0] = a0[0] + 1.0; b0[1] = a0[1] + 1.0;
+        c0[0] = a0[0] * 2.0; c0[1] = a0[1] * 2.0;
+        d0[0] = a0[0] + 3.0; d0[1] = a0[1] + 3.0;
+        b1[0] = a1[0] + 1.0; b1[1] = a1[1] + 1.0;
+        c1[0] = a1[0] * 2.0; c1[1] = a1[1] * 2.0;
+        d1[0] = a1[0] + 3.0; d1[1] = a1[1] + 3.0;
+        *(d2_t *)&B[i] = b0; *(d2_t *)&C[i] = c0; *(d2_t *)&D[i] = d0;
+        *(d2_t *)&B[i + 2 * G] = b1; *(d2_t *)&C[i + 2 * G] = c1;
+        *(d2_t *)&D[i + 2 * G] = d1;
+    }
+    for (; i + 2 <= n; i += 2 * G) {
+        d2_t a = *(const d2_t *)&A[i];
+        d2_t b, c, d;
+        b[0] = a[0] + 1.0; b[1] = a[1] + 1.0;
+        c[0] = a[0] * 2.0; c[1] = a[1] * 2.0;
+        d[0] = a[0] + 3.0; d[1] = a[1] + 3.0;
+        *(d2_t *)&B[i] = b; *(d2_t *)&C[i] = c; *(d2_t *)&D[i] = d;
+    }
+}
+
 template <typename F>
 static double timeit(F f, int iters) {
     hipEvent_t e0, e1;
@@ -193,5 +242,9 @@ int main(int argc, char **argv) {
         hipLaunchKernelGGL(k_flagship_nt, dim3(grid), dim3(256), 0, 0, A, B, C, D, n));
     RUN("flagship_v4", n * 32.0,
         hipLaunchKernelGGL(k_flagship_v4, dim3(grid), dim3(256), 0, 0, A, B, C, D, n));
+    RUN("copy_u4", n * 16.0,
+        hipLaunchKernelGGL(k_copy_u4, dim3(grid), dim3(256), 0, 0, A, B, n));
+    RUN("1r3w_u2", n * 32.0,
+        hipLaunchKernelGGL(k_1r3w_u2, dim3(grid), dim3(256), 0, 0, A, B, C, D, n));
     return 0;
 }
