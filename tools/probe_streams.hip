@@ -148,17 +148,17 @@ __global__ __launch_bounds__(256) void k_copy_u4(const double *__restrict__ A,
                                                  double *__restrict__ B, i64 n) {
     i64 i = ((i64)blockIdx.x * 256 + threadIdx.x) * 2;
     const i64 G = (i64)gridDim.x * 256 * 2;
-    for (; i + 6 * G + 2 <= n; i += 8 * G) {
+    for (; i + 3 * G + 2 <= n; i += 4 * G) {
         d2_t a0 = *(const d2_t *)&A[i];
-        d2_t a1 = *(const d2_t *)&A[i + 2 * G];
-        d2_t a2 = *(const d2_t *)&A[i + 4 * G];
-        d2_t a3 = *(const d2_t *)&A[i + 6 * G];
+        d2_t a1 = *(const d2_t *)&A[i + G];
+        d2_t a2 = *(const d2_t *)&A[i + 2 * G];
+        d2_t a3 = *(const d2_t *)&A[i + 3 * G];
         *(d2_t *)&B[i] = a0;
-        *(d2_t *)&B[i + 2 * G] = a1;
-        *(d2_t *)&B[i + 4 * G] = a2;
-        *(d2_t *)&B[i + 6 * G] = a3;
+        *(d2_t *)&B[i + G] = a1;
+        *(d2_t *)&B[i + 2 * G] = a2;
+        *(d2_t *)&B[i + 3 * G] = a3;
     }
-    for (; i + 2 <= n; i += 2 * G) *(d2_t *)&B[i] = *(const d2_t *)&A[i];
+    for (; i + 2 <= n; i += G) *(d2_t *)&B[i] = *(const d2_t *)&A[i];
 }
 
 __global__ __launch_bounds__(256) void k_1r3w_u2(const double *__restrict__ A,
@@ -167,9 +167,9 @@ __global__ __launch_bounds__(256) void k_1r3w_u2(const double *__restrict__ A,
                                                  double *__restrict__ D, i64 n) {
     i64 i = ((i64)blockIdx.x * 256 + threadIdx.x) * 2;
     const i64 G = (i64)gridDim.x * 256 * 2;
-    for (; i + 2 * G + 2 <= n; i += 4 * G) {
+    for (; i + G + 2 <= n; i += 2 * G) {
         d2_t a0 = *(const d2_t *)&A[i];
-        d2_t a1 = *(const d2_t *)&A[i + 2 * G];
+        d2_t a1 = *(const d2_t *)&A[i + G];
         d2_t b0, c0, d0, b1, c1, d1;
         b0[0] = a0[0] + 1.0; b0[1] = a0[1] + 1.0;
         c0[0] = a0[0] * 2.0; c0[1] = a0[1] * 2.0;
@@ -178,10 +178,10 @@ __global__ __launch_bounds__(256) void k_1r3w_u2(const double *__restrict__ A,
         c1[0] = a1[0] * 2.0; c1[1] = a1[1] * 2.0;
         d1[0] = a1[0] + 3.0; d1[1] = a1[1] + 3.0;
         *(d2_t *)&B[i] = b0; *(d2_t *)&C[i] = c0; *(d2_t *)&D[i] = d0;
-        *(d2_t *)&B[i + 2 * G] = b1; *(d2_t *)&C[i + 2 * G] = c1;
-        *(d2_t *)&D[i + 2 * G] = d1;
+        *(d2_t *)&B[i + G] = b1; *(d2_t *)&C[i + G] = c1;
+        *(d2_t *)&D[i + G] = d1;
     }
-    for (; i + 2 <= n; i += 2 * G) {
+    for (; i + 2 <= n; i += G) {
         d2_t a = *(const d2_t *)&A[i];
         d2_t b, c, d;
         b[0] = a[0] + 1.0; b[1] = a[1] + 1.0;
