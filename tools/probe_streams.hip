@@ -191,6 +191,50 @@ __global__ __launch_bounds__(256) void k_1r3w_u2(const double *__restrict__ A,
     }
 }
 
+__global__ __launch_bounds__(1024) void k_copy_b1024(const double *__restrict__ A,
+                                                     double *__restrict__ B, i64 n) {
+    i64 vb = ((i64)blockIdx.x * 1024 + threadIdx.x) * 2;
+    const i64 xs = (i64)gridDim.x * 1024 * 2;
+    for (; vb + 2 <= n; vb += xs) *(d2_t *)&B[vb] = *(const d2_t *)&A[vb];
+}
+
+__global__ __launch_bounds__(256) void k_copy_ntl(const double *__restrict__ A,
+                                                  double *__restrict__ B, i64 n) {
+    LOOP(d2_t a = __builtin_nontemporal_load((const d2_t *)&A[vb]);
+         __builtin_nontemporal_store(a, (d2_t *)&B[vb]);)
+}
+
+typedef __attribute__((ext_vector_type(4))) float f4v;
+__global__ __launch_bounds__(256) void k_1r3w_sc1(const double *__restrict__ A,
+                                                  double *__restrict__ B,
+                                                  double *__restrict__ C,
+                                                  double *__restrict__ D, i64 n) {
+    auto rb = __builtin_amdgcn_make_buffer_rsrc((void*)B, 0, (unsigned)(n * 8), 0x00020000);
+    auto rc = __builtin_amdgcn_make_buffer_rsrc((void*)C, 0, (unsigned)(n * 8), 0x00020000);
+    auto rd = __builtin_amdgcn_make_buffer_rsrc((void*)D, 0, (unsigned)(n * 8), 0x00020000);
+    LOOP(d2_t a = *(const d2_t *)&A[vb];
+         d2_t b, c, d;
+         b[0] = a[0] + 1.0; b[1] = a[1] + 1.0;
+         c[0] = a[0] * 2.0; c[1] = a[1] * 2.0;
+         d[0] = a[0] + 3.0; d[1] = a[1] + 3.0;
+         unsigned off = (unsigned)(vb * 8);
+         __builtin_amdgcn_raw_buffer_store_b128(__builtin_bit_cast(f4v, b), rb, off, 0, 16);
+         __builtin_amdgcn_raw_buffer_store_b128(__builtin_bit_cast(f4v, c), rc, off, 0, 16);
+         __builtin_amdgcn_raw_buffer_store_b128(__builtin_bit_cast(f4v, d), rd, off, 0, 16);)
+}
+
+__global__ __launch_bounds__(256) void k_1r3w_ntl(const double *__restrict__ A,
+                                                  double *__restrict__ B,
+                                                  double *__restrict__ C,
+                                                  double *__restrict__ D, i64 n) {
+    LOOP(d2_t a = __builtin_nontemporal_load((const d2_t *)&A[vb]);
+         d2_t b, c, d;
+         b[0] = a[0] + 1.0; b[1] = a[1] + 1.0;
+         c[0] = a[0] * 2.0; c[1] = a[1] * 2.0;
+         d[0] = a[0] + 3.0; d[1] = a[1] + 3.0;
+         *(d2_t *)&B[vb] = b; *(d2_t *)&C[vb] = c; *(d2_t *)&D[vb] = d;)
+}
+
 template <typename F>
 static double timeit(F f, int iters) {
     hipEvent_t e0, e1;
@@ -246,5 +290,13 @@ int main(int argc, char **argv) {
         hipLaunchKernelGGL(k_copy_u4, dim3(grid), dim3(256), 0, 0, A, B, n));
     RUN("1r3w_u2", n * 32.0,
         hipLaunchKernelGGL(k_1r3w_u2, dim3(grid), dim3(256), 0, 0, A, B, C, D, n));
+    RUN("copy_b1024", n * 16.0,
+        hipLaunchKernelGGL(k_copy_b1024, dim3(grid / 4), dim3(1024), 0, 0, A, B, n));
+    RUN("copy_ntl", n * 16.0,
+        hipLaunchKernelGGL(k_copy_ntl, dim3(grid), dim3(256), 0, 0, A, B, n));
+    RUN("1r3w_sc1", n * 32.0,
+        hipLaunchKernelGGL(k_1r3w_sc1, dim3(grid), dim3(256), 0, 0, A, B, C, D, n));
+    RUN("1r3w_ntl", n * 32.0,
+        hipLaunchKernelGGL(k_1r3w_ntl, dim3(grid), dim3(256), 0, 0, A, B, C, D, n));
     return 0;
 }
