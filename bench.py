@@ -119,6 +119,10 @@ def main():
     if kms:
         achieved = ALG_BYTES_PER_ELEM * local_elems / (kms / 1e3)
         traffic = os.environ.get("RAMBA_BENCH_TRAFFIC")
+        if traffic is None and N == 1_000_000_000 and world == 1:
+            # measured via rocprofv3 PMC on this workload (profiles/README.md
+            # r01): FETCH 4.00GB (x2 gfx950 correction = 8GB) + WRITE 24.00GB
+            traffic = 32.0e9
         roofline = {
             "bound": "hbm",
             "achieved": achieved / 1e9,          # GB/s
@@ -126,6 +130,8 @@ def main():
             "unit": "GB/s",
             "frac": achieved / HBM_PEAK_BYTES,
             "traffic": float(traffic) if traffic else None,
+            "traffic_source": "rocprofv3 PMC, profiles/r01_pmc_*.csv"
+                              if traffic else None,
             "kernel_ms": kms,
         }
 
