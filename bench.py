@@ -28,7 +28,6 @@ ROOT = os.path.dirname(os.path.abspath(__file__))
 sys.path.insert(0, ROOT)
 
 HBM_PEAK_BYTES = 8.0e12          # MI355X HBM3E spec peak
-ALG_BYTES_PER_ELEM = 32          # 8 read (A) + 24 stored (B, C, D)
 
 
 def main():
@@ -40,6 +39,12 @@ def main():
     ap.add_argument("--check", action="store_true",
                     help="verify a small slice against NumPy first")
     ap.add_argument("--no-cpu-baseline", action="store_true")
+    ap.add_argument("--workload", default="flagship",
+                    choices=["flagship", "reduce", "stencil"],
+                    help="flagship = BASELINE configs[1] (the judged line); "
+                         "reduce = configs[2] 1e9 fp64 sum; "
+                         "stencil = configs[3] 4096^2 fp32 5-pt Laplacian")
+    ap.add_argument("--stencil-n", type=int, default=4096)
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -73,16 +78,59 @@ def main():
         if rank == 0:
             print("check ok", file=sys.stderr)
 
-    N = args.elems
-    A = ra.arange(N) / 1000.0
-    ra.sync()
-
-    def step():
-        B = ra.sin(A)
-        C = ra.cos(A)
-        D = B * B + C ** 2
+    # ---- workload definitions (BASELINE.json configs) ---------------------
+    if args.workload == "flagship":
+        N = args.elems
+        alg_bytes_per_elem = 32
+        A = ra.arange(N) / 1000.0
         ra.sync()
-        return B, C, D
+
+        def step():
+            B = ra.sin(A)
+            C = ra.cos(A)
+            D = B * B + C ** 2
+            ra.sync()
+            return B, C, D
+        metric = ("GElem/s, 1e9-elem fused sin²+cos² fp64 "
+                  "(BASELINE configs[1])")
+        wl_desc = ("1e9-elem fp64 arange→sin²+cos², one shard per GPU, "
+                   "32 B/elem algorithmic")
+    elif args.workload == "reduce":
+        N = args.elems
+        alg_bytes_per_elem = 8
+        A0 = ra.arange(N) / 1000.0
+        A = ra.sin(A0) ** 2 + ra.cos(A0) ** 2
+        ra.sync()
+        del A0
+
+        def step():
+            s = A.sum()
+            assert abs(float(s) - N) < 1e-3 * N
+            return s
+        metric = ("GElem/s, 1e9-elem fp64 global sum via RCCL allreduce "
+                  "(BASELINE configs[2])")
+        wl_desc = "1e9-elem fp64 sum(), 8 B/elem read + one 8 B allreduce"
+    else:  # stencil
+        S = args.stencil_n
+        N = S * S
+        alg_bytes_per_elem = 8
+        A = ra.fromfunction(lambda x, y: x + y, (S, S), dtype=np.float32)
+        B = ra.zeros((S, S), dtype=np.float32)
+        ra.sync()
+        bufs = [A, B]
+
+        def step():
+            src, dst = bufs
+            dst[1:-1, 1:-1] = (src[:-2, 1:-1] + src[2:, 1:-1]
+                               + src[1:-1, :-2] + src[1:-1, 2:]
+                               - 4.0 * src[1:-1, 1:-1])
+            bufs.reverse()
+            ra.sync()
+            return dst
+        metric = (f"GElem/s, {S}x{S} fp32 5-pt Laplacian stencil "
+                  "(BASELINE configs[3])")
+        wl_desc = (f"{S}^2 fp32 5-point stencil, 2-D blocks + halo "
+                   "exchange, 8 B/elem algorithmic")
 
     keep = None
     for _ in range(args.warmup):
@@ -113,13 +161,16 @@ def main():
     local_elems = 0
     eb = rt.core_box(A.bdarray, rank)
     if eb is not None:
-        local_elems = int(eb[1, 0] - eb[0, 0] + 1)
+        local_elems = 1
+        for i in range(eb.shape[1]):
+            local_elems *= int(eb[1, i] - eb[0, i] + 1)
     kms = min(backend.kernel_times_ms) if backend.kernel_times_ms else None
     roofline = None
     if kms:
-        achieved = ALG_BYTES_PER_ELEM * local_elems / (kms / 1e3)
+        achieved = alg_bytes_per_elem * local_elems / (kms / 1e3)
         traffic = os.environ.get("RAMBA_BENCH_TRAFFIC")
-        if traffic is None and N == 1_000_000_000 and world == 1:
+        if traffic is None and args.workload == "flagship" \
+                and N == 1_000_000_000 and world == 1:
             # measured via rocprofv3 PMC on this workload (profiles/README.md
             # r01): FETCH 4.00GB (x2 gfx950 correction = 8GB) + WRITE 24.00GB
             traffic = 32.0e9
@@ -137,7 +188,8 @@ def main():
 
     # ---- cpu_baseline leg (rank 0, N=1 only): oracle/fused_cpu.c ----------
     cpu_baseline = None
-    if rank == 0 and world == 1 and not args.no_cpu_baseline:
+    if rank == 0 and world == 1 and not args.no_cpu_baseline \
+            and args.workload == "flagship":
         exe = os.path.join(ROOT, "oracle", "_build", "fused_cpu")
         if os.path.exists(exe):
             sample = min(N, 200_000_000)
@@ -163,8 +215,7 @@ def main():
 
     if rank == 0:
         line = {
-            "metric": "GElem/s, 1e9-elem fused sin²+cos² fp64 "
-                      "(BASELINE configs[1])",
+            "metric": metric,
             "value": value,
             "unit": "GElem/s",
             "n_gpus": world,
@@ -177,8 +228,7 @@ def main():
             "dtype": "f64",
             "data": "synthetic",
             "config": {
-                "workload": "1e9-elem fp64 arange→sin²+cos², one shard "
-                            "per GPU, 32 B/elem algorithmic",
+                "workload": wl_desc,
                 "elems": N,
                 "parallelism": f"shard{world}",
             },
