@@ -140,7 +140,7 @@ class NumpyBackend:
 
     # -- kernel execution ------------------------------------------------------
 
-    def launch(self, plan):
+    def launch(self, plan, recipe=None):
         env = {"__scalars__": plan.scalars}
         for op in plan.operands:
             if op.kind == "temp":

@@ -284,8 +284,13 @@ class HipBackend:
         self.kernels[gk.key] = gk
         return gk
 
-    def launch(self, plan):
-        gk = self._get_kernel(plan)
+    def launch(self, plan, recipe=None):
+        if recipe is not None and recipe.backend_kernel is not None:
+            gk = recipe.backend_kernel
+        else:
+            gk = self._get_kernel(plan)
+            if recipe is not None:
+                recipe.backend_kernel = gk
         nd = gk.nd
         shape = plan.itershape
         V = gk.vec

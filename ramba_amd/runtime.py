@@ -57,6 +57,20 @@ class KernelPlan:
         self.temps = {}             # temp_key -> (shape, dtype) buffers
 
 
+class Recipe:
+    """Cached execution geometry for one fused-group structure."""
+    __slots__ = ("plan", "temp_specs", "comm_msgs", "temp_geom",
+                 "adopted_divs", "backend_kernel")
+
+    def __init__(self):
+        self.plan = None
+        self.temp_specs = []
+        self.comm_msgs = []
+        self.temp_geom = {}
+        self.adopted_divs = None
+        self.backend_kernel = None
+
+
 class Runtime:
     def __init__(self, backend, rank=0, world=1):
         self.backend = backend
@@ -118,22 +132,110 @@ class Runtime:
         return np.array([lo, hi])
 
     # ------------------------------------------------------------------
+    # Execution-recipe cache: iterating workloads (the bench loops, stencil
+    # sweeps) re-issue structurally identical fused groups every step; all
+    # the box math, operand addressing and kernel lookup depend only on the
+    # structure, so they are computed once and re-bound to the new backing
+    # arrays on later steps.  The structural signature pins everything the
+    # geometry depends on (views, divisions CONTENT, liveness, statements).
+    # ------------------------------------------------------------------
+
+    def _group_signature(self, group, live):
+        parts = [group.shape, self.world,
+                 ("part", group.part_view, group.part_divs.tobytes(),
+                  group.flex)]
+        for name, oi in group.arr_vars.items():
+            parts.append((name, name in live, oi.bd.shape, str(oi.bd.dtype),
+                          oi.bd.border, oi.bd.divisions.tobytes(),
+                          oi.bd.is_flex, oi.bd.constructed, oi.view,
+                          oi.written))
+        parts.append(tuple(sorted((n, str(dt))
+                                  for n, (v, dt) in group.scalars.items())))
+        parts.append(tuple((s.acc, s.kind, str(s.dtype))
+                           for s, _ in group.reductions))
+        parts.append(tuple((st.target, st.expr) for st in group.statements))
+        return tuple(parts)
+
     def execute_group(self, group):
         live, dead = deferred.compute_live_vars(group)
+        cache = getattr(self, "_recipe_cache", None)
+        if cache is None:
+            cache = self._recipe_cache = {}
+        try:
+            sig = self._group_signature(group, live)
+        except TypeError:
+            sig = None
+        recipe = cache.get(sig) if sig is not None else None
+        if recipe is not None:
+            self._run_recipe(recipe, group, live)
+            return
+        recipe = self._build_and_run(group, live, dead)
+        if sig is not None and recipe is not None:
+            if len(cache) > 128:
+                cache.clear()
+            cache[sig] = recipe
+
+    def _run_recipe(self, recipe, group, live):
+        """Re-run a cached recipe against the current group's backing
+        arrays (pointers differ; geometry is identical by signature)."""
+        # flex adoption + allocation
+        for name, oi in live.items():
+            if oi.bd.is_flex and oi.bd.shape == group.shape \
+                    and recipe.adopted_divs is not None:
+                oi.bd.divisions = recipe.adopted_divs
+                oi.bd.flex = False
+            if not oi.bd.constructed:
+                self.backend.alloc_container(oi.bd, self)
+                oi.bd.constructed = True
+        # rebind operand backing arrays
+        plan = recipe.plan
+        plan.scalars = dict(group.scalars)
+        for op in plan.operands:
+            op.bd = live[op.name].bd
+        # temps + local copies
+        for (vname, shape, dtype, owner, need, part) in recipe.temp_specs:
+            self.backend.alloc_temp(vname, shape, dtype)
+            bd = live[owner].bd
+            if part is not None and bd.constructed:
+                self.backend.copy_container_to_temp(bd, self, part, vname,
+                                                    need)
+        if recipe.comm_msgs:
+            msgs = [(dst, src, live[owner].bd, bx, tgt)
+                    for (dst, src, owner, bx, tgt) in recipe.comm_msgs]
+            self._do_comms(msgs, recipe.temp_geom)
+        if plan.itershape is not None:
+            partials = self.backend.launch(plan, recipe)
+        else:
+            partials = [np.asarray(ir.reduction_init(spec.kind, spec.dtype),
+                                   dtype=spec.dtype)[()]
+                        for spec in plan.reductions]
+        for (spec, pend), val in zip(group.reductions, partials):
+            pend.partial = np.asarray(val, dtype=spec.dtype)[()]
+        self.backend.free_temps()
+        for bd in group.delete_bds:
+            if bd.constructed:
+                self.backend.free_container(bd)
+                bd.constructed = False
+
+    def _build_and_run(self, group, live, dead):
         eboxes = group.exec_boxes()
+        recipe = Recipe()
 
         # flex arrays adopt the group's partition (ramba.py:8093-8101)
         nd = len(group.shape)
+        adopted = None
         for oi in live.values():
             if oi.bd.is_flex and oi.bd.shape == group.shape:
-                divs = np.zeros((self.world, 2, nd), dtype=np.int64)
-                for r, b in enumerate(eboxes):
-                    if b is None:
-                        divs[r, 1, :] = -1
-                    else:
-                        divs[r] = b
-                oi.bd.divisions = divs
+                if adopted is None:
+                    adopted = np.zeros((self.world, 2, nd), dtype=np.int64)
+                    for r, b in enumerate(eboxes):
+                        if b is None:
+                            adopted[r, 1, :] = -1
+                        else:
+                            adopted[r] = b
+                oi.bd.divisions = adopted
                 oi.bd.flex = False
+        recipe.adopted_divs = adopted
 
         # allocate + mark constructed (creation on first use, ramba.py:3506)
         for oi in live.values():
@@ -149,7 +251,7 @@ class Runtime:
 
         # ---- per-gid needed boxes on every rank (deterministic everywhere)
         # needed[r] : gid -> (container_fill_box or None, {var: temp box})
-        comm_msgs = []   # (dst, src, bd, box, target) target: ("border",) or ("temp", var)
+        comm_msgs = []   # (dst, src, owner_var, box, target)
         my_plans = {}
         for r in range(self.world):
             ib = eboxes[r]
@@ -159,6 +261,7 @@ class Runtime:
                 if not vars_here:
                     continue
                 bd = vars_here[0].bd
+                owner = vars_here[0].name
                 core = self.core_box(bd, r)
                 bbox = self.border_box(bd, r)
                 fit_boxes = []
@@ -184,7 +287,7 @@ class Runtime:
                     lo = np.min([b[0] for b in fit_boxes], axis=0)
                     hi = np.max([b[1] for b in fit_boxes], axis=0)
                     fill = np.array([lo, hi])
-                per_gid[gid] = (bd, fill, temp_vars)
+                per_gid[gid] = (bd, owner, fill, temp_vars)
                 # build transfer list
                 if fill is not None:
                     missing = box_subtract(fill, core) if core is not None \
@@ -195,7 +298,8 @@ class Runtime:
                                 continue
                             part = box_intersect(mbox, self.core_box(bd, s))
                             if part is not None:
-                                comm_msgs.append((r, s, bd, part, ("border", gid)))
+                                comm_msgs.append(
+                                    (r, s, owner, part, ("border", gid)))
                 for vname, need in temp_vars.items():
                     for s in range(self.world):
                         part = box_intersect(need, self.core_box(bd, s))
@@ -203,13 +307,13 @@ class Runtime:
                             continue
                         if s == r:
                             continue  # local part copied below
-                        comm_msgs.append((r, s, bd, part, ("temp", vname)))
+                        comm_msgs.append((r, s, owner, part, ("temp", vname)))
             if r == self.rank:
                 my_plans = per_gid
 
         # ---- allocate temp operand buffers + copy local parts
         temp_geom = {}   # var -> (need box, strides)
-        for gid, (bd, fill, temp_vars) in my_plans.items():
+        for gid, (bd, owner, fill, temp_vars) in my_plans.items():
             for vname, need in temp_vars.items():
                 shape = box_shape(need)
                 self.backend.alloc_temp(vname, shape, bd.dtype)
@@ -219,13 +323,19 @@ class Runtime:
                 temp_geom[vname] = (need, tuple(strides))
                 core = self.core_box(bd, self.rank)
                 part = box_intersect(need, core)
+                recipe.temp_specs.append(
+                    (vname, shape, bd.dtype, owner, need, part))
                 if part is not None and bd.constructed:
                     self.backend.copy_container_to_temp(bd, self, part,
                                                         vname, need)
+        recipe.temp_geom = temp_geom
+        recipe.comm_msgs = comm_msgs
 
         # ---- execute transfers
         if comm_msgs:
-            self._do_comms(comm_msgs, temp_geom)
+            msgs = [(dst, src, live[owner].bd, bx, tgt)
+                    for (dst, src, owner, bx, tgt) in comm_msgs]
+            self._do_comms(msgs, temp_geom)
 
         # ---- operand descriptors for the kernel
         ib = eboxes[self.rank]
@@ -253,11 +363,13 @@ class Runtime:
                         ib[0], cstrides, d[0], border)
                     plan.operands.append(OperandPlan(
                         name, "container", oi.bd, None, off0, s, oi.dtype))
-            partials = self.backend.launch(plan)
+            partials = self.backend.launch(plan, recipe)
         else:
+            plan.itershape = None
             partials = [np.asarray(ir.reduction_init(spec.kind, spec.dtype),
                                    dtype=spec.dtype)[()]
                         for spec in plan.reductions]
+        recipe.plan = plan
 
         for (spec, pend), val in zip(group.reductions, partials):
             pend.partial = np.asarray(val, dtype=spec.dtype)[()]
@@ -269,7 +381,7 @@ class Runtime:
             if bd.constructed:
                 self.backend.free_container(bd)
                 bd.constructed = False
-        add_time("execute_group", 0.0)
+        return recipe
 
     # ------------------------------------------------------------------
     def _do_comms(self, msgs, temp_geom):

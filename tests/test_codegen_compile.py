@@ -31,7 +31,7 @@ class CompileCheckBackend(NumpyBackend):
         self.lib.rt_last_error.restype = ctypes.c_char_p
         self.compiled = {}
 
-    def launch(self, plan):
+    def launch(self, plan, recipe=None):
         gk = codegen.generate(plan)
         if gk.key not in self.compiled:
             rc = self.lib.rt_compile_check(gk.source.encode())
@@ -44,7 +44,7 @@ class CompileCheckBackend(NumpyBackend):
                     "hiprtc rejected finish kernel:\n"
                     + self.lib.rt_last_error().decode())
             self.compiled[gk.key] = gk
-        return super().launch(plan)
+        return super().launch(plan, recipe)
 
 
 @pytest.fixture(scope="module")
