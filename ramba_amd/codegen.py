@@ -55,37 +55,83 @@ def is_float(dt):
 # ---------------------------------------------------------------------------
 
 class OpndClass:
-    __slots__ = ("name", "dtype", "inner", "outer_g", "loaded", "stored")
-    # inner: 'v' vector (unit stride, aligned), 'u' unit stride scalar,
-    #        'z' zero stride, 'g' general stride
+    __slots__ = ("name", "dtype", "inner", "outer_g", "rel")
+    # inner: 'v' aligned vector (unit stride, rel 0), 'u' unit stride with a
+    #        structural misalignment `rel` relative to the anchor (reads use
+    #        two aligned vector loads + constant lane extract), 'z' zero
+    #        stride, 'g' general stride
 
-    def __init__(self, name, dtype, inner, outer_g):
+    def __init__(self, name, dtype, inner, outer_g, rel=0):
         self.name = name
         self.dtype = np.dtype(dtype)
         self.inner = inner
         self.outer_g = outer_g   # tuple of bools: outer axis stride passed?
-        self.loaded = False
-        self.stored = False
+        self.rel = rel
 
 
 def classify_plan(plan, vec):
-    """Structural classification of every live operand."""
+    """Structural classification of every live operand.
+
+    The vectorised grid is ALIGNED to the anchor operand (the first
+    unit-stride written stream): its stores become aligned vector stores;
+    unit-stride reads at a different alignment carry a structural `rel`
+    shift and are served by two aligned vector loads with a constant lane
+    extract (measured 2.4-3x over per-lane scalar loads, which produce a
+    stride-V lane pattern -- tools/probe_stencil).
+
+    Returns (classes, novec, anchor_name): novec means a written stream
+    cannot be aligned (or a temp operand would read out of bounds) and the
+    kernel must be built with vec=1."""
     nd = len(plan.itershape)
-    out = []
+    written = {st.target for st in plan.statements}
+    anchor = None
+    w_off = None
     for op in plan.operands:
-        s = op.strides
-        inner = s[nd - 1] if nd else 0
-        if inner == 1 and (op.offset0 % vec) == 0 and vec > 1:
-            icl = "v"
+        inner = op.strides[nd - 1] if nd else 0
+        if inner == 1 and op.name in written:
+            anchor, w_off = op.name, op.offset0
+            break
+    if anchor is None:
+        for op in plan.operands:
+            inner = op.strides[nd - 1] if nd else 0
+            if inner == 1:
+                anchor, w_off = op.name, op.offset0
+                break
+    if w_off is None:
+        w_off = 0
+    out = []
+    novec = False
+    for op in plan.operands:
+        st = op.strides
+        inner = st[nd - 1] if nd else 0
+        rel = 0
+        if inner == 1 and vec > 1:
+            rel = (op.offset0 - w_off) % vec
+            icl = "v" if rel == 0 else "u"
+            if rel != 0 and (op.name in written or op.kind == "temp"):
+                # cannot mis-store / temp buffers carry no safety pads
+                novec = True
         elif inner == 1:
             icl = "u"
         elif inner == 0:
             icl = "z"
         else:
             icl = "g"
-        outer_g = tuple(s[d] != 0 for d in range(nd - 1))
-        out.append(OpndClass(op.name, op.dtype, icl, outer_g))
-    return out
+        outer_g = tuple(st[d] != 0 for d in range(nd - 1))
+        out.append(OpndClass(op.name, op.dtype, icl, outer_g, rel))
+    return out, novec, anchor
+
+
+def plan_structure(plan):
+    """(vec, classes, anchor, key) — shared by generate() and the backend's
+    kernel-cache fast path."""
+    vec = decide_vec(plan)
+    classes, novec, anchor = classify_plan(plan, vec)
+    if novec and vec > 1:
+        vec = 1
+        classes, _, anchor = classify_plan(plan, 1)
+    key = structural_key(plan, classes, vec)
+    return vec, classes, anchor, key
 
 
 # ---------------------------------------------------------------------------
@@ -455,13 +501,15 @@ class GeneratedKernel:
         self.nred = nred
         self.handle = None
         self.finish_handle = None
+        self.anchor = None
 
 
 def structural_key(plan, classes, vec):
     h = hashlib.sha256()
     h.update(f"nd={len(plan.itershape)};vec={vec};".encode())
     for c in classes:
-        h.update(f"op:{c.name}:{c.dtype}:{c.inner}:{c.outer_g};".encode())
+        h.update(f"op:{c.name}:{c.dtype}:{c.inner}:{c.rel}:"
+                 f"{c.outer_g};".encode())
     for n, (v, dt) in sorted(plan.scalars.items()):
         h.update(f"sc:{n}:{dt};".encode())
     for n, dt in sorted(plan.dead_vars.items()):
@@ -555,6 +603,11 @@ class KernelGen:
             if vec_lane is not None and c.inner == "v":
                 em.lines.append(
                     f"      {ct} {ldv} = vv_{name}[{vec_lane}];")
+            elif vec_lane is not None and c.inner == "u" and c.rel:
+                k = vec_lane + c.rel
+                src = f"ul_{name}[{k}]" if k < self.vec \
+                    else f"uh_{name}[{k - self.vec}]"
+                em.lines.append(f"      {ct} {ldv} = {src};")
             else:
                 em.lines.append(
                     f"      {ct} {ldv} = a.{name}_p["
@@ -647,6 +700,9 @@ class KernelGen:
             else:
                 L.append(f"  i64 {n};")
                 fields.append(("scalar_i", n))
+        if V > 1:
+            L.append("  i64 lead;")
+            fields.append(("lead", None))
         nred = len(self.plan.reductions)
         if nred:
             L.append("  char* partials;")
@@ -675,8 +731,9 @@ class KernelGen:
                      f" i{oy} += gridDim.y) {{")
             ind += 2
         x = nd - 1
-        L.append(f"{' '*ind}i64 vb = ((i64)blockIdx.x * 256 + threadIdx.x) "
-                 f"* {V};")
+        lead = "a.lead" if V > 1 else "0"
+        L.append(f"{' '*ind}i64 vb = {lead} + ((i64)blockIdx.x * 256 "
+                 f"+ threadIdx.x) * {V};")
         L.append(f"{' '*ind}const i64 xs = (i64)gridDim.x * 256 * {V};")
         L.append(f"{' '*ind}for (; vb + {V} <= a.n{x}; vb += xs) {{")
         # vector preloads
@@ -688,6 +745,15 @@ class KernelGen:
                 L.append(f"{' '*body_ind}const {vt} vv_{name} = "
                          f"*(const {vt}*)&a.{name}_p["
                          f"{self.addr_expr(name, 'vb')}];")
+            elif c.inner == "u" and V > 1 and c.rel:
+                # two aligned vectors + constant lane extract (classify doc)
+                vt = self.vec_type(c.dtype)
+                L.append(f"{' '*body_ind}const {vt} ul_{name} = "
+                         f"*(const {vt}*)&a.{name}_p["
+                         f"{self.addr_expr(name, 'vb')} - {c.rel}];")
+                L.append(f"{' '*body_ind}const {vt} uh_{name} = "
+                         f"*(const {vt}*)&a.{name}_p["
+                         f"{self.addr_expr(name, 'vb')} - {c.rel} + {V}];")
         for name in sorted(self.written_ops):
             c = self.classes[name]
             if c.inner == "v":
@@ -705,9 +771,18 @@ class KernelGen:
                 L.append(f"{' '*body_ind}*({vt}*)&a.{name}_p["
                          f"{self.addr_expr(name, 'vb')}] = sv_{name};")
         L.append(f"{' '*ind}}}")
-        # tail
+        # scalar edges: [0, lead) prologue and [tstart, n) tail
         if V > 1:
-            L.append(f"{' '*ind}const i64 tstart = a.n{x} / {V} * {V};")
+            L.append(f"{' '*ind}if (a.lead > 0 && blockIdx.x == 0 && "
+                     f"threadIdx.x == 0) {{")
+            L.append(f"{' '*ind}  i64 pe = a.lead < a.n{x} ? a.lead : "
+                     f"a.n{x};")
+            L.append(f"{' '*ind}  for (i64 te = 0; te < pe; ++te) {{")
+            L.append(self.gen_lane_body("_P", "te", ind + 4))
+            L.append(f"{' '*ind}  }}")
+            L.append(f"{' '*ind}}}")
+            L.append(f"{' '*ind}const i64 tstart = a.n{x} > a.lead ? "
+                     f"a.lead + (a.n{x} - a.lead) / {V} * {V} : a.n{x};")
             L.append(f"{' '*ind}if (vb == tstart && tstart < a.n{x}) {{")
             L.append(f"{' '*ind}  for (i64 te = tstart; te < a.n{x}; ++te) {{")
             L.append(self.gen_lane_body("_T", "te", ind + 4))
@@ -844,14 +919,13 @@ def generate(plan):
     nd = len(plan.itershape)
     if nd < 1 or nd > 3:
         raise NotImplementedError(f"{nd}-d iteration spaces")
-    vec = decide_vec(plan)
-    classes = classify_plan(plan, vec)
-    # if nothing vectorises, a vec build only costs code size; keep vec
-    key = structural_key(plan, classes, vec)
+    vec, classes, anchor, key = plan_structure(plan)
     gen = KernelGen(plan, classes, vec)
     source, kmain, kfinish, finish_source, fields = gen.generate(key)
-    return GeneratedKernel(key, source, kmain, kfinish, finish_source,
-                           fields, vec, nd, len(plan.reductions))
+    gk = GeneratedKernel(key, source, kmain, kfinish, finish_source,
+                         fields, vec, nd, len(plan.reductions))
+    gk.anchor = anchor
+    return gk
 
 
 def pack_args(gk, plan, ptr_of):
@@ -859,6 +933,9 @@ def pack_args(gk, plan, ptr_of):
     of an operand buffer; special names '__partials__'."""
     out = bytearray()
     opmap = {o.name: o for o in plan.operands}
+    lead = 0
+    if gk.anchor is not None and gk.vec > 1:
+        lead = (-opmap[gk.anchor].offset0) % gk.vec
     for kind, payload in gk.fields:
         if kind == "iter_n":
             out += struct.pack("<q", plan.itershape[payload])
@@ -877,6 +954,8 @@ def pack_args(gk, plan, ptr_of):
         elif kind == "scalar_i":
             v, dt = plan.scalars[payload]
             out += struct.pack("<q", int(v))
+        elif kind == "lead":
+            out += struct.pack("<q", lead)
         elif kind == "partials":
             out += struct.pack("<Q", ptr_of("__partials__"))
         elif kind == "npartials":

@@ -259,9 +259,7 @@ class HipBackend:
 
     def _get_kernel(self, plan):
         # fast path: structural key without source generation
-        vec = codegen.decide_vec(plan)
-        classes = codegen.classify_plan(plan, vec)
-        key = codegen.structural_key(plan, classes, vec)
+        vec, classes, anchor, key = codegen.plan_structure(plan)
         cached = self.kernels.get(key)
         if cached is not None:
             return cached

@@ -87,7 +87,11 @@ class Runtime:
         nd = len(bd.shape)
         align = max(1, 128 // bd.dtype.itemsize)
         pads = [bd.border] * nd
-        pads[nd - 1] = -(-bd.border // align) * align if bd.border else 0
+        # the last axis always gets at least one full granule of pad on
+        # both sides: the rel-shifted vector reads of the code generator
+        # (codegen.classify_plan) may touch up to vec-1 elements beyond a
+        # view's range
+        pads[nd - 1] = max(-(-bd.border // align) * align, align)
         return tuple(pads)
 
     def shard_geometry(self, bd, r=None):
@@ -106,8 +110,10 @@ class Runtime:
         cshape = []
         for i in range(nd):
             sz = int(d[1, i] - d[0, i] + 1) + pads[i] + bd.border
-            if i == nd - 1 and nd > 1:
-                sz = -(-sz // align) * align
+            if i == nd - 1:
+                # round up + one extra granule of tail slack (rel-shifted
+                # vector reads; keeps outer-axis strides line-aligned)
+                sz = -(-sz // align) * align + align
             cshape.append(sz)
         cshape = tuple(cshape)
         cstrides = [1] * nd
