@@ -67,6 +67,14 @@ int rt_copy_box(uintptr_t stream, void *dst, const void *src, int nd,
                 const int64_t *src_strides, int64_t dst_off, int64_t src_off,
                 int elemsize);
 
+/* Typed combining box copy: dst = dst OP src (axis-reduction merge).
+ * dtype: 0=f64 1=f32 2=i64 3=i32 4=i16 5=i8 6=u8;
+ * op: 0=add 1=mul 2=min 3=max 4=logical_and 5=logical_or. */
+int rt_combine_box(uintptr_t stream, void *dst, const void *src, int nd,
+                   const int64_t *shape, const int64_t *dst_strides,
+                   const int64_t *src_strides, int64_t dst_off,
+                   int64_t src_off, int dtype, int op);
+
 int rt_stream_sync(uintptr_t stream);
 int rt_device_sync(void);
 

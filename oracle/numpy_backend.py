@@ -161,3 +161,48 @@ class NumpyBackend:
 
     def sync(self):
         pass
+
+    # -- axis reductions -------------------------------------
+
+    def fill_container(self, bd, rt, value):
+        c = self.containers.get(bd.gid)
+        if c is not None:
+            c[...] = value
+
+    def axis_reduce_partial(self, bd, off0, strides, lb, axes, kind,
+                            out_dtype):
+        from ramba_amd.shardview import box_shape as _bs
+        base = self._cont(bd)
+        flat = base.reshape(-1)
+        isz = base.itemsize
+        view = np.lib.stride_tricks.as_strided(
+            flat[off0:] if off0 else flat, shape=_bs(lb),
+            strides=tuple(s * isz for s in strides))
+        comb, _ = ir.REDUCTIONS[kind]
+        uf = ir.BINOPS[comb]
+        part = uf.reduce(view.astype(out_dtype, copy=False), axis=axes,
+                         keepdims=True)
+        self.temps["__axred__"] = np.ascontiguousarray(part,
+                                                       dtype=out_dtype)
+
+    def pack_temp_box(self, vname, rel_box):
+        import torch
+        t = self.temps[vname]
+        sl = tuple(slice(int(rel_box[0, i]), int(rel_box[1, i]) + 1)
+                   for i in range(t.ndim))
+        return torch.from_numpy(np.ascontiguousarray(t[sl]))
+
+    def _combine(self, dst, src, kind):
+        comb, _ = ir.REDUCTIONS[kind]
+        dst[...] = ir.BINOPS[comb](dst, src)
+
+    def combine_box_into_container(self, bd, rt, box, buf, kind):
+        self._combine(self._cont(bd)[rt.container_slice(bd, box)],
+                      buf.numpy(), kind)
+
+    def combine_temp_into_container(self, bd, rt, box, vname, rel_box, kind):
+        t = self.temps[vname]
+        sl = tuple(slice(int(rel_box[0, i]), int(rel_box[1, i]) + 1)
+                   for i in range(t.ndim))
+        self._combine(self._cont(bd)[rt.container_slice(bd, box)], t[sl],
+                      kind)

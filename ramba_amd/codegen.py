@@ -972,3 +972,179 @@ def pack_finish_args(partials_ptr, npartials, out_ptrs):
     for p in out_ptrs:
         out += struct.pack("<Q", p)
     return bytes(out)
+
+
+# ---------------------------------------------------------------------------
+# axis-reduction kernels (sum/prod/min/max/any/all along
+# axes; replaces the reference's axis_reduce loops, ramba/ramba.py:8231-8244)
+# ---------------------------------------------------------------------------
+
+def generate_axis_reduce(nd, axes, in_dtype, out_dtype, kind):
+    """Local phase: reduce a strided view over `axes` into a contiguous
+    partial buffer covering the local out box.
+
+    Two shapes:
+    - the innermost view axis is kept: lanes map to contiguous out elements
+      (coalesced reads), reduced axes become serial inner loops;
+    - the innermost view axis is reduced: one 64-lane wave per out element,
+      lanes split the innermost reduction (coalesced), `__shfl_down` tree
+      combines (the wave idiom of the CDNA4 guide, Appendix B).
+
+    Args struct (packed by pack_axis_reduce_args):
+      i64 oe{j} for each out axis (out extents, in axis order)
+      i64 ke{j} for each reduced axis (extents)
+      T* in; i64 in_off; i64 in_s{d} for every view axis d
+      O* out  (contiguous over out extents, row-major)
+    """
+    axes = tuple(sorted(axes))
+    out_axes = [d for d in range(nd) if d not in axes]
+    it = ctype(in_dtype)
+    ot = ctype(out_dtype)
+    comb, _ = ir.REDUCTIONS[kind]
+    lane_split = (nd - 1) in axes
+
+    def combc(a, b):
+        if comb == "add":
+            return f"({a}) + ({b})"
+        if comb == "mul":
+            return f"({a}) * ({b})"
+        if comb in ("minimum", "maximum"):
+            lt = "<" if comb == "minimum" else ">"
+            if np.dtype(out_dtype).kind == "f":
+                return (f"(__builtin_isnan((double)({a})) || (({a}) {lt} "
+                        f"({b}))) ? ({ot})({a}) : ({ot})({b})")
+            return f"(({a}) {lt} ({b})) ? ({ot})({a}) : ({ot})({b})"
+        if comb == "logical_and":
+            return f"((({a}) != 0) && (({b}) != 0)) ? 1 : 0"
+        return f"((({a}) != 0) || (({b}) != 0)) ? 1 : 0"
+
+    init = _axinit(kind, out_dtype)
+    L = [PREAMBLE]
+    L.append("struct AxArgs {")
+    fields = []
+    for j, d in enumerate(out_axes):
+        L.append(f"  i64 oe{j};")
+        fields.append(("oe", d))
+    for j, d in enumerate(axes):
+        L.append(f"  i64 ke{j};")
+        fields.append(("ke", d))
+    L.append(f"  {it}* __restrict__ in;")
+    fields.append(("in_ptr", None))
+    L.append("  i64 in_off;")
+    fields.append(("in_off", None))
+    for d in range(nd):
+        L.append(f"  i64 in_s{d};")
+        fields.append(("in_s", d))
+    L.append(f"  {ot}* __restrict__ out;")
+    fields.append(("out_ptr", None))
+    L.append("};")
+
+    key = hashlib.sha256(
+        f"axred:{nd}:{axes}:{in_dtype}:{out_dtype}:{kind}".encode()
+    ).hexdigest()[:20]
+    kname = f"ax_{key}"
+    L.append(f'extern "C" __global__ void __launch_bounds__(256) '
+             f"{kname}(AxArgs a) {{")
+    # total out elements and per-thread mapping
+    tot = " * ".join([f"a.oe{j}" for j in range(len(out_axes))]) or "1"
+    L.append(f"  const i64 nout = {tot};")
+    if lane_split:
+        L.append("  const int lane = threadIdx.x & 63;")
+        L.append("  i64 w = ((i64)blockIdx.x * 256 + threadIdx.x) >> 6;")
+        L.append("  const i64 ws = ((i64)gridDim.x * 256) >> 6;")
+        L.append("  for (; w < nout; w += ws) {")
+        L.append("    i64 rem = w;")
+    else:
+        L.append("  i64 o = (i64)blockIdx.x * 256 + threadIdx.x;")
+        L.append("  const i64 os = (i64)gridDim.x * 256;")
+        L.append("  for (; o < nout; o += os) {")
+        L.append("    i64 rem = o;")
+    # decompose out index (row-major over out extents)
+    L.append("    i64 base = a.in_off;")
+    for j in range(len(out_axes) - 1, -1, -1):
+        d = out_axes[j]
+        L.append(f"    {{ i64 ix = rem % a.oe{j}; rem /= a.oe{j}; "
+                 f"base += ix * a.in_s{d}; }}")
+    L.append(f"    {ot} acc = {init};")
+    # reduction loops
+    if lane_split:
+        inner = axes[-1]
+        outer_red = axes[:-1]
+        ind = "    "
+        for j, d in enumerate(outer_red):
+            L.append(f"{ind}for (i64 k{j} = 0; k{j} < a.ke{j}; ++k{j}) {{")
+            ind += "  "
+        ki = len(axes) - 1
+        L.append(f"{ind}for (i64 kk = lane; kk < a.ke{ki}; kk += 64) {{")
+        addr = "base" + "".join(
+            f" + k{j} * a.in_s{d}" for j, d in enumerate(outer_red))
+        if kind in ("all", "any"):
+            L.append(f"{ind}  {ot} v = (a.in[{addr} + kk * a.in_s{inner}]"
+                     f" != 0) ? 1 : 0;")
+        else:
+            L.append(f"{ind}  {ot} v = ({ot})a.in[{addr} + kk * "
+                     f"a.in_s{inner}];")
+        L.append(f"{ind}  acc = {combc('acc', 'v')};")
+        L.append(f"{ind}}}")
+        for j in range(len(outer_red)):
+            ind = ind[:-2]
+            L.append(f"{ind}}}")
+        # wave tree
+        L.append("    for (int off = 32; off > 0; off >>= 1) {")
+        if np.dtype(out_dtype).itemsize < 4:
+            L.append(f"      {ot} other = ({ot})__shfl_down((int)acc, off, 64);")
+        else:
+            L.append(f"      {ot} other = __shfl_down(acc, off, 64);")
+        L.append(f"      acc = {combc('acc', 'other')};")
+        L.append("    }")
+        L.append("    if (lane == 0) a.out[w] = acc;")
+    else:
+        ind = "    "
+        for j, d in enumerate(axes):
+            L.append(f"{ind}for (i64 k{j} = 0; k{j} < a.ke{j}; ++k{j}) {{")
+            ind += "  "
+        addr = "base" + "".join(
+            f" + k{j} * a.in_s{d}" for j, d in enumerate(axes))
+        if kind in ("all", "any"):
+            L.append(f"{ind}{ot} v = (a.in[{addr}] != 0) ? 1 : 0;")
+        else:
+            L.append(f"{ind}{ot} v = ({ot})a.in[{addr}];")
+        L.append(f"{ind}acc = {combc('acc', 'v')};")
+        for j in range(len(axes)):
+            ind = ind[:-2]
+            L.append(f"{ind}}}")
+        L.append("    a.out[o] = acc;")
+    L.append("  }")
+    L.append("}")
+    return key, "\n".join(L), kname, fields, lane_split
+
+
+def _axinit(kind, dtype):
+    v = ir.reduction_init(kind, dtype)
+    ct = ctype(dtype)
+    if isinstance(v, float) and np.isinf(v):
+        return ("-__builtin_inf()" if v < 0 else "__builtin_inf()") \
+            if ct == "double" else \
+            ("-__builtin_inff()" if v < 0 else "__builtin_inff()")
+    if isinstance(v, bool):
+        return "1" if v else "0"
+    return f"({ct})({v!r})"
+
+
+def pack_axis_reduce_args(fields, out_extents_by_axis, red_extents_by_axis,
+                          in_ptr, in_off, in_strides, out_ptr):
+    out = bytearray()
+    for kind, d in fields:
+        if kind == "oe":
+            out += struct.pack("<q", out_extents_by_axis[d])
+        elif kind == "ke":
+            out += struct.pack("<q", red_extents_by_axis[d])
+        elif kind == "in_ptr":
+            out += struct.pack("<Q", in_ptr)
+        elif kind == "in_off":
+            out += struct.pack("<q", in_off)
+        elif kind == "in_s":
+            out += struct.pack("<q", in_strides[d])
+        elif kind == "out_ptr":
+            out += struct.pack("<Q", out_ptr)
+    return bytes(out)

@@ -252,6 +252,122 @@ int box_copy_launch(uintptr_t stream, void *dst, const void *src,
 
 }  // namespace
 
+// ---------------------------------------------------------------------------
+// typed combining box copy: dst = dst OP src  (axis-reduction cross-rank
+// merge — replaces the reference's internal_reduction2 slice-combining,
+// ramba/ramba.py:5818-5849)
+// ---------------------------------------------------------------------------
+
+namespace {
+
+template <typename T, int OP>
+__global__ __launch_bounds__(256) void box_combine_kernel(
+    T *__restrict__ dst, const T *__restrict__ src, BoxArgs a) {
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < a.n; i += stride) {
+        int64_t rem = i, so = 0, dofs = 0;
+        for (int d = a.nd - 1; d >= 0; --d) {
+            int64_t idx = rem % a.shape[d];
+            rem /= a.shape[d];
+            so += idx * a.sstr[d];
+            dofs += idx * a.dstr[d];
+        }
+        T s = src[so], d0 = dst[dofs];
+        if (OP == 0) dst[dofs] = d0 + s;
+        else if (OP == 1) dst[dofs] = d0 * s;
+        else if (OP == 2) dst[dofs] = s < d0 ? s : d0;
+        else if (OP == 3) dst[dofs] = s > d0 ? s : d0;
+        else if (OP == 4) dst[dofs] = (d0 != (T)0 && s != (T)0) ? (T)1 : (T)0;
+        else dst[dofs] = (d0 != (T)0 || s != (T)0) ? (T)1 : (T)0;
+    }
+}
+
+template <typename T>
+int box_combine_launch(uintptr_t stream, void *dst, const void *src,
+                       const BoxArgs &a, int op) {
+    int64_t blocks = (a.n + 255) / 256;
+    if (blocks > 2048) blocks = 2048;
+    if (blocks < 1) blocks = 1;
+    hipStream_t st = reinterpret_cast<hipStream_t>(stream);
+    T *d = static_cast<T *>(dst);
+    const T *s = static_cast<const T *>(src);
+    switch (op) {
+#define CASE(n)                                                            \
+    case n:                                                                \
+        hipLaunchKernelGGL((box_combine_kernel<T, n>), dim3((unsigned)blocks), \
+                           dim3(256), 0, st, d, s, a);                     \
+        break;
+        CASE(0) CASE(1) CASE(2) CASE(3) CASE(4) CASE(5)
+#undef CASE
+        default:
+            set_error("rt_combine_box: bad op");
+            return 1;
+    }
+    hipError_t e = hipGetLastError();
+    if (e != hipSuccess) {
+        set_error(std::string("box_combine launch: ") + hipGetErrorString(e));
+        return 1;
+    }
+    return 0;
+}
+
+}  // namespace
+
+// dtype: 0=f64 1=f32 2=i64 3=i32 4=i16 5=i8 6=u8; op: 0=add 1=mul 2=min
+// 3=max 4=logical_and 5=logical_or
+extern "C" int rt_combine_box(uintptr_t stream, void *dst, const void *src,
+                              int nd, const int64_t *shape,
+                              const int64_t *dst_strides,
+                              const int64_t *src_strides, int64_t dst_off,
+                              int64_t src_off, int dtype, int op) {
+    if (nd < 1 || nd > 4) {
+        set_error("rt_combine_box: nd out of range");
+        return 1;
+    }
+    BoxArgs a;
+    a.nd = nd;
+    a.n = 1;
+    for (int d = 0; d < nd; ++d) {
+        a.shape[d] = shape[d];
+        a.dstr[d] = dst_strides[d];
+        a.sstr[d] = src_strides[d];
+        a.n *= shape[d];
+    }
+    for (int d = nd; d < 4; ++d) {
+        a.shape[d] = 1;
+        a.dstr[d] = 0;
+        a.sstr[d] = 0;
+    }
+    if (a.n == 0) return 0;
+    switch (dtype) {
+        case 0: return box_combine_launch<double>(
+            stream, static_cast<char *>(dst) + dst_off * 8,
+            static_cast<const char *>(src) + src_off * 8, a, op);
+        case 1: return box_combine_launch<float>(
+            stream, static_cast<char *>(dst) + dst_off * 4,
+            static_cast<const char *>(src) + src_off * 4, a, op);
+        case 2: return box_combine_launch<int64_t>(
+            stream, static_cast<char *>(dst) + dst_off * 8,
+            static_cast<const char *>(src) + src_off * 8, a, op);
+        case 3: return box_combine_launch<int32_t>(
+            stream, static_cast<char *>(dst) + dst_off * 4,
+            static_cast<const char *>(src) + src_off * 4, a, op);
+        case 4: return box_combine_launch<int16_t>(
+            stream, static_cast<char *>(dst) + dst_off * 2,
+            static_cast<const char *>(src) + src_off * 2, a, op);
+        case 5: return box_combine_launch<int8_t>(
+            stream, static_cast<char *>(dst) + dst_off,
+            static_cast<const char *>(src) + src_off, a, op);
+        case 6: return box_combine_launch<uint8_t>(
+            stream, static_cast<char *>(dst) + dst_off,
+            static_cast<const char *>(src) + src_off, a, op);
+        default:
+            set_error("rt_combine_box: bad dtype");
+            return 1;
+    }
+}
+
 extern "C" int rt_copy_box(uintptr_t stream, void *dst, const void *src,
                            int nd, const int64_t *shape,
                            const int64_t *dst_strides,

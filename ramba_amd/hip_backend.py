@@ -44,6 +44,13 @@ def _load_lib():
                                 ctypes.POINTER(ctypes.c_int64),
                                 ctypes.POINTER(ctypes.c_int64),
                                 ctypes.c_int64, ctypes.c_int64, ctypes.c_int]
+    lib.rt_combine_box.argtypes = [ctypes.c_size_t, ctypes.c_void_p,
+                                   ctypes.c_void_p, ctypes.c_int,
+                                   ctypes.POINTER(ctypes.c_int64),
+                                   ctypes.POINTER(ctypes.c_int64),
+                                   ctypes.POINTER(ctypes.c_int64),
+                                   ctypes.c_int64, ctypes.c_int64,
+                                   ctypes.c_int, ctypes.c_int]
     lib.rt_stream_sync.argtypes = [ctypes.c_size_t]
     return lib
 
@@ -347,3 +354,106 @@ class HipBackend:
 
     def sync(self):
         self.torch.cuda.synchronize()
+
+
+# ---------------------------------------------------------------------------
+# axis reductions: local strided-reduce kernel + typed
+# combining box merges (rt_combine_box)
+# ---------------------------------------------------------------------------
+
+_DT_ENUM = {"float64": 0, "float32": 1, "int64": 2, "int32": 3, "int16": 4,
+            "int8": 5, "uint8": 6, "bool": 6}
+_OP_ENUM = {"sum": 0, "prod": 1, "min": 2, "max": 3, "all": 4, "any": 5}
+
+
+def _hb_fill_container(self, bd, rt, value):
+    c = self.containers.get(bd.gid)
+    if c is not None:
+        c.fill_(float(value) if c.is_floating_point() else int(value))
+
+
+def _hb_axis_reduce_partial(self, bd, off0, strides, lb, axes, kind,
+                            out_dtype):
+    from .shardview import box_shape as _bs
+    from . import codegen as cg
+    nd = lb.shape[1]
+    ck = ("axred", nd, tuple(sorted(axes)), str(bd.dtype), str(out_dtype),
+          kind)
+    cached = self.kernels.get(ck)
+    if cached is None:
+        key, source, kname, fields, lane_split = cg.generate_axis_reduce(
+            nd, axes, bd.dtype, out_dtype, kind)
+        h = ctypes.c_void_p()
+        self._check(self.lib.rt_kernel_get(
+            key.encode(), source.encode(), kname.encode(), ctypes.byref(h)),
+            "rt_kernel_get(axred)")
+        cached = (h.value, fields, lane_split)
+        self.kernels[ck] = cached
+    handle, fields, lane_split = cached
+    shape = _bs(lb)
+    kd_shape = tuple(1 if d in axes else shape[d] for d in range(nd))
+    out_t = self.torch.empty(kd_shape, dtype=self._tdt(out_dtype),
+                             device="cuda")
+    self.temps["__axred__"] = out_t
+    out_extents = {d: shape[d] for d in range(nd) if d not in axes}
+    red_extents = {d: shape[d] for d in axes}
+    nout = 1
+    for v in out_extents.values():
+        nout *= v
+    args = cg.pack_axis_reduce_args(
+        fields, out_extents, red_extents,
+        self._cont(bd).data_ptr(), off0, strides, out_t.data_ptr())
+    if lane_split:
+        gx = max(1, min(4096, (nout * 64 + 255) // 256))
+    else:
+        gx = max(1, min(4096, (nout + 255) // 256))
+    self._check(self.lib.rt_launch(
+        ctypes.c_void_p(handle), gx, 1, 1, 256, self._stream(), args,
+        len(args)), "rt_launch(axred)")
+
+
+def _hb_pack_temp_box(self, vname, rel_box):
+    from .shardview import box_shape as _bs
+    t = self.temps[vname]
+    shape = _bs(rel_box)
+    msg = self.torch.empty(shape, dtype=t.dtype, device="cuda")
+    ts = t.stride()
+    src_off = sum(int(rel_box[0, i]) * ts[i] for i in range(len(shape)))
+    self._copy(msg, msg.stride(), 0, t, ts, src_off, shape,
+               t.element_size())
+    return msg
+
+
+def _hb_combine(self, dst_t, dst_strides, dst_off, src_t, shape, np_dtype,
+                kind):
+    self._check(self.lib.rt_combine_box(
+        self._stream(), ctypes.c_void_p(dst_t.data_ptr()),
+        ctypes.c_void_p(src_t.data_ptr()), len(shape), _i64arr(shape),
+        _i64arr(dst_strides), _i64arr(src_t.stride()), int(dst_off), 0,
+        _DT_ENUM[str(np_dtype)], _OP_ENUM[kind]), "rt_combine_box")
+
+
+def _hb_combine_box_into_container(self, bd, rt, box, buf, kind):
+    from .shardview import box_shape as _bs
+    dst_off, cs = self._box_off(bd, rt, box)
+    self._combine(self._cont(bd), cs, dst_off, buf, _bs(box), bd.dtype,
+                  kind)
+
+
+def _hb_combine_temp_into_container(self, bd, rt, box, vname, rel_box, kind):
+    from .shardview import box_shape as _bs
+    t = self.temps[vname]
+    shape = _bs(rel_box)
+    # pack the temp slice contiguous first (combine kernel takes any
+    # strides, but reuse the packed path for simplicity)
+    sl = self.pack_temp_box(vname, rel_box)
+    dst_off, cs = self._box_off(bd, rt, box)
+    self._combine(self._cont(bd), cs, dst_off, sl, shape, bd.dtype, kind)
+
+
+HipBackend.fill_container = _hb_fill_container
+HipBackend.axis_reduce_partial = _hb_axis_reduce_partial
+HipBackend.pack_temp_box = _hb_pack_temp_box
+HipBackend._combine = _hb_combine
+HipBackend.combine_box_into_container = _hb_combine_box_into_container
+HipBackend.combine_temp_into_container = _hb_combine_temp_into_container

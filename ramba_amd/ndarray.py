@@ -265,48 +265,76 @@ class ndarray:
     # -- reductions (axis=None; reference array_unaryop reduction path,
     #    ramba.py:5890-5920 + internal_reduction2b 5852-5863) ---------------
 
-    def _reduce(self, kind, dtype=None):
+    def _reduce(self, kind, dtype=None, axis=None, keepdims=False):
         dt = self.dtype if dtype is None else np.dtype(dtype)
+        if dtype is None and kind in ("sum", "prod") \
+                and dt.kind in "bi" and dt.itemsize < 8:
+            # NumPy promotes small-int/bool sums to the platform int
+            dt = np.dtype(np.int64)
         if kind in ("all", "any"):
             dt = ir.BOOL
-        src = self
-        if kind in ("all", "any") and self.dtype != ir.BOOL:
-            src = ir.Un("logical_not", ir.Un("logical_not", self, ir.BOOL),
-                        ir.BOOL)
-        pend = deferred.add_reduction(src, kind, dt)
+        # axis normalisation (reference array_unaryop, ramba.py:5896-5903:
+        # an axis tuple covering every dim collapses to the axis-None path)
+        if axis is not None:
+            axes = (axis,) if isinstance(axis, (int, np.integer)) else \
+                tuple(axis)
+            axes = tuple(sorted(a + self.ndim if a < 0 else a for a in axes))
+            if len(axes) == self.ndim:
+                axis = None
+        if axis is None:
+            src = self
+            if kind in ("all", "any") and self.dtype != ir.BOOL:
+                src = ir.Un("logical_not",
+                            ir.Un("logical_not", self, ir.BOOL), ir.BOOL)
+            pend = deferred.add_reduction(src, kind, dt)
+            deferred.flush()
+            rt = deferred.get_runtime()
+            val = rt.finish_reduction(pend)
+            if keepdims:
+                return np.asarray(val).reshape((1,) * self.ndim)
+            return val
+        # axis reduction
         deferred.flush()
         rt = deferred.get_runtime()
-        return rt.finish_reduction(pend)
+        kd_shape = tuple(1 if i in axes else self.shape[i]
+                         for i in range(self.ndim))
+        out_bd = rt.reduce_axes_op(self, axes, kind, dt, kd_shape)
+        res = ndarray(out_bd, View.identity(kd_shape))
+        if not keepdims:
+            res = res[tuple(0 if i in axes else slice(None)
+                            for i in range(self.ndim))]
+        return res
 
-    def sum(self, axis=None, dtype=None, **kw):
-        assert axis is None, "axis reductions are queued (SURVEY §8f n1)"
-        return self._reduce("sum", dtype)
+    def sum(self, axis=None, dtype=None, keepdims=False, **kw):
+        return self._reduce("sum", dtype, axis=axis, keepdims=keepdims)
 
-    def prod(self, axis=None, dtype=None, **kw):
-        assert axis is None
-        return self._reduce("prod", dtype)
+    def prod(self, axis=None, dtype=None, keepdims=False, **kw):
+        return self._reduce("prod", dtype, axis=axis, keepdims=keepdims)
 
-    def min(self, axis=None, **kw):
-        assert axis is None
-        return self._reduce("min")
+    def min(self, axis=None, keepdims=False, **kw):
+        return self._reduce("min", axis=axis, keepdims=keepdims)
 
-    def max(self, axis=None, **kw):
-        assert axis is None
-        return self._reduce("max")
+    def max(self, axis=None, keepdims=False, **kw):
+        return self._reduce("max", axis=axis, keepdims=keepdims)
 
-    def all(self, axis=None, **kw):
-        assert axis is None
-        return self._reduce("all")
+    def all(self, axis=None, keepdims=False, **kw):
+        return self._reduce("all", axis=axis, keepdims=keepdims)
 
-    def any(self, axis=None, **kw):
-        assert axis is None
-        return self._reduce("any")
+    def any(self, axis=None, keepdims=False, **kw):
+        return self._reduce("any", axis=axis, keepdims=keepdims)
 
-    def mean(self, axis=None, **kw):
-        assert axis is None
+    def mean(self, axis=None, keepdims=False, **kw):
         dt = np.dtype(np.float32) if self.dtype == np.float32 \
             else np.dtype(np.float64)
-        return self._reduce("sum", dt) / self.size
+        s = self._reduce("sum", dt, axis=axis, keepdims=keepdims)
+        if axis is None:
+            return s / self.size
+        axes = (axis,) if isinstance(axis, (int, np.integer)) else \
+            tuple(axis)
+        cnt = 1
+        for a in axes:
+            cnt *= self.shape[a + self.ndim if a < 0 else a]
+        return s * (1.0 / cnt)
 
 
 # -- attach the op tables (reference make_method loops, ramba.py:7893-7973) --

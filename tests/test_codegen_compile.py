@@ -120,3 +120,29 @@ def test_misc_unops_compile(rac):
         return (np_.sqrt(a) + np_.exp(-a) + np_.log(a) + np_.tanh(a)
                 + np_.arctan(a))
     run_both(impl, rac, tol=1e-13)
+
+
+def test_axis_reduce_kernels_compile():
+    """Every axis-reduce kernel shape compiles for gfx950 (hiprtc)."""
+    import ctypes
+    import itertools
+    from ramba_amd.codegen import generate_axis_reduce
+    lib = ctypes.CDLL(LIBPATH)
+    lib.rt_compile_check.argtypes = [ctypes.c_char_p]
+    lib.rt_last_error.restype = ctypes.c_char_p
+    cases = [
+        (2, (0,), np.int64, np.int64, "sum"),
+        (2, (1,), np.float64, np.float64, "sum"),     # lane-split
+        (2, (0,), np.float32, np.float32, "max"),
+        (2, (1,), np.int32, np.int32, "min"),
+        (3, (1,), np.float64, np.float64, "sum"),
+        (3, (0, 2), np.int64, np.int64, "prod"),      # mixed + lane-split
+        (2, (1,), np.int64, np.bool_, "any"),
+        (2, (0,), np.float64, np.bool_, "all"),
+    ]
+    for nd, axes, idt, odt, kind in cases:
+        key, src, kname, fields, ls = generate_axis_reduce(
+            nd, axes, np.dtype(idt), np.dtype(odt), kind)
+        rc = lib.rt_compile_check(src.encode())
+        assert rc == 0, (f"{nd}d axes={axes} {kind}: "
+                         + lib.rt_last_error().decode() + "\n" + src)

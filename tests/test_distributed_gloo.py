@@ -161,3 +161,18 @@ def test_broadcast_2d_spmd():
         b = np_.arange(50) * 1.0
         return a + b
     """, world=2)
+
+
+@pytest.mark.parametrize("world", [2, 4])
+def test_axis_reduction_spmd(world):
+    run_spmd("""
+        a = np_.fromfunction(lambda x, y: x * 101 + y, (101, 77),
+                             dtype=np.int64)
+        r0 = a.sum(axis=0)
+        r1 = a.sum(axis=1)
+        m = a[3:90, 5:70].max(axis=0)
+        if np_ is np:
+            return np.concatenate([r0, r1, m])
+        import numpy as _np
+        return _np.concatenate([r0.asarray(), r1.asarray(), m.asarray()])
+    """, world=world)

@@ -221,3 +221,49 @@ class TestNativeLoaded:
             else:
                 np.testing.assert_allclose(got, dat["out"], rtol=tol,
                                            atol=tol, err_msg=name)
+
+
+class TestAxisReduction:
+    def test_sum_axis0_int(self, ra_gpu):
+        def impl(np_):
+            a = np_.fromfunction(lambda x, y: x * 513 + y, (512, 513),
+                                 dtype=np.int64)
+            return a.sum(axis=0)
+        run_both(impl, ra_gpu)
+
+    def test_sum_axis1_lane_split(self, ra_gpu):
+        def impl(np_):
+            a = np_.fromfunction(lambda x, y: x * 0.25 + y * 0.125,
+                                 (300, 4097), dtype=np.float64)
+            return a.sum(axis=1)
+        run_both(impl, ra_gpu, tol=1e-12)
+
+    def test_min_max_axis(self, ra_gpu):
+        def impl(np_):
+            a = np_.fromfunction(lambda x, y: (x * 37 + y * 11) % 1013,
+                                 (257, 129), dtype=np.int64)
+            return a.min(axis=0) + a.max(axis=1).sum()
+        run_both(impl, ra_gpu)
+
+    def test_keepdims_broadcast(self, ra_gpu):
+        def impl(np_):
+            a = np_.fromfunction(lambda x, y: x + y * 0.5, (64, 100),
+                                 dtype=np.float64)
+            return a - a.mean(axis=1, keepdims=True)
+        run_both(impl, ra_gpu, tol=1e-12)
+
+    def test_3d_axis(self, ra_gpu):
+        def impl(np_):
+            a = np_.fromfunction(lambda x, y, z: x * 1000 + y * 31 + z,
+                                 (16, 31, 64), dtype=np.int64)
+            return a.sum(axis=1)
+        run_both(impl, ra_gpu)
+
+    def test_any_all_axis(self, ra_gpu):
+        def impl(np_):
+            a = np_.fromfunction(lambda x, y: x * 13 + y, (90, 110),
+                                 dtype=np.int64)
+            m = (a % 7) == 0
+            return np.array([int(m.any(axis=0).sum()),
+                             int(m.all(axis=1).sum())])
+        run_both(impl, ra_gpu)

@@ -319,3 +319,72 @@ class TestLifetime:
         assert gid in backend.containers
         del a
         assert gid not in backend.containers
+
+
+class TestAxisReduction:
+    """SURVEY §8f n1 (reference axis_reduce, ramba.py:8231-8244 +
+    TestReduction axis cases, test_distributed_array.py:1308)."""
+
+    def test_sum_axis0(self, ra):
+        def impl(np_):
+            a = np_.fromfunction(lambda x, y: x * 31 + y, (16, 31),
+                                 dtype=np.int64)
+            return a.sum(axis=0)
+        run_both(impl, ra)
+
+    def test_sum_axis1(self, ra):
+        def impl(np_):
+            a = np_.fromfunction(lambda x, y: x * 0.5 + y * 0.25, (40, 52),
+                                 dtype=np.float64)
+            return a.sum(axis=1)
+        run_both(impl, ra, tol=1e-13)
+
+    def test_min_max_axis(self, ra):
+        def impl(np_):
+            a = np_.fromfunction(lambda x, y: (x * 37 + y * 11) % 97,
+                                 (23, 45), dtype=np.int64)
+            return a.min(axis=0) + a.max(axis=1).sum()
+        run_both(impl, ra)
+
+    def test_keepdims(self, ra):
+        def impl(np_):
+            a = np_.fromfunction(lambda x, y: x + y, (8, 9),
+                                 dtype=np.float64)
+            return a.sum(axis=1, keepdims=True) + a
+        run_both(impl, ra)
+
+    def test_axis_of_view(self, ra):
+        def impl(np_):
+            a = np_.fromfunction(lambda x, y: x * 100 + y, (30, 30),
+                                 dtype=np.int64)
+            return a[5:25, 3:29].sum(axis=0)
+        run_both(impl, ra)
+
+    def test_negative_axis(self, ra):
+        def impl(np_):
+            a = np_.fromfunction(lambda x, y: x - y, (12, 13),
+                                 dtype=np.int64)
+            return a.sum(axis=-1)
+        run_both(impl, ra)
+
+    def test_mean_axis(self, ra):
+        def impl(np_):
+            a = np_.fromfunction(lambda x, y: x * 2.0 + y, (10, 20),
+                                 dtype=np.float64)
+            return a.mean(axis=0)
+        run_both(impl, ra, tol=1e-13)
+
+    def test_any_all_axis(self, ra):
+        def impl(np_):
+            a = np_.fromfunction(lambda x, y: x * 13 + y, (9, 11),
+                                 dtype=np.int64)
+            m = (a % 7) == 0
+            return np.array([(m.any(axis=0)).sum(), (m.all(axis=1)).sum()])
+        run_both(impl, ra)
+
+    def test_3d_axis(self, ra):
+        def impl(np_):
+            a = np_.fromfunction(lambda x, y, z: x * 100 + y * 10 + z,
+                                 (6, 7, 8), dtype=np.int64)
+            return a.sum(axis=1)
+        run_both(impl, ra)
