@@ -75,6 +75,16 @@ int rt_combine_box(uintptr_t stream, void *dst, const void *src, int nd,
                    const int64_t *src_strides, int64_t dst_off,
                    int64_t src_off, int dtype, int op);
 
+/* cumsum phases (local scan; cross-rank offset fixed up by the host):
+ * phase 1: block sums into `bsums[nblocks]`; phase 2: one-block exclusive
+ * scan of bsums in place + grand total into `total`; phase 3: apply with
+ * `fbase`/`ibase` (the rank's global offset).  dtype: 0=f64 1=f32 2=i64
+ * 3=i32. */
+int rt_cumsum(uintptr_t stream, const void *in, int64_t in_off,
+              int64_t in_stride, int64_t n, void *out, int64_t out_off,
+              void *bsums, int64_t nblocks, void *total, double fbase,
+              int64_t ibase, int dtype, int phase);
+
 int rt_stream_sync(uintptr_t stream);
 int rt_device_sync(void);
 

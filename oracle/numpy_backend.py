@@ -206,3 +206,31 @@ class NumpyBackend:
                    for i in range(t.ndim))
         self._combine(self._cont(bd)[rt.container_slice(bd, box)], t[sl],
                       kind)
+
+    # -- cumsum ---------------------------------------------
+
+    def cumsum_local_phase12(self, bd, off0, stride, n, out_dtype):
+        base = self._cont(bd).reshape(-1)
+        isz = base.itemsize
+        view = np.lib.stride_tricks.as_strided(
+            base[off0:] if off0 else base, shape=(n,),
+            strides=(stride * isz,))
+        cs = np.cumsum(view, dtype=out_dtype)
+        self.temps["__cumsum__"] = cs
+        return cs[-1] if n else np.asarray(0, dtype=out_dtype)[()]
+
+    def cumsum_local_phase3(self, bd, off0, stride, n, out_bd, out_off,
+                            offset, out_dtype):
+        cs = self.temps["__cumsum__"]
+        out = self.containers[out_bd.gid].reshape(-1)
+        out[out_off:out_off + n] = cs + offset
+
+    def allgather_scalars(self, val, dtype):
+        if self.rt.world == 1:
+            return [val]
+        import torch
+        dist = self._d()
+        t = torch.from_numpy(np.asarray([val], dtype=dtype))
+        outs = [torch.empty_like(t) for _ in range(self.rt.world)]
+        dist.all_gather(outs, t)
+        return [o.numpy()[0] for o in outs]

@@ -404,3 +404,179 @@ extern "C" int rt_copy_box(uintptr_t stream, void *dst, const void *src,
             return 1;
     }
 }
+
+// ---------------------------------------------------------------------------
+// cumsum (replaces the reference's scumulative local-prefix +
+// cross-worker fixup chain, ramba/ramba.py:10057-10171 / 3378-3460):
+// three-phase local scan (block sums -> block-sum scan + total -> apply),
+// the cross-rank offset comes from an allgather in the host runtime.
+// ---------------------------------------------------------------------------
+
+namespace {
+
+constexpr int SCAN_THREADS = 256;
+constexpr int SCAN_ITEMS = 16;
+constexpr int SCAN_CHUNK = SCAN_THREADS * SCAN_ITEMS;
+
+template <typename T>
+__global__ __launch_bounds__(256) void cumsum_k1(const T *__restrict__ in,
+                                                 int64_t in_off,
+                                                 int64_t in_stride, int64_t n,
+                                                 T *__restrict__ bsums) {
+    int64_t blk = blockIdx.x;
+    int64_t nblocks = gridDim.x;
+    for (; blk * SCAN_CHUNK < n; blk += nblocks) {
+        int64_t base = blk * SCAN_CHUNK;
+        T acc = (T)0;
+        for (int j = 0; j < SCAN_ITEMS; ++j) {
+            int64_t i = base + j * SCAN_THREADS + threadIdx.x;
+            if (i < n) acc += in[in_off + i * in_stride];
+        }
+        // wave + LDS tree sum
+        for (int off = 32; off > 0; off >>= 1) acc += __shfl_down(acc, off, 64);
+        __shared__ T lds[4];
+        if ((threadIdx.x & 63) == 0) lds[threadIdx.x >> 6] = acc;
+        __syncthreads();
+        if (threadIdx.x == 0)
+            bsums[blk] = lds[0] + lds[1] + lds[2] + lds[3];
+        __syncthreads();
+    }
+}
+
+// one block: in-place EXCLUSIVE scan of bsums; writes the grand total
+template <typename T>
+__global__ __launch_bounds__(256) void cumsum_k2(T *__restrict__ bsums,
+                                                 int64_t nblocks,
+                                                 T *__restrict__ total) {
+    // each thread owns a contiguous range of block sums
+    int64_t per = (nblocks + SCAN_THREADS - 1) / SCAN_THREADS;
+    int64_t lo = threadIdx.x * per;
+    int64_t hi = lo + per < nblocks ? lo + per : nblocks;
+    T s = (T)0;
+    for (int64_t i = lo; i < hi; ++i) s += bsums[i];
+    // exclusive scan of the 256 per-thread sums (wave shfl_up + LDS)
+    T x = s;
+    int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
+    for (int off = 1; off < 64; off <<= 1) {
+        T y = __shfl_up(x, off, 64);
+        if (lane >= off) x += y;
+    }
+    __shared__ T wsum[4];
+    if (lane == 63) wsum[wid] = x;
+    __syncthreads();
+    T wbase = (T)0;
+    for (int w = 0; w < wid; ++w) wbase += wsum[w];
+    T excl = wbase + x - s;   // exclusive prefix of this thread's range
+    // rewrite this thread's range as exclusive prefixes
+    T run = excl;
+    for (int64_t i = lo; i < hi; ++i) {
+        T v = bsums[i];
+        bsums[i] = run;
+        run += v;
+    }
+    // thread 255's final `run` is the grand total (its range is the last
+    // non-empty one, or empty with excl == sum of everything before)
+    if (threadIdx.x == SCAN_THREADS - 1 && total) *total = run;
+}
+
+template <typename T>
+__global__ __launch_bounds__(256) void cumsum_k3(const T *__restrict__ in,
+                                                 int64_t in_off,
+                                                 int64_t in_stride, int64_t n,
+                                                 T *__restrict__ out,
+                                                 int64_t out_off,
+                                                 const T *__restrict__ bsums,
+                                                 T base) {
+    int64_t blk = blockIdx.x;
+    int64_t nblocks = gridDim.x;
+    for (; blk * SCAN_CHUNK < n; blk += nblocks) {
+        int64_t b0 = blk * SCAN_CHUNK;
+        int64_t t0 = b0 + (int64_t)threadIdx.x * SCAN_ITEMS;
+        // per-thread sum of its contiguous 16 elements
+        T s = (T)0;
+        for (int j = 0; j < SCAN_ITEMS; ++j) {
+            int64_t i = t0 + j;
+            if (i < n) s += in[in_off + i * in_stride];
+        }
+        // exclusive scan across the block's threads
+        T x = s;
+        int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
+        for (int off = 1; off < 64; off <<= 1) {
+            T y = __shfl_up(x, off, 64);
+            if (lane >= off) x += y;
+        }
+        __shared__ T wsum[4];
+        __syncthreads();
+        if (lane == 63) wsum[wid] = x;
+        __syncthreads();
+        T wbase = (T)0;
+        for (int w = 0; w < wid; ++w) wbase += wsum[w];
+        T run = base + bsums[blk] + wbase + x - s;
+        for (int j = 0; j < SCAN_ITEMS; ++j) {
+            int64_t i = t0 + j;
+            if (i < n) {
+                run += in[in_off + i * in_stride];
+                out[out_off + i] = run;
+            }
+        }
+        __syncthreads();
+    }
+}
+
+template <typename T>
+int cumsum_launch(uintptr_t stream, const void *in, int64_t in_off,
+                  int64_t in_stride, int64_t n, void *out, int64_t out_off,
+                  void *bsums, int64_t nblocks, void *total, double fbase,
+                  int64_t ibase, int phase) {
+    hipStream_t st = reinterpret_cast<hipStream_t>(stream);
+    unsigned grid = (unsigned)nblocks;
+    if (phase == 1) {
+        hipLaunchKernelGGL((cumsum_k1<T>), dim3(grid), dim3(SCAN_THREADS), 0,
+                           st, static_cast<const T *>(in), in_off, in_stride,
+                           n, static_cast<T *>(bsums));
+    } else if (phase == 2) {
+        hipLaunchKernelGGL((cumsum_k2<T>), dim3(1), dim3(SCAN_THREADS), 0, st,
+                           static_cast<T *>(bsums), nblocks,
+                           static_cast<T *>(total));
+    } else {
+        T base = (T)fbase;
+        if (sizeof(T) >= 4 && (T)0.5 == 0) base = (T)ibase;  // integer T
+        hipLaunchKernelGGL((cumsum_k3<T>), dim3(grid), dim3(SCAN_THREADS), 0,
+                           st, static_cast<const T *>(in), in_off, in_stride,
+                           n, static_cast<T *>(out), out_off,
+                           static_cast<const T *>(bsums), base);
+    }
+    hipError_t e = hipGetLastError();
+    if (e != hipSuccess) {
+        set_error(std::string("cumsum launch: ") + hipGetErrorString(e));
+        return 1;
+    }
+    return 0;
+}
+
+}  // namespace
+
+// dtype: 0=f64 1=f32 2=i64 3=i32; phase 1/2/3 (see kernel comments)
+extern "C" int rt_cumsum(uintptr_t stream, const void *in, int64_t in_off,
+                         int64_t in_stride, int64_t n, void *out,
+                         int64_t out_off, void *bsums, int64_t nblocks,
+                         void *total, double fbase, int64_t ibase,
+                         int dtype, int phase) {
+    switch (dtype) {
+        case 0: return cumsum_launch<double>(stream, in, in_off, in_stride, n,
+                                             out, out_off, bsums, nblocks,
+                                             total, fbase, ibase, phase);
+        case 1: return cumsum_launch<float>(stream, in, in_off, in_stride, n,
+                                            out, out_off, bsums, nblocks,
+                                            total, fbase, ibase, phase);
+        case 2: return cumsum_launch<int64_t>(stream, in, in_off, in_stride,
+                                              n, out, out_off, bsums, nblocks,
+                                              total, fbase, ibase, phase);
+        case 3: return cumsum_launch<int32_t>(stream, in, in_off, in_stride,
+                                              n, out, out_off, bsums, nblocks,
+                                              total, fbase, ibase, phase);
+        default:
+            set_error("rt_cumsum: bad dtype");
+            return 1;
+    }
+}

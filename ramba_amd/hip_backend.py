@@ -51,6 +51,12 @@ def _load_lib():
                                    ctypes.POINTER(ctypes.c_int64),
                                    ctypes.c_int64, ctypes.c_int64,
                                    ctypes.c_int, ctypes.c_int]
+    lib.rt_cumsum.argtypes = [ctypes.c_size_t, ctypes.c_void_p,
+                              ctypes.c_int64, ctypes.c_int64, ctypes.c_int64,
+                              ctypes.c_void_p, ctypes.c_int64,
+                              ctypes.c_void_p, ctypes.c_int64,
+                              ctypes.c_void_p, ctypes.c_double,
+                              ctypes.c_int64, ctypes.c_int, ctypes.c_int]
     lib.rt_stream_sync.argtypes = [ctypes.c_size_t]
     return lib
 
@@ -457,3 +463,73 @@ HipBackend.pack_temp_box = _hb_pack_temp_box
 HipBackend._combine = _hb_combine
 HipBackend.combine_box_into_container = _hb_combine_box_into_container
 HipBackend.combine_temp_into_container = _hb_combine_temp_into_container
+
+
+# -- cumsum --------------------------------------------------
+
+_CS_DT = {"float64": 0, "float32": 1, "int64": 2, "int32": 3}
+_SCAN_CHUNK = 4096
+
+
+def _hb_cumsum_local_phase12(self, bd, off0, stride, n, out_dtype):
+    if str(np.dtype(out_dtype)) not in _CS_DT:
+        raise NotImplementedError(f"cumsum dtype {out_dtype}")
+    dt = _CS_DT[str(np.dtype(out_dtype))]
+    nblocks = max(1, (n + _SCAN_CHUNK - 1) // _SCAN_CHUNK)
+    cont = self._cont(bd)
+    # input must be read as out_dtype: if dtypes differ, cast via a temp
+    src = cont
+    if np.dtype(bd.dtype) != np.dtype(out_dtype):
+        # pack (handles negative strides) then cast
+        tmp = self.torch.empty(n, dtype=cont.dtype, device="cuda")
+        self._copy(tmp, (1,), 0, cont, (stride,), off0, (n,),
+                   cont.element_size())
+        src = tmp.to(self._tdt(out_dtype))
+        off0, stride = 0, 1
+        self.temps["__cs_src__"] = src
+    bsums = self.torch.empty(nblocks, dtype=self._tdt(out_dtype),
+                             device="cuda")
+    total = self.torch.empty(1, dtype=self._tdt(out_dtype), device="cuda")
+    self.temps["__cs_bsums__"] = bsums
+    self.temps["__cs_state__"] = (src, off0, stride, n, nblocks, dt)
+    rc = self.lib.rt_cumsum(
+        self._stream(), ctypes.c_void_p(src.data_ptr()), off0, stride, n,
+        None, 0, ctypes.c_void_p(bsums.data_ptr()),
+        min(nblocks, 4096), None, 0.0, 0, dt, 1)
+    self._check(rc, "rt_cumsum(1)")
+    rc = self.lib.rt_cumsum(
+        self._stream(), None, 0, 0, 0, None, 0,
+        ctypes.c_void_p(bsums.data_ptr()), nblocks,
+        ctypes.c_void_p(total.data_ptr()), 0.0, 0, dt, 2)
+    self._check(rc, "rt_cumsum(2)")
+    return np.asarray(total.cpu().numpy()[0], dtype=out_dtype)[()]
+
+
+def _hb_cumsum_local_phase3(self, bd, off0, stride, n, out_bd, out_off,
+                            offset, out_dtype):
+    src, off0_, stride_, n_, nblocks, dt = self.temps["__cs_state__"]
+    bsums = self.temps["__cs_bsums__"]
+    out = self._cont(out_bd)
+    fb = float(offset) if np.dtype(out_dtype).kind == "f" else 0.0
+    ib = int(offset) if np.dtype(out_dtype).kind != "f" else 0
+    rc = self.lib.rt_cumsum(
+        self._stream(), ctypes.c_void_p(src.data_ptr()), off0_, stride_, n_,
+        ctypes.c_void_p(out.data_ptr()), out_off,
+        ctypes.c_void_p(bsums.data_ptr()), min(nblocks, 4096), None,
+        fb, ib, dt, 3)
+    self._check(rc, "rt_cumsum(3)")
+
+
+def _hb_allgather_scalars(self, val, dtype):
+    if self.rt.world == 1:
+        return [val]
+    dist = self._d()
+    t = self.torch.from_numpy(np.asarray([val], dtype=dtype)).to("cuda")
+    outs = [self.torch.empty_like(t) for _ in range(self.rt.world)]
+    dist.all_gather(outs, t)
+    return [np.asarray(o.cpu().numpy()[0], dtype=dtype)[()] for o in outs]
+
+
+HipBackend.cumsum_local_phase12 = _hb_cumsum_local_phase12
+HipBackend.cumsum_local_phase3 = _hb_cumsum_local_phase3
+HipBackend.allgather_scalars = _hb_allgather_scalars

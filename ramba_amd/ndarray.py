@@ -323,6 +323,19 @@ class ndarray:
     def any(self, axis=None, keepdims=False, **kw):
         return self._reduce("any", axis=axis, keepdims=keepdims)
 
+    def cumsum(self, axis=None, dtype=None, **kw):
+        """1-D cumulative sum (reference scumulative/cumsum,
+        ramba.py:10057-10171, 9675)."""
+        assert self.ndim == 1 and axis in (None, 0), \
+            "cumsum: 1-D only (reference scumulative)"
+        dt = self.dtype if dtype is None else np.dtype(dtype)
+        if dtype is None and dt.kind in "bi" and dt.itemsize < 8:
+            dt = np.dtype(np.int64)   # NumPy platform-int promotion
+        deferred.flush()
+        rt = deferred.get_runtime()
+        out_bd = rt.cumsum_op(self, dt)
+        return ndarray(out_bd, View.identity(self.shape))
+
     def mean(self, axis=None, keepdims=False, **kw):
         dt = np.dtype(np.float32) if self.dtype == np.float32 \
             else np.dtype(np.float64)
@@ -637,6 +650,13 @@ def _module_binop(name):
 
 
 sum = _module_reduction("sum")
+
+
+def cumsum(x, axis=None, dtype=None):
+    if isinstance(x, ndarray):
+        return x.cumsum(axis=axis, dtype=dtype)
+    return np.cumsum(x, axis=axis, dtype=dtype)
+
 prod = _module_reduction("prod")
 amin = _module_reduction("min")
 amax = _module_reduction("max")

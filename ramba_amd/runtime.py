@@ -586,3 +586,58 @@ def reduce_axes_op(self, arr, axes, kind, out_dtype, kd_shape):
 
 
 Runtime.reduce_axes_op = reduce_axes_op
+
+
+# ---------------------------------------------------------------------------
+# cumsum (replaces the reference scumulative local-prefix +
+# sequential cross-worker fixup chain, ramba.py:10057-10171/3378-3460 — the
+# chain becomes an allgather of rank totals)
+# ---------------------------------------------------------------------------
+
+def cumsum_op(self, arr, out_dtype):
+    from .shardview import exec_boxes as _eb, View as _View
+    bd, v = arr.bdarray, arr.view
+    assert v.ndim == 1, "cumsum is 1-D (reference scumulative)"
+    lbs = _eb(v, bd.divisions)
+    # result partitioned by the input's exec boxes (no data exchange)
+    nd = 1
+    divs = np.zeros((self.world, 2, nd), dtype=np.int64)
+    starts = []
+    for r, b in enumerate(lbs):
+        if b is None:
+            divs[r, 1, :] = -1
+            starts.append(None)
+        else:
+            divs[r] = b
+            starts.append(int(b[0, 0]))
+    out_bd = deferred.bdarray(v.shape, out_dtype, divs, default_border,
+                              flex=False)
+    self.backend.alloc_container(out_bd, self)
+    out_bd.constructed = True
+
+    lb = lbs[self.rank]
+    local_total = np.asarray(0, dtype=out_dtype)[()]
+    if lb is not None:
+        d_, cshape, cstrides, pads = self.shard_geometry(bd)
+        off0, strides = v.operand_addressing(lb[0], cstrides, d_[0], pads)
+        n_local = int(lb[1, 0] - lb[0, 0] + 1)
+        local_total = self.backend.cumsum_local_phase12(
+            bd, off0, strides[0], n_local, out_dtype)
+    # cross-rank exclusive prefix of totals, ordered by global start
+    totals = self.backend.allgather_scalars(local_total, out_dtype)
+    offset = np.asarray(0, dtype=out_dtype)[()]
+    my_start = starts[self.rank]
+    if my_start is not None:
+        for r in range(self.world):
+            if starts[r] is not None and starts[r] < my_start:
+                offset = np.asarray(offset + totals[r], dtype=out_dtype)[()]
+    if lb is not None:
+        dd, _, ocs, opads = self.shard_geometry(out_bd)
+        out_off = opads[0]
+        self.backend.cumsum_local_phase3(bd, off0, strides[0], n_local,
+                                         out_bd, out_off, offset, out_dtype)
+    self.backend.free_temps()
+    return out_bd
+
+
+Runtime.cumsum_op = cumsum_op

@@ -176,3 +176,16 @@ def test_axis_reduction_spmd(world):
         import numpy as _np
         return _np.concatenate([r0.asarray(), r1.asarray(), m.asarray()])
     """, world=world)
+
+
+@pytest.mark.parametrize("world", [2, 4])
+def test_cumsum_spmd(world):
+    run_spmd("""
+        a = np_.arange(5001) * 1.0
+        c = a.cumsum()
+        d = np_.arange(4000).cumsum()
+        if np_ is np:
+            return np.concatenate([c, d * 1.0])
+        import numpy as _np
+        return _np.concatenate([c.asarray(), d.asarray() * 1.0])
+    """, world=world, tol=1e-12)

@@ -267,3 +267,31 @@ class TestAxisReduction:
             return np.array([int(m.any(axis=0).sum()),
                              int(m.all(axis=1).sum())])
         run_both(impl, ra_gpu)
+
+
+class TestCumsum:
+    def test_cumsum_large_int(self, ra_gpu):
+        run_both(lambda np_: np_.arange(1_000_000).cumsum(), ra_gpu)
+
+    def test_cumsum_float(self, ra_gpu):
+        run_both(lambda np_: (np_.arange(2_000_000) * 1e-6).cumsum(),
+                 ra_gpu, tol=1e-10)
+
+    def test_cumsum_int32_promotes(self, ra_gpu):
+        def impl(np_):
+            a = np_.arange(100_000).astype(np.int32)
+            return a.cumsum()
+        r, n = run_both(impl, ra_gpu)
+        assert r.dtype == n.dtype == np.int64
+
+    def test_cumsum_of_strided_view(self, ra_gpu):
+        def impl(np_):
+            a = np_.arange(300_000)
+            return a[1000:250_000:7].cumsum()
+        run_both(impl, ra_gpu)
+
+    def test_cumsum_downstream_fusion(self, ra_gpu):
+        def impl(np_):
+            c = np_.arange(50_000).cumsum()
+            return (c % 997) + c[10]
+        run_both(impl, ra_gpu)

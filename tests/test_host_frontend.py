@@ -388,3 +388,27 @@ class TestAxisReduction:
                                  (6, 7, 8), dtype=np.int64)
             return a.sum(axis=1)
         run_both(impl, ra)
+
+
+class TestCumsum:
+    """SURVEY §8f n2 (reference scumulative/cumsum, TestReduction cumsum
+    KAT, test_distributed_array.py:1368)."""
+
+    def test_cumsum_int(self, ra):
+        run_both(lambda np_: np_.arange(1000).cumsum(), ra)
+
+    def test_cumsum_float(self, ra):
+        run_both(lambda np_: (np_.arange(2000) * 0.25).cumsum(), ra,
+                 tol=1e-12)
+
+    def test_cumsum_of_view(self, ra):
+        def impl(np_):
+            a = np_.arange(3000)
+            return a[100:2500:3].cumsum()
+        run_both(impl, ra)
+
+    def test_cumsum_fused_downstream(self, ra):
+        def impl(np_):
+            c = np_.arange(500).cumsum()
+            return c * 2 + 1
+        run_both(impl, ra)
