@@ -184,6 +184,19 @@ class ndarray:
 
     # -- conversion ----------------------------------------------------------
 
+    def __float__(self):
+        assert self.size == 1, "only 1-element arrays convert to scalars"
+        return float(self.asarray().reshape(())[()])
+
+    def __int__(self):
+        assert self.size == 1
+        return int(self.asarray().reshape(())[()])
+
+    def __bool__(self):
+        assert self.size == 1, ("truth value of a multi-element array is "
+                                "ambiguous")
+        return bool(self.asarray().reshape(())[()])
+
     def asarray(self):
         deferred.flush()
         rt = deferred.get_runtime()

@@ -412,3 +412,34 @@ class TestCumsum:
             c = np_.arange(500).cumsum()
             return c * 2 + 1
         run_both(impl, ra)
+
+
+class TestGatherViews:
+    """asarray of raw views — the exit/parity boundary (ramba.py:5735)."""
+
+    def test_gather_slice_view(self, ra):
+        run_both(lambda np_: np_.arange(1000)[100:900:3], ra)
+
+    def test_gather_negative_step_view(self, ra):
+        run_both(lambda np_: np_.arange(500)[::-2], ra)
+
+    def test_gather_transpose_view(self, ra):
+        def impl(np_):
+            return np_.fromfunction(lambda x, y: x * 20 + y, (15, 20),
+                                    dtype=np.int64).T
+        run_both(impl, ra)
+
+    def test_gather_broadcast_view(self, ra):
+        def impl(np_):
+            a = np_.arange(7)
+            if np_ is np:
+                return np.broadcast_to(a, (5, 7)).copy()
+            return np_.broadcast_to(a, (5, 7))
+        run_both(impl, ra)
+
+    def test_scalar_dunders(self, ra):
+        a = ra.arange(10) * 2.5
+        v = a[20:21] if False else a[4:5]
+        assert float(v) == 10.0
+        assert int(v) == 10
+        assert bool(v)
