@@ -115,8 +115,11 @@ class HipBackend:
     def init_process_group(self, rank, world):
         import torch.distributed as dist
         self._dist = dist
+        # RCCL in production; RAMBA_PG_BACKEND=gloo lets tests run several
+        # ranks against ONE GPU (gloo needs host tensors -> staging below)
+        self._pg = os.environ.get("RAMBA_PG_BACKEND", "nccl")
         if not dist.is_initialized():
-            dist.init_process_group("nccl", rank=rank, world_size=world)
+            dist.init_process_group(self._pg, rank=rank, world_size=world)
 
     def _d(self):
         if self._dist is None:
@@ -204,13 +207,24 @@ class HipBackend:
             return
         dist = self._d()
         self.torch.cuda.synchronize()  # pack kernels complete before NCCL
+        staged = getattr(self, "_pg", "nccl") != "nccl"
+        if staged:
+            sends = [(d, b.cpu()) for (d, b) in sends]
+            hrecvs = [(s, b, b.cpu()) for (s, b) in recvs]
         ops = []
         for (dst, buf) in sends:
             ops.append(dist.P2POp(dist.isend, buf, dst))
-        for (src, buf) in recvs:
-            ops.append(dist.P2POp(dist.irecv, buf, src))
+        if staged:
+            for (src, _, hb) in hrecvs:
+                ops.append(dist.P2POp(dist.irecv, hb, src))
+        else:
+            for (src, buf) in recvs:
+                ops.append(dist.P2POp(dist.irecv, buf, src))
         for req in dist.batch_isend_irecv(ops):
             req.wait()
+        if staged:
+            for (_, dbuf, hb) in hrecvs:
+                dbuf.copy_(hb.to(dbuf.device))
 
     def unpack_box_to_container(self, bd, rt, box, buf):
         cont = self._cont(bd)
@@ -261,7 +275,9 @@ class HipBackend:
         dt = v.dtype
         if dt == np.bool_:
             v = v.astype(np.uint8)
-        t = self.torch.from_numpy(v.reshape(1).copy()).to("cuda")
+        t = self.torch.from_numpy(v.reshape(1).copy())
+        if getattr(self, "_pg", "nccl") == "nccl":
+            t = t.to("cuda")
         dist.all_reduce(t, op=getattr(dist.ReduceOp, self._RED_MAP[kind]))
         out = t.cpu().numpy()[0]
         if dt == np.bool_:
@@ -524,7 +540,9 @@ def _hb_allgather_scalars(self, val, dtype):
     if self.rt.world == 1:
         return [val]
     dist = self._d()
-    t = self.torch.from_numpy(np.asarray([val], dtype=dtype)).to("cuda")
+    t = self.torch.from_numpy(np.asarray([val], dtype=dtype))
+    if getattr(self, "_pg", "nccl") == "nccl":
+        t = t.to("cuda")
     outs = [self.torch.empty_like(t) for _ in range(self.rt.world)]
     dist.all_gather(outs, t)
     return [np.asarray(o.cpu().numpy()[0], dtype=dtype)[()] for o in outs]
