@@ -272,6 +272,8 @@ int main(int argc, char **argv) {
     printf("%-14s %8.3f ms   %8.1f GB/s (algorithmic %.1f GB)\n", name,  \
            ms, (bytes) / ms * 1e-6, (bytes) / 1e9);
 
+    RUN("memcpyDtoD", n * 16.0,
+        CHECK(hipMemcpyAsync(B, A, n * 8, hipMemcpyDeviceToDevice, 0)));
     RUN("copy_1r1w", n * 16.0,
         hipLaunchKernelGGL(k_copy, dim3(grid), dim3(256), 0, 0, A, B, n));
     RUN("store3", n * 24.0,
