@@ -164,6 +164,24 @@ class ndarray:
         # slice branch of setitem_array_executor (ramba.py:6273-6296)
         if self.readonly:
             raise ValueError("assignment destination is read-only")
+        # boolean-mask write (reference maskarray guard,
+        # ramba.py:8476-8478): a[mask] = v  ->  a = where(mask, v, a)
+        if isinstance(index, ndarray) and index.dtype == np.bool_:
+            if not index.broadcastable_to(self.shape):
+                raise IndexError("mask shape does not match array shape")
+            mask = index.broadcast_to(self.shape) \
+                if index.shape != self.shape else index
+            if isinstance(value, np.ndarray):
+                value = fromarray(value)
+            if isinstance(value, ndarray) and value.shape != self.shape:
+                value = value.broadcast_to(self.shape)
+            dt = self.dtype
+            deferred.add_op(self, "=",
+                            ir.Where(mask, ir.Cast(value, dt)
+                                     if isinstance(value, ndarray) else value,
+                                     self, dt),
+                            empty_like=empty_like)
+            return
         view = self[index] if not (isinstance(index, tuple) and index == ()) \
             else self
         if isinstance(view, (np.generic, numbers.Number)):

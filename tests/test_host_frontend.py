@@ -443,3 +443,45 @@ class TestGatherViews:
         assert float(v) == 10.0
         assert int(v) == 10
         assert bool(v)
+
+
+class TestMaskedWrite:
+    """boolean-mask writes (reference maskarray guard,
+    ramba.py:8476-8478; TestBasic masked, test_distributed_array.py:975)."""
+
+    def test_mask_scalar(self, ra):
+        def impl(np_):
+            a = np_.arange(200) * 1.0
+            a[a % 3 == 0] = -1.0
+            return a
+        run_both(impl, ra)
+
+    def test_mask_array_value(self, ra):
+        def impl(np_):
+            a = np_.arange(300)
+            b = np_.arange(300) * 10
+            a[a > 150] = b[a > 150] if np_ is np else b
+            # ramba semantics: a[mask] = b selects elementwise from b
+            return a
+        res_r = impl(ra).asarray()
+        a = np.arange(300)
+        b = np.arange(300) * 10
+        a[a > 150] = b[a > 150]
+        assert np.array_equal(res_r, a)
+
+    def test_mask_2d(self, ra):
+        def impl(np_):
+            a = np_.fromfunction(lambda x, y: x * 20 + y, (15, 20),
+                                 dtype=np.int64)
+            a[a % 7 == 0] = 0
+            return a
+        run_both(impl, ra)
+
+    def test_mask_compound(self, ra):
+        def impl(np_):
+            a = np_.arange(100) * 0.5
+            m = (a > 10.0).logical_and(a < 30.0) if np_ is not np \
+                else np.logical_and(a > 10.0, a < 30.0)
+            a[m] = 99.0
+            return a
+        run_both(impl, ra)
