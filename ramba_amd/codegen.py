@@ -55,18 +55,25 @@ def is_float(dt):
 # ---------------------------------------------------------------------------
 
 class OpndClass:
-    __slots__ = ("name", "dtype", "inner", "outer_g", "rel")
+    __slots__ = ("name", "dtype", "inner", "outer_g", "rel", "base", "delta")
     # inner: 'v' aligned vector (unit stride, rel 0), 'u' unit stride with a
     #        structural misalignment `rel` relative to the anchor (reads use
     #        two aligned vector loads + constant lane extract), 'z' zero
     #        stride, 'g' general stride
+    # base/delta: same-array fold — this operand addresses base's buffer at
+    #        a small structural offset delta, so overlapping vector loads
+    #        CSE (a 5-pt stencil's x±1 shifted loads share the center's
+    #        aligned vectors — probe_stencil measured the win)
 
-    def __init__(self, name, dtype, inner, outer_g, rel=0):
+    def __init__(self, name, dtype, inner, outer_g, rel=0, base=None,
+                 delta=0):
         self.name = name
         self.dtype = np.dtype(dtype)
         self.inner = inner
         self.outer_g = outer_g   # tuple of bools: outer axis stride passed?
         self.rel = rel
+        self.base = base
+        self.delta = delta
 
 
 def classify_plan(plan, vec):
@@ -119,6 +126,26 @@ def classify_plan(plan, vec):
             icl = "g"
         outer_g = tuple(st[d] != 0 for d in range(nd - 1))
         out.append(OpndClass(op.name, op.dtype, icl, outer_g, rel))
+    # same-array offset folding: container operands of the same backing
+    # array with identical strides and a small offset difference share the
+    # base operand's pointer/offset/stride args; their own offset becomes
+    # a structural literal delta
+    by_group = {}
+    for c, op in zip(out, plan.operands):
+        if op.kind != "container" or c.inner not in ("v", "u"):
+            continue
+        gkey = (id(op.bd), op.strides)
+        bases = by_group.setdefault(gkey, [])
+        folded = False
+        for (bc, bop) in bases:
+            delta = op.offset0 - bop.offset0
+            if abs(delta) <= 4 * vec:
+                c.base = bc.name
+                c.delta = int(delta)
+                folded = True
+                break
+        if not folded:
+            bases.append((c, op))
     return out, novec, anchor
 
 
@@ -508,8 +535,8 @@ def structural_key(plan, classes, vec):
     h = hashlib.sha256()
     h.update(f"nd={len(plan.itershape)};vec={vec};".encode())
     for c in classes:
-        h.update(f"op:{c.name}:{c.dtype}:{c.inner}:{c.rel}:"
-                 f"{c.outer_g};".encode())
+        h.update(f"op:{c.name}:{c.dtype}:{c.inner}:{c.rel}:{c.base}:"
+                 f"{c.delta}:{c.outer_g};".encode())
     for n, (v, dt) in sorted(plan.scalars.items()):
         h.update(f"sc:{n}:{dt};".encode())
     for n, dt in sorted(plan.dead_vars.items()):
@@ -567,20 +594,32 @@ class KernelGen:
 
     # -- addressing ---------------------------------------------------------
 
-    def addr_expr(self, name, inner_expr):
-        """element index expression into operand `name`'s buffer."""
+    def eff(self, name):
+        """arg-owning operand: the fold base, or the operand itself."""
         c = self.classes[name]
-        parts = [f"a.{name}_off"]
+        return c.base if c.base is not None else name
+
+    def addr_parts(self, name, inner_expr):
+        """(symbolic parts, constant) of the element index into `name`'s
+        buffer — the constant is kept separate so equal addresses CSE."""
+        c = self.classes[name]
+        eff = self.eff(name)
+        parts = [f"a.{eff}_off"]
+        const = int(c.delta)
         for d in range(self.nd - 1):
             if c.outer_g[d]:
-                parts.append(f"i{d} * a.{name}_s{d}")
-        op = next(o for o in self.plan.operands if o.name == name)
-        inner = "v" if c.inner in ("v", "u") else (
-            None if c.inner == "z" else "g")
+                parts.append(f"i{d} * a.{eff}_s{d}")
         if c.inner in ("v", "u"):
             parts.append(f"({inner_expr})")
         elif c.inner == "g":
-            parts.append(f"({inner_expr}) * a.{name}_sx")
+            parts.append(f"({inner_expr}) * a.{eff}_sx")
+        return parts, const
+
+    def addr_expr(self, name, inner_expr, extra_const=0):
+        parts, const = self.addr_parts(name, inner_expr)
+        const += extra_const
+        if const:
+            parts = parts + [f"({const})"]
         return " + ".join(parts)
 
     # -- body generation ------------------------------------------------------
@@ -602,15 +641,17 @@ class KernelGen:
             ldv = self._load_name(em, name)
             if vec_lane is not None and c.inner == "v":
                 em.lines.append(
-                    f"      {ct} {ldv} = vv_{name}[{vec_lane}];")
+                    f"      {ct} {ldv} = "
+                    f"{self.preload[name][1]}[{vec_lane}];")
             elif vec_lane is not None and c.inner == "u" and c.rel:
                 k = vec_lane + c.rel
-                src = f"ul_{name}[{k}]" if k < self.vec \
-                    else f"uh_{name}[{k - self.vec}]"
+                _, ul, uh = self.preload[name]
+                src = f"{ul}[{k}]" if k < self.vec \
+                    else f"{uh}[{k - self.vec}]"
                 em.lines.append(f"      {ct} {ldv} = {src};")
             else:
                 em.lines.append(
-                    f"      {ct} {ldv} = a.{name}_p["
+                    f"      {ct} {ldv} = a.{self.eff(name)}_p["
                     f"{self.addr_expr(name, inner_expr)}];")
         # statements
         last_val = {}
@@ -640,8 +681,8 @@ class KernelGen:
                 em.lines.append(f"      sv_{name}[{vec_lane}] = {final};")
             else:
                 em.lines.append(
-                    f"      a.{name}_p[{self.addr_expr(name, inner_expr)}]"
-                    f" = {final};")
+                    f"      a.{self.eff(name)}_p["
+                    f"{self.addr_expr(name, inner_expr)}] = {final};")
         pad = " " * (indent - 6)
         return "\n".join(pad + ln.lstrip() if False else
                          (" " * indent) + ln.strip() for ln in em.lines)
@@ -680,6 +721,8 @@ class KernelGen:
             L.append(f"  i64 gs{d};")
             fields.append(("iter_gs", d))
         for c in self.class_list:
+            if c.base is not None:
+                continue  # folded: shares the base operand's args
             ct = ctype(c.dtype)
             L.append(f"  {ct}* __restrict__ {c.name}_p;")
             fields.append(("ptr", c.name))
@@ -736,24 +779,36 @@ class KernelGen:
                  f"+ threadIdx.x) * {V};")
         L.append(f"{' '*ind}const i64 xs = (i64)gridDim.x * 256 * {V};")
         L.append(f"{' '*ind}for (; vb + {V} <= a.n{x}; vb += xs) {{")
-        # vector preloads
+        # vector preloads, CSE'd by (pointer, address) — folded same-array
+        # operands share overlapping vectors
         body_ind = ind + 2
+        self.preload = {}
+        cse = {}
+        nload = [0]
+
+        def vec_load(name, extra_const):
+            parts, const = self.addr_parts(name, "vb")
+            const += extra_const
+            kk = (self.eff(name), tuple(parts), const)
+            v = cse.get(kk)
+            if v is None:
+                vt = self.vec_type(self.classes[name].dtype)
+                nload[0] += 1
+                v = f"vv{nload[0]}"
+                addr = " + ".join(parts + ([f"({const})"] if const else []))
+                L.append(f"{' '*body_ind}const {vt} {v} = "
+                         f"*(const {vt}*)&a.{self.eff(name)}_p[{addr}];")
+                cse[kk] = v
+            return v
+
         for name in sorted(self.read_ops):
             c = self.classes[name]
             if c.inner == "v":
-                vt = self.vec_type(c.dtype)
-                L.append(f"{' '*body_ind}const {vt} vv_{name} = "
-                         f"*(const {vt}*)&a.{name}_p["
-                         f"{self.addr_expr(name, 'vb')}];")
+                self.preload[name] = ("v", vec_load(name, 0))
             elif c.inner == "u" and V > 1 and c.rel:
-                # two aligned vectors + constant lane extract (classify doc)
-                vt = self.vec_type(c.dtype)
-                L.append(f"{' '*body_ind}const {vt} ul_{name} = "
-                         f"*(const {vt}*)&a.{name}_p["
-                         f"{self.addr_expr(name, 'vb')} - {c.rel}];")
-                L.append(f"{' '*body_ind}const {vt} uh_{name} = "
-                         f"*(const {vt}*)&a.{name}_p["
-                         f"{self.addr_expr(name, 'vb')} - {c.rel} + {V}];")
+                self.preload[name] = (
+                    "u", vec_load(name, -c.rel),
+                    vec_load(name, -c.rel + V))
         for name in sorted(self.written_ops):
             c = self.classes[name]
             if c.inner == "v":
@@ -768,7 +823,7 @@ class KernelGen:
             c = self.classes[name]
             if c.inner == "v":
                 vt = self.vec_type(c.dtype)
-                L.append(f"{' '*body_ind}*({vt}*)&a.{name}_p["
+                L.append(f"{' '*body_ind}*({vt}*)&a.{self.eff(name)}_p["
                          f"{self.addr_expr(name, 'vb')}] = sv_{name};")
         L.append(f"{' '*ind}}}")
         # scalar edges: [0, lead) prologue and [tstart, n) tail
