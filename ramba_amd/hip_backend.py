@@ -326,6 +326,10 @@ class HipBackend:
         gx = max(1, min(cap, (nx + 256 * V - 1) // (256 * V)))
         gy = max(1, min(8192, shape[nd - 2])) if nd >= 2 else 1
         gz = max(1, min(64, shape[0])) if nd >= 3 else 1
+        if gk.nred:
+            # bound the partials array: the finish kernel is one block
+            gy = max(1, min(gy, 8192 // gx))
+            gz = max(1, min(gz, max(1, 8192 // (gx * gy))))
 
         partials = None
         outs = []
