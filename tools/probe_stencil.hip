@@ -128,6 +128,49 @@ __global__ __launch_bounds__(256) void st_ycarry(
     }
 }
 
+// v4: unrolled y-block, addresses from ONE base pointer with literal row
+// deltas -- tests whether LLVM GVN dedupes the overlapping vector loads
+// across unrolled rows (the codegen-friendly alternative to an explicit
+// register window)
+template <int YB>
+__global__ __launch_bounds__(256) void st_yunroll(
+    const float *__restrict__ A, float *__restrict__ B, int n, i64 pitch) {
+    i64 x4 = ((i64)blockIdx.x * 256 + threadIdx.x) * 4;
+    if (x4 + 4 > n) return;
+    i64 y0 = (i64)blockIdx.y * YB + 1;
+    if (y0 + YB > n - 1) {
+        for (i64 y = y0; y < n - 1; ++y) {
+            f4_t a0 = *(const f4_t *)&A[(y - 1) * pitch + x4];
+            f4_t a1 = *(const f4_t *)&A[y * pitch + x4];
+            f4_t a2 = *(const f4_t *)&A[(y + 1) * pitch + x4];
+            float left = x4 > 0 ? A[y * pitch + x4 - 1] : 0.0f;
+            float right = x4 + 4 < n ? A[y * pitch + x4 + 4] : 0.0f;
+            f4_t out;
+            out[0] = a0[0] + a2[0] + left + a1[1] - 4.0f * a1[0];
+            out[1] = a0[1] + a2[1] + a1[0] + a1[2] - 4.0f * a1[1];
+            out[2] = a0[2] + a2[2] + a1[1] + a1[3] - 4.0f * a1[2];
+            out[3] = a0[3] + a2[3] + a1[2] + right - 4.0f * a1[3];
+            if (x4 >= 1 && x4 + 4 <= n - 1) *(f4_t *)&B[y * pitch + x4] = out;
+        }
+        return;
+    }
+#pragma unroll
+    for (int ky = 0; ky < YB; ++ky) {
+        i64 y = y0 + ky;
+        f4_t a0 = *(const f4_t *)&A[(y - 1) * pitch + x4];
+        f4_t a1 = *(const f4_t *)&A[y * pitch + x4];
+        f4_t a2 = *(const f4_t *)&A[(y + 1) * pitch + x4];
+        float left = x4 > 0 ? A[y * pitch + x4 - 1] : 0.0f;
+        float right = x4 + 4 < n ? A[y * pitch + x4 + 4] : 0.0f;
+        f4_t out;
+        out[0] = a0[0] + a2[0] + left + a1[1] - 4.0f * a1[0];
+        out[1] = a0[1] + a2[1] + a1[0] + a1[2] - 4.0f * a1[1];
+        out[2] = a0[2] + a2[2] + a1[1] + a1[3] - 4.0f * a1[2];
+        out[3] = a0[3] + a2[3] + a1[2] + right - 4.0f * a1[3];
+        if (x4 >= 1 && x4 + 4 <= n - 1) *(f4_t *)&B[y * pitch + x4] = out;
+    }
+}
+
 template <typename F>
 static double timeit(F f, int iters) {
     hipEvent_t e0, e1;
@@ -179,6 +222,10 @@ int main(int argc, char **argv) {
     RUN("ycarry8", hipLaunchKernelGGL(st_ycarry<8>, dim3(gx4, (n + yb - 1) / yb),
                                       dim3(256), 0, 0, A, B, n, pitch));
     RUN("ycarry16", hipLaunchKernelGGL(st_ycarry<16>, dim3(gx4, (n + 15) / 16),
+                                       dim3(256), 0, 0, A, B, n, pitch));
+    RUN("yunroll4", hipLaunchKernelGGL(st_yunroll<4>, dim3(gx4, (n + 3) / 4),
+                                       dim3(256), 0, 0, A, B, n, pitch));
+    RUN("yunroll8", hipLaunchKernelGGL(st_yunroll<8>, dim3(gx4, (n + 7) / 8),
                                        dim3(256), 0, 0, A, B, n, pitch));
     return 0;
 }
