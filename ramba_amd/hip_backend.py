@@ -324,7 +324,9 @@ class HipBackend:
         nx = shape[nd - 1]
         cap = int(os.environ.get("RAMBA_GRID_CAP", "16384"))
         gx = max(1, min(cap, (nx + 256 * V - 1) // (256 * V)))
-        gy = max(1, min(8192, shape[nd - 2])) if nd >= 2 else 1
+        yb = getattr(gk, "yblock", 1)
+        gy = max(1, min(8192, (shape[nd - 2] + yb - 1) // yb)) \
+            if nd >= 2 else 1
         gz = max(1, min(64, shape[0])) if nd >= 3 else 1
         if gk.nred:
             # bound the partials array: the finish kernel is one block
