@@ -316,6 +316,19 @@ class LaneEmitter:
                   "sinh": "sinh", "cosh": "cosh", "tanh": "tanh",
                   "arcsin": "asin", "arccos": "acos", "arctan": "atan",
                   "exp": "exp", "log": "log"}
+        if op in ("floor", "ceil", "trunc", "rint"):
+            if src_dt.kind in "iub":
+                return self.fresh(ct, f"({a})")
+            fn = {"floor": "floor", "ceil": "ceil", "trunc": "trunc",
+                  "rint": "rint"}[op]
+            return self.fresh(ct, f"{fn}{sfx}(({ct})({a}))")
+        if op == "sign":
+            if src_dt.kind == "f":
+                return self.fresh(
+                    ct, f"__builtin_isnan((double)({a})) ? ({ct})({a}) : "
+                        f"(({a}) > 0 ? ({ct})1 : (({a}) < 0 ? ({ct})-1 "
+                        f": ({ct})0))")
+            return self.fresh(ct, f"(({a}) > 0) - (({a}) < 0)")
         if op in simple:
             return self.fresh(ct, f"{simple[op]}{sfx}(({ct})({a}))")
         if op == "neg":

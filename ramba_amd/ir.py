@@ -132,6 +132,8 @@ UNOPS = {
     "sinh": np.sinh, "cosh": np.cosh, "tanh": np.tanh,
     "arcsin": np.arcsin, "arccos": np.arccos, "arctan": np.arctan,
     "neg": np.negative, "exp": np.exp, "log": np.log,
+    "floor": np.floor, "ceil": np.ceil, "trunc": np.trunc,
+    "rint": np.rint, "sign": np.sign,
     "isnan": np.isnan, "isinf": np.isinf, "isfinite": np.isfinite,
     "logical_not": np.logical_not, "invert": np.invert,
 }

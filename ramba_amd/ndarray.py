@@ -354,6 +354,15 @@ class ndarray:
     def any(self, axis=None, keepdims=False, **kw):
         return self._reduce("any", axis=axis, keepdims=keepdims)
 
+    def clip(self, lo=None, hi=None, **kw):
+        """reference TestBasic clip (test_distributed_array.py)."""
+        out = self
+        if lo is not None:
+            out = out.maximum(lo)
+        if hi is not None:
+            out = out.minimum(hi)
+        return out
+
     def cumsum(self, axis=None, dtype=None, **kw):
         """1-D cumulative sum (reference scumulative/cumsum,
         ramba.py:10057-10171, 9675)."""
@@ -406,7 +415,8 @@ _UNOP_METHODS = {
     "arccos": "arccos", "arctan": "arctan", "__neg__": "neg",
     "exp": "exp", "log": "log", "isnan": "isnan", "isinf": "isinf",
     "isfinite": "isfinite", "logical_not": "logical_not",
-    "__invert__": "invert",
+    "__invert__": "invert", "floor": "floor", "ceil": "ceil",
+    "trunc": "trunc", "rint": "rint", "sign": "sign",
 }
 
 for _m, _op in _BINOP_METHODS.items():
@@ -658,6 +668,17 @@ isnan = _module_unop("isnan")
 isinf = _module_unop("isinf")
 isfinite = _module_unop("isfinite")
 logical_not = _module_unop("logical_not")
+floor = _module_unop("floor")
+ceil = _module_unop("ceil")
+trunc = _module_unop("trunc")
+rint = _module_unop("rint")
+sign = _module_unop("sign")
+
+
+def clip(a, lo, hi, **kw):
+    if isinstance(a, ndarray):
+        return a.clip(lo, hi)
+    return np.clip(a, lo, hi)
 
 
 def _module_reduction(name):

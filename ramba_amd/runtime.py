@@ -18,6 +18,7 @@ container that covers its division box plus a border ring
 """
 
 import os
+import time
 
 import numpy as np
 
@@ -164,6 +165,15 @@ class Runtime:
         return tuple(parts)
 
     def execute_group(self, group):
+        # timing accumulators: the analog of the reference's add_time
+        # summaries (ramba/ramba.py:945-1022, RAMBA_TIMING)
+        t0 = time.perf_counter()
+        try:
+            self._execute_group(group)
+        finally:
+            add_time("run_deferred_ops", time.perf_counter() - t0)
+
+    def _execute_group(self, group):
         live, dead = deferred.compute_live_vars(group)
         cache = getattr(self, "_recipe_cache", None)
         if cache is None:
@@ -399,6 +409,7 @@ class Runtime:
             return (dst, src, bd.gid, tuple(bx[0]), tuple(bx[1]), tgt[0],
                     str(tgt[1]))
         msgs = sorted(msgs, key=key)
+        tc0 = time.perf_counter()
         sends, recvs = [], []
         for m in msgs:
             dst, src, bd, bx, tgt = m
@@ -416,6 +427,7 @@ class Runtime:
                 vname = tgt[1]
                 need, _ = temp_geom[vname]
                 self.backend.unpack_box_to_temp(vname, need, bx, buf)
+        add_time("part_exchange", time.perf_counter() - tc0)
 
     # ------------------------------------------------------------------
     def finish_reduction(self, pend):

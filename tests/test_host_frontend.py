@@ -485,3 +485,80 @@ class TestMaskedWrite:
             a[m] = 99.0
             return a
         run_both(impl, ra)
+
+
+class TestFusionCount:
+    """The reference asserts fusion via timing (TestFusion,
+    test_distributed_array.py:112-200: 10 fused updates < 2x one update);
+    we assert the launch count directly."""
+
+    def _count_launches(self, ra, prog):
+        be = ra._deferred.get_runtime().backend
+        calls = []
+        orig = be.launch
+
+        def spy(plan, recipe=None):
+            calls.append(len(plan.statements))
+            return orig(plan, recipe)
+
+        be.launch = spy
+        try:
+            prog()
+        finally:
+            be.launch = orig
+        return calls
+
+    def test_flagship_is_one_kernel(self, ra):
+        def prog():
+            A = ra.arange(4000) / 1000.0
+            ra.sync()
+            B = ra.sin(A)
+            C = ra.cos(A)
+            D = B * B + C ** 2
+            ra.sync()
+            return D
+        calls = self._count_launches(ra, prog)
+        assert len(calls) == 2, calls   # arange fill + ONE fused chain
+
+    def test_ten_updates_fuse(self, ra):
+        def prog():
+            a = ra.arange(1000) * 1.0
+            ra.sync()
+            for _ in range(10):
+                a += 1.5
+            ra.sync()
+        calls = self._count_launches(ra, prog)
+        assert len(calls) == 2, calls   # all ten updates in one kernel
+
+    def test_stencil_is_one_kernel(self, ra):
+        import numpy as _np
+
+        def prog():
+            A = ra.arange(500) * 1.0
+            B = ra.zeros(500)
+            ra.sync()
+            B[2:-2] = (0.1 * A[:-4] + 0.2 * A[1:-3] + 0.4 * A[2:-2]
+                       + 0.2 * A[3:-1] + 0.1 * A[4:])
+            ra.sync()
+        calls = self._count_launches(ra, prog)
+        assert len(calls) == 2, calls
+
+    def test_reduction_fuses_with_producer(self, ra):
+        def prog():
+            A = ra.arange(3000) / 1000.0
+            ra.sync()
+            s = (ra.sin(A) ** 2 + ra.cos(A) ** 2).sum()
+        calls = self._count_launches(ra, prog)
+        assert len(calls) == 2, calls   # producer + reduction in ONE kernel
+
+
+def test_timing_accumulators(ra):
+    """§5 aux: add_time/get_timing analog (reference ramba.py:945-1022)."""
+    ra.reset_timing()
+    a = ra.arange(1000) / 10.0
+    b = ra.sin(a)
+    ra.sync()
+    t = ra.get_timing()
+    assert "run_deferred_ops" in t
+    count, secs = t["run_deferred_ops"]
+    assert count >= 1 and secs >= 0

@@ -151,3 +151,28 @@ class TestOpMatrixGpu:
                     + (np_.sqrt(f) + np_.tanh(f)
                        + np_.exp(-f * 0.001)).astype(np.int64))
         run_both(impl, ra_gpu)
+
+
+class TestRoundingOps:
+    @pytest.mark.parametrize("op", ["floor", "ceil", "trunc", "rint",
+                                    "sign"])
+    def test_float(self, ra, op):
+        def impl(np_):
+            a = np_.arange(-100, 100) * 0.37
+            return getattr(np_, op)(a)
+        run_both(impl, ra)
+
+    @pytest.mark.parametrize("op", ["floor", "ceil", "sign"])
+    def test_int(self, ra, op):
+        def impl(np_):
+            a = np_.arange(-50, 50)
+            return getattr(np_, op)(a)
+        run_both(impl, ra)
+
+    def test_clip(self, ra):
+        def impl(np_):
+            a = np_.arange(200) - 100
+            if np_ is np:
+                return np.clip(a, -20, 55)
+            return a.clip(-20, 55)
+        run_both(impl, ra)
