@@ -166,6 +166,11 @@ def compute_regular_schedule(num_workers, size, dims_do_not_distribute=()):
     on a single node).
     """
     num_dim = len(size)
+    if any(s == 0 for s in size):
+        # zero-size arrays: every rank empty
+        divisions = np.zeros((num_workers, 2, num_dim), dtype=np.int64)
+        divisions[:, 1, :] = -1
+        return divisions
     best, best_value = None, math.inf
     for factored in get_dim_factors(num_workers, num_dim):
         ok = True
@@ -188,8 +193,14 @@ def compute_regular_schedule(num_workers, size, dims_do_not_distribute=()):
         if surface < best_value:
             best_value = surface
             best = factored
-    assert best is not None, (
-        f"no valid partition of {size} over {num_workers} workers")
+    if best is None:
+        # array too small to split over every worker (reference
+        # make_uni_dist / do_not_distribute, common.py): rank 0 owns it
+        divisions = np.zeros((num_workers, 2, num_dim), dtype=np.int64)
+        divisions[:, 1, :] = -1
+        divisions[0, 0, :] = 0
+        divisions[0, 1, :] = np.array(size, dtype=np.int64) - 1
+        return divisions
     divisions = np.empty((num_workers, 2, num_dim), dtype=np.int64)
     create_divisions(divisions, size, best)
     return divisions

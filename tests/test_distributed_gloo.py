@@ -189,3 +189,14 @@ def test_cumsum_spmd(world):
         import numpy as _np
         return _np.concatenate([c.asarray(), d.asarray() * 1.0])
     """, world=world, tol=1e-12)
+
+
+def test_tiny_array_uni_dist():
+    # array smaller than the worker count: rank 0 owns it (reference
+    # make_uni_dist); arithmetic and gather still work on every rank
+    run_spmd("""
+        a = np_.arange(3)
+        b = a * 2 + 1
+        z = np_.arange(0)
+        return b
+    """, world=4)

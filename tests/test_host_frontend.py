@@ -562,3 +562,10 @@ def test_timing_accumulators(ra):
     assert "run_deferred_ops" in t
     count, secs = t["run_deferred_ops"]
     assert count >= 1 and secs >= 0
+
+
+def test_zero_size_arrays(ra):
+    assert ra.arange(0).asarray().shape == (0,)
+    assert (ra.arange(0) + 1.0).asarray().shape == (0,)
+    a = ra.zeros(5)
+    assert a[2:2].asarray().shape == (0,)
