@@ -10,13 +10,12 @@ materialisation rule, ramba.py:8123) and the group is handed to the runtime
 """
 
 import numbers
-import weakref
 
 import numpy as np
 
 from . import ir
-from .common import dprint, add_time, default_border, default_divisions
-from .shardview import View, exec_boxes, exec_boxes_eq
+from .common import dprint
+from .shardview import exec_boxes, exec_boxes_eq
 
 
 _gid_counter = [0]

@@ -13,7 +13,6 @@ import os
 import numpy as np
 
 from . import codegen
-from .common import dprint
 from .shardview import box_shape
 
 LIBPATH = os.path.join(os.path.dirname(os.path.abspath(__file__)), "_lib",

@@ -9,8 +9,8 @@ oracle).  The op vocabulary mirrors the reference's op tables
 (ramba/ramba.py:7893-7918 binops, 7949-7973 unaries, 7981-7993 reductions).
 """
 
-from dataclasses import dataclass, field
-from typing import Any, Optional, Tuple
+from dataclasses import dataclass
+from typing import Any
 
 import numpy as np
 

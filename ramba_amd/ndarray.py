@@ -12,7 +12,7 @@ import numbers
 import numpy as np
 
 from . import deferred, ir
-from .common import default_border, default_divisions, dprint
+from .common import default_border, default_divisions
 from .shardview import View
 
 

@@ -23,10 +23,9 @@ import time
 import numpy as np
 
 from . import deferred, ir
-from .common import (dprint, default_border, add_time,
-                     default_divisions)
-from .shardview import (box_contains, box_empty, box_eq, box_intersect,
-                        box_shape, box_subtract, View)
+from .common import add_time, default_border, default_divisions
+from .shardview import (box_contains, box_intersect, box_shape,
+                        box_subtract)
 
 
 class OperandPlan:
@@ -153,13 +152,13 @@ class Runtime:
                  ("part", group.part_view, group.part_divs.tobytes(),
                   group.flex)]
         for name, oi in group.arr_vars.items():
-            parts.append((name, name in live, oi.bd.shape, str(oi.bd.dtype),
+            parts.append((name, name in live, oi.bd.shape, oi.bd.dtype,
                           oi.bd.border, oi.bd.divisions.tobytes(),
                           oi.bd.is_flex, oi.bd.constructed, oi.view,
                           oi.written))
-        parts.append(tuple(sorted((n, str(dt))
+        parts.append(tuple(sorted((n, dt.name)
                                   for n, (v, dt) in group.scalars.items())))
-        parts.append(tuple((s.acc, s.kind, str(s.dtype))
+        parts.append(tuple((s.acc, s.kind, s.dtype)
                            for s, _ in group.reductions))
         parts.append(tuple((st.target, st.expr) for st in group.statements))
         return tuple(parts)
@@ -607,7 +606,7 @@ Runtime.reduce_axes_op = reduce_axes_op
 # ---------------------------------------------------------------------------
 
 def cumsum_op(self, arr, out_dtype):
-    from .shardview import exec_boxes as _eb, View as _View
+    from .shardview import exec_boxes as _eb
     bd, v = arr.bdarray, arr.view
     assert v.ndim == 1, "cumsum is 1-D (reference scumulative)"
     lbs = _eb(v, bd.divisions)
