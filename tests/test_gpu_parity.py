@@ -295,3 +295,10 @@ class TestCumsum:
             c = np_.arange(50_000).cumsum()
             return (c % 997) + c[10]
         run_both(impl, ra_gpu)
+
+
+def test_fuzz_programs_gpu(ra_gpu):
+    """Seeded random-program parity sweep through the HIP path."""
+    from fuzz_programs import check_seed
+    for seed in range(240):
+        check_seed(ra_gpu, seed)

@@ -569,3 +569,10 @@ def test_zero_size_arrays(ra):
     assert (ra.arange(0) + 1.0).asarray().shape == (0,)
     a = ra.zeros(5)
     assert a[2:2].asarray().shape == (0,)
+
+
+def test_fuzz_programs_cpu(ra):
+    """Seeded random-program parity sweep (tests/fuzz_programs.py)."""
+    from fuzz_programs import check_seed
+    for seed in range(120):
+        check_seed(ra, seed)
