@@ -71,5 +71,10 @@ _deferred.get_runtime = _get_runtime_auto
 
 def shutdown():
     _deferred.flush()
+    from .common import ntiming, get_timing
+    if ntiming >= 1:
+        print("[ramba_amd timing] tag: (count, seconds)")
+        for tag, (c, secs) in sorted(get_timing().items()):
+            print(f"  {tag}: ({c}, {secs:.6f})")
     _initialized["done"] = False
     _deferred.set_runtime(None)
