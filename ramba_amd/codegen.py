@@ -1105,7 +1105,8 @@ def pack_finish_args(partials_ptr, npartials, out_ptrs):
 # axes; replaces the reference's axis_reduce loops, ramba/ramba.py:8231-8244)
 # ---------------------------------------------------------------------------
 
-def generate_axis_reduce(nd, axes, in_dtype, out_dtype, kind):
+def generate_axis_reduce(nd, axes, in_dtype, out_dtype, kind,
+                         chunked=False):
     """Local phase: reduce a strided view over `axes` into a contiguous
     partial buffer covering the local out box.
 
@@ -1128,6 +1129,7 @@ def generate_axis_reduce(nd, axes, in_dtype, out_dtype, kind):
     ot = ctype(out_dtype)
     comb, _ = ir.REDUCTIONS[kind]
     lane_split = (nd - 1) in axes
+    assert not (chunked and (lane_split or len(axes) != 1))
 
     def combc(a, b):
         if comb == "add":
@@ -1163,16 +1165,24 @@ def generate_axis_reduce(nd, axes, in_dtype, out_dtype, kind):
         fields.append(("in_s", d))
     L.append(f"  {ot}* __restrict__ out;")
     fields.append(("out_ptr", None))
+    if chunked:
+        # split the (single) reduced axis into nchunk ragged chunks of
+        # clen; chunk is an extra slowest OUT axis (parallelism for
+        # small-nout reductions like sum(axis=0))
+        L.append("  i64 nchunk; i64 clen; i64 ktot;")
+        fields.append(("chunk", None))
     L.append("};")
 
     key = hashlib.sha256(
-        f"axred:{nd}:{axes}:{in_dtype}:{out_dtype}:{kind}".encode()
-    ).hexdigest()[:20]
+        f"axred:{nd}:{axes}:{in_dtype}:{out_dtype}:{kind}:{chunked}"
+        .encode()).hexdigest()[:20]
     kname = f"ax_{key}"
     L.append(f'extern "C" __global__ void __launch_bounds__(256) '
              f"{kname}(AxArgs a) {{")
     # total out elements and per-thread mapping
     tot = " * ".join([f"a.oe{j}" for j in range(len(out_axes))]) or "1"
+    if chunked:
+        tot = f"({tot}) * a.nchunk"
     L.append(f"  const i64 nout = {tot};")
     if lane_split:
         L.append("  const int lane = threadIdx.x & 63;")
@@ -1185,12 +1195,14 @@ def generate_axis_reduce(nd, axes, in_dtype, out_dtype, kind):
         L.append("  const i64 os = (i64)gridDim.x * 256;")
         L.append("  for (; o < nout; o += os) {")
         L.append("    i64 rem = o;")
-    # decompose out index (row-major over out extents)
+    # decompose out index (row-major over out extents; chunk slowest)
     L.append("    i64 base = a.in_off;")
     for j in range(len(out_axes) - 1, -1, -1):
         d = out_axes[j]
         L.append(f"    {{ i64 ix = rem % a.oe{j}; rem /= a.oe{j}; "
                  f"base += ix * a.in_s{d}; }}")
+    if chunked:
+        L.append("    const i64 chunk = rem;")
     L.append(f"    {ot} acc = {init};")
     # reduction loops
     if lane_split:
@@ -1224,6 +1236,20 @@ def generate_axis_reduce(nd, axes, in_dtype, out_dtype, kind):
         L.append(f"      acc = {combc('acc', 'other')};")
         L.append("    }")
         L.append("    if (lane == 0) a.out[w] = acc;")
+    elif chunked:
+        d = axes[0]
+        L.append("    const i64 k0 = chunk * a.clen;")
+        L.append("    const i64 k1 = k0 + a.clen < a.ktot ? k0 + a.clen "
+                 ": a.ktot;")
+        L.append("    for (i64 kk = k0; kk < k1; ++kk) {")
+        if kind in ("all", "any"):
+            L.append(f"      {ot} v = (a.in[base + kk * a.in_s{d}] != 0) "
+                     "? 1 : 0;")
+        else:
+            L.append(f"      {ot} v = ({ot})a.in[base + kk * a.in_s{d}];")
+        L.append(f"      acc = {combc('acc', 'v')};")
+        L.append("    }")
+        L.append("    a.out[o] = acc;")
     else:
         ind = "    "
         for j, d in enumerate(axes):
@@ -1258,7 +1284,8 @@ def _axinit(kind, dtype):
 
 
 def pack_axis_reduce_args(fields, out_extents_by_axis, red_extents_by_axis,
-                          in_ptr, in_off, in_strides, out_ptr):
+                          in_ptr, in_off, in_strides, out_ptr,
+                          chunk_spec=None):
     out = bytearray()
     for kind, d in fields:
         if kind == "oe":
@@ -1273,4 +1300,7 @@ def pack_axis_reduce_args(fields, out_extents_by_axis, red_extents_by_axis,
             out += struct.pack("<q", in_strides[d])
         elif kind == "out_ptr":
             out += struct.pack("<Q", out_ptr)
+        elif kind == "chunk":
+            nchunk, clen, ktot = chunk_spec
+            out += struct.pack("<qqq", nchunk, clen, ktot)
     return bytes(out)

@@ -487,17 +487,30 @@ __global__ __launch_bounds__(256) void cumsum_k3(const T *__restrict__ in,
                                                  int64_t out_off,
                                                  const T *__restrict__ bsums,
                                                  T base) {
+    // stage the chunk through LDS so global loads AND stores stay
+    // coalesced (per-lane contiguous-16 global stores are partial-line
+    // read-modify-write: measured 572 GB/s -> LDS-staged ~streaming
+    // rate).  One pad element per 16 keeps the per-thread contiguous
+    // reads off a 32-way bank conflict.  ~35 KB of LDS for fp64 chunks.
+#define LIDX(g) ((g) + ((g) >> 4))
+    __shared__ T lds[SCAN_CHUNK + SCAN_THREADS];
     int64_t blk = blockIdx.x;
     int64_t nblocks = gridDim.x;
     for (; blk * SCAN_CHUNK < n; blk += nblocks) {
         int64_t b0 = blk * SCAN_CHUNK;
-        int64_t t0 = b0 + (int64_t)threadIdx.x * SCAN_ITEMS;
-        // per-thread sum of its contiguous 16 elements
-        T s = (T)0;
+        // coalesced load into LDS
         for (int j = 0; j < SCAN_ITEMS; ++j) {
-            int64_t i = t0 + j;
-            if (i < n) s += in[in_off + i * in_stride];
+            int64_t i = b0 + j * SCAN_THREADS + threadIdx.x;
+            if (i < n) lds[LIDX(j * SCAN_THREADS + threadIdx.x)] =
+                in[in_off + i * in_stride];
         }
+        __syncthreads();
+        // per-thread sum of its contiguous 16 LDS elements
+        int64_t l0 = (int64_t)threadIdx.x * SCAN_ITEMS;
+        int64_t lmax = n - b0;
+        T s = (T)0;
+        for (int j = 0; j < SCAN_ITEMS; ++j)
+            if (l0 + j < lmax) s += lds[LIDX(l0 + j)];
         // exclusive scan across the block's threads
         T x = s;
         int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
@@ -506,18 +519,23 @@ __global__ __launch_bounds__(256) void cumsum_k3(const T *__restrict__ in,
             if (lane >= off) x += y;
         }
         __shared__ T wsum[4];
-        __syncthreads();
         if (lane == 63) wsum[wid] = x;
         __syncthreads();
         T wbase = (T)0;
         for (int w = 0; w < wid; ++w) wbase += wsum[w];
         T run = base + bsums[blk] + wbase + x - s;
         for (int j = 0; j < SCAN_ITEMS; ++j) {
-            int64_t i = t0 + j;
-            if (i < n) {
-                run += in[in_off + i * in_stride];
-                out[out_off + i] = run;
+            if (l0 + j < lmax) {
+                run += lds[LIDX(l0 + j)];
+                lds[LIDX(l0 + j)] = run;
             }
+        }
+        __syncthreads();
+        // coalesced store from LDS
+        for (int j = 0; j < SCAN_ITEMS; ++j) {
+            int64_t i = b0 + j * SCAN_THREADS + threadIdx.x;
+            if (i < n) out[out_off + i] =
+                lds[LIDX(j * SCAN_THREADS + threadIdx.x)];
         }
         __syncthreads();
     }

@@ -146,3 +146,12 @@ def test_axis_reduce_kernels_compile():
         rc = lib.rt_compile_check(src.encode())
         assert rc == 0, (f"{nd}d axes={axes} {kind}: "
                          + lib.rt_last_error().decode() + "\n" + src)
+    # chunked stage-1 variants (small-nout parallel split)
+    for nd, axes, idt, odt, kind in [
+            (2, (0,), np.float64, np.float64, "sum"),
+            (2, (0,), np.int64, np.int64, "max"),
+            (3, (0,), np.float32, np.float32, "sum")]:
+        key, src, kname, fields, ls = generate_axis_reduce(
+            nd, axes, np.dtype(idt), np.dtype(odt), kind, chunked=True)
+        rc = lib.rt_compile_check(src.encode())
+        assert rc == 0, lib.rt_last_error().decode() + "\n" + src

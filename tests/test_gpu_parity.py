@@ -302,3 +302,21 @@ def test_fuzz_programs_gpu(ra_gpu):
     from fuzz_programs import check_seed
     for seed in range(240):
         check_seed(ra_gpu, seed)
+
+
+def test_large_axis0_sum_chunked(ra_gpu):
+    """sum(axis=0) with K large enough to take the two-stage chunked
+    path (parallelism fix for small-nout reductions)."""
+    def impl(np_):
+        a = np_.fromfunction(lambda x, y: (x * 7 + y * 3) % 1000,
+                             (8192, 4099), dtype=np.int64)
+        return a.sum(axis=0)
+    run_both(impl, ra_gpu)
+
+
+def test_large_axis0_max_chunked_float(ra_gpu):
+    def impl(np_):
+        a = np_.fromfunction(lambda x, y: ((x * 131 + y * 17) % 977) * 0.5,
+                             (4096, 3001), dtype=np.float64)
+        return a.max(axis=0)
+    run_both(impl, ra_gpu)
