@@ -76,7 +76,10 @@ class HipBackend:
                 "ramba_amd HIP backend requires an AMD GPU "
                 "(torch.cuda.is_available() is False); no CPU fallback.")
         self.lib = _load_lib()
-        self.device = int(os.environ.get("LOCAL_RANK", "0")) \
+        # RAMBA_DEVICE overrides (multi-rank-on-one-GPU dry-runs);
+        # production multi-GPU uses torchrun's LOCAL_RANK
+        self.device = int(os.environ.get(
+            "RAMBA_DEVICE", os.environ.get("LOCAL_RANK", "0"))) \
             if device is None else device
         torch.cuda.set_device(self.device)
         self._check(self.lib.rt_init(self.device), "rt_init")
