@@ -625,16 +625,27 @@ class KernelGen:
         c = self.classes[name]
         return c.base if c.base is not None else name
 
-    def addr_parts(self, name, inner_expr):
+    def addr_parts(self, name, inner_expr, row=None):
         """(symbolic parts, constant) of the element index into `name`'s
-        buffer — the constant is kept separate so equal addresses CSE."""
+        buffer — the constant is kept separate so equal addresses CSE.
+
+        `row` = (base_expr, ky): in the unrolled y-block the row index is
+        base_expr + ky, so ky and the fold's krow collapse into one
+        literal and equal absolute rows produce IDENTICAL part strings
+        (true cross-row vector sharing)."""
         c = self.classes[name]
         eff = self.eff(name)
         parts = [f"a.{eff}_off"]
         const = int(c.delta)
         for d in range(self.nd - 1):
             if c.outer_g[d]:
-                if d == 0 and c.krow:
+                if d == 0 and row is not None:
+                    base_expr, ky = row
+                    k = ky + c.krow
+                    parts.append(
+                        f"({base_expr} + {k}) * a.{eff}_s{d}" if k else
+                        f"({base_expr}) * a.{eff}_s{d}")
+                elif d == 0 and c.krow:
                     parts.append(f"(i0 + ({c.krow})) * a.{eff}_s{d}")
                 else:
                     parts.append(f"i{d} * a.{eff}_s{d}")
@@ -644,8 +655,8 @@ class KernelGen:
             parts.append(f"({inner_expr}) * a.{eff}_sx")
         return parts, const
 
-    def addr_expr(self, name, inner_expr, extra_const=0):
-        parts, const = self.addr_parts(name, inner_expr)
+    def addr_expr(self, name, inner_expr, extra_const=0, row=None):
+        parts, const = self.addr_parts(name, inner_expr, row=row)
         const += extra_const
         if const:
             parts = parts + [f"({const})"]
@@ -654,14 +665,18 @@ class KernelGen:
     # -- body generation ------------------------------------------------------
 
     def gen_lane_body(self, tag, inner_expr, indent, vec_lane=None,
-                      svp=""):
+                      svp="", row=None):
         """Emit loads + statements + stores for ONE element.
 
         vec_lane: (vecvar-suffix, lane index) when inside the vectorised
         body -- loads come from preloaded vectors."""
         idx = []
         for d in range(self.nd - 1):
-            idx.append(f"a.gs{d} + i{d}")
+            if d == 0 and row is not None:
+                idx.append(f"a.gs0 + {row[0]} + {row[1]}"
+                           if row[1] else f"a.gs0 + {row[0]}")
+            else:
+                idx.append(f"a.gs{d} + i{d}")
         idx.append(f"a.gs{self.nd-1} + ({inner_expr})" if self.nd else "0")
         em = LaneEmitter(self, tag, idx)
         # loads
@@ -682,7 +697,7 @@ class KernelGen:
             else:
                 em.lines.append(
                     f"      {ct} {ldv} = a.{self.eff(name)}_p["
-                    f"{self.addr_expr(name, inner_expr)}];")
+                    f"{self.addr_expr(name, inner_expr, row=row)}];")
         # statements
         last_val = {}
         for (tgt, accinfo, expr) in self.ssa:
@@ -713,7 +728,8 @@ class KernelGen:
             else:
                 em.lines.append(
                     f"      a.{self.eff(name)}_p["
-                    f"{self.addr_expr(name, inner_expr)}] = {final};")
+                    f"{self.addr_expr(name, inner_expr, row=row)}] = "
+                    f"{final};")
         pad = " " * (indent - 6)
         return "\n".join(pad + ln.lstrip() if False else
                          (" " * indent) + ln.strip() for ln in em.lines)
@@ -737,23 +753,28 @@ class KernelGen:
         raise NotImplementedError(comb)
 
 
-    def emit_quad(self, L, body_ind, tagp):
+    def emit_quad(self, L, body_ind, tagp, row=None, shared=None):
         """Preloads + per-lane bodies + vector stores for one (x, row)
-        position.  `tagp` keeps locals unique across unrolled rows."""
+        position.  `tagp` keeps locals unique across unrolled rows.
+        `row`/`shared`: unrolled y-blocks share one load-CSE dict across
+        the rows, with addresses keyed by the folded absolute row — the
+        3 column vectors of a 5-pt stencil load 6 times per 4 rows
+        instead of 12 (the probe's ycarry structure, generically)."""
         V = self.vec
         self.preload = {}
-        cse = {}
-        nload = [0]
+        cse = {} if shared is None else shared
+        nload = self._nload = getattr(self, "_nload", [0])             if shared is not None else [0]
 
         def vec_load(name, extra_const):
-            parts, const = self.addr_parts(name, "vb")
+            parts, const = self.addr_parts(name, "vb", row=row)
             const += extra_const
             kk = (self.eff(name), tuple(parts), const)
             v = cse.get(kk)
             if v is None:
                 vt = self.vec_type(self.classes[name].dtype)
                 nload[0] += 1
-                v = f"vv{tagp}_{nload[0]}"
+                v = f"vq{nload[0]}" if shared is not None \
+                    else f"vv{tagp}_{nload[0]}"
                 addr = " + ".join(parts + ([f"({const})"] if const else []))
                 L.append(f"{' '*body_ind}const {vt} {v} = "
                          f"*(const {vt}*)&a.{self.eff(name)}_p[{addr}];")
@@ -777,14 +798,15 @@ class KernelGen:
             L.append(f"{' '*body_ind}{{ // lane {lane}")
             L.append(self.gen_lane_body(f"{tagp}_L{lane}", f"vb + {lane}",
                                         body_ind + 2, vec_lane=lane,
-                                        svp=tagp))
+                                        svp=tagp, row=row))
             L.append(f"{' '*body_ind}}}")
         for name, _ in sorted(self.written_ops.items()):
             c = self.classes[name]
             if c.inner == "v":
                 vt = self.vec_type(c.dtype)
                 L.append(f"{' '*body_ind}*({vt}*)&a.{self.eff(name)}_p["
-                         f"{self.addr_expr(name, 'vb')}] = sv{tagp}_{name};")
+                         f"{self.addr_expr(name, 'vb', row=row)}] = "
+                         f"sv{tagp}_{name};")
 
     # -- full source -----------------------------------------------------------
 
@@ -874,10 +896,13 @@ class KernelGen:
                      f"{yblock};")
             L.append(f"{' '*body_ind}for (; yb0 + {yblock} <= a.n0; "
                      f"yb0 += ys) {{")
+            # NOTE: no braces between rows — the shared vq* load vars
+            # must stay in scope across the unrolled rows
+            shared = {}
+            self._nload = [0]
             for ky in range(yblock):
-                L.append(f"{' '*(body_ind+2)}{{ const i64 i0 = yb0 + {ky};")
-                self.emit_quad(L, body_ind + 4, f"_Y{ky}")
-                L.append(f"{' '*(body_ind+2)}}}")
+                self.emit_quad(L, body_ind + 2, f"_Y{ky}",
+                               row=("yb0", ky), shared=shared)
             L.append(f"{' '*body_ind}}}")
             L.append(f"{' '*body_ind}for (i64 i0r = yb0; i0r < a.n0; "
                      f"++i0r) {{ const i64 i0 = i0r;")
