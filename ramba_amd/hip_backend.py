@@ -22,6 +22,11 @@ TORCH_DTYPE = None  # filled at init
 
 
 def _load_lib():
+    # on-disk kernel cache: pre-populated by the CPU suite's compile
+    # checks; ships with the snapshot so GPU boxes skip hiprtc entirely
+    kc = os.path.join(os.path.dirname(os.path.abspath(__file__)), "_kcache")
+    os.environ.setdefault("RAMBA_KCACHE", kc)
+    os.makedirs(os.environ["RAMBA_KCACHE"], exist_ok=True)
     if not os.path.exists(LIBPATH):
         raise RuntimeError(
             f"ramba_amd HIP runtime not built: {LIBPATH} missing. "

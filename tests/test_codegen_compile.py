@@ -20,12 +20,16 @@ from conftest import run_both
 
 class CompileCheckBackend(NumpyBackend):
     """Oracle backend that additionally feeds every generated kernel
-    through hiprtc for gfx950 (compile-only)."""
+    through hiprtc for gfx950 (compile-only; also seeds the on-disk
+    kernel cache that ships to GPU boxes)."""
 
     def __init__(self):
         super().__init__()
         assert os.path.exists(LIBPATH), \
             f"{LIBPATH} missing — run __graft_entry__.build() first"
+        kc = os.path.join(os.path.dirname(LIBPATH), "..", "_kcache")
+        os.environ.setdefault("RAMBA_KCACHE", os.path.abspath(kc))
+        os.makedirs(os.environ["RAMBA_KCACHE"], exist_ok=True)
         self.lib = ctypes.CDLL(LIBPATH)
         self.lib.rt_compile_check.argtypes = [ctypes.c_char_p]
         self.lib.rt_last_error.restype = ctypes.c_char_p
