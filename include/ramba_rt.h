@@ -85,6 +85,16 @@ int rt_cumsum(uintptr_t stream, const void *in, int64_t in_off,
               void *bsums, int64_t nblocks, void *total, double fbase,
               int64_t ibase, int dtype, int phase);
 
+/* single-pass decoupled-lookback cumsum (one read + one write per
+ * element; the chained cross-workgroup hand-off follows the CDNA4
+ * guide's agent-atomic discipline).  agg/inc: u64[nchunks]; flag:
+ * u32[nchunks] and ticket: u32[1], both zeroed by the host before every
+ * launch. */
+int rt_cumsum_scan(uintptr_t stream, const void *in, int64_t in_off,
+                   int64_t in_stride, int64_t n, void *out, int64_t out_off,
+                   void *agg, void *inc, void *flag, void *ticket,
+                   double fbase, int64_t ibase, int dtype);
+
 int rt_stream_sync(uintptr_t stream);
 int rt_device_sync(void);
 

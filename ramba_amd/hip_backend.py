@@ -61,6 +61,13 @@ def _load_lib():
                               ctypes.c_void_p, ctypes.c_int64,
                               ctypes.c_void_p, ctypes.c_double,
                               ctypes.c_int64, ctypes.c_int, ctypes.c_int]
+    lib.rt_cumsum_scan.argtypes = [ctypes.c_size_t, ctypes.c_void_p,
+                                   ctypes.c_int64, ctypes.c_int64,
+                                   ctypes.c_int64, ctypes.c_void_p,
+                                   ctypes.c_int64, ctypes.c_void_p,
+                                   ctypes.c_void_p, ctypes.c_void_p,
+                                   ctypes.c_void_p, ctypes.c_double,
+                                   ctypes.c_int64, ctypes.c_int]
     lib.rt_stream_sync.argtypes = [ctypes.c_size_t]
     return lib
 
@@ -548,13 +555,8 @@ _CS_DT = {"float64": 0, "float32": 1, "int64": 2, "int32": 3}
 _SCAN_CHUNK = 4096
 
 
-def _hb_cumsum_local_phase12(self, bd, off0, stride, n, out_dtype):
-    if str(np.dtype(out_dtype)) not in _CS_DT:
-        raise NotImplementedError(f"cumsum dtype {out_dtype}")
-    dt = _CS_DT[str(np.dtype(out_dtype))]
-    nblocks = max(1, (n + _SCAN_CHUNK - 1) // _SCAN_CHUNK)
+def _hb_cumsum_src(self, bd, off0, stride, n, out_dtype):
     cont = self._cont(bd)
-    # input must be read as out_dtype: if dtypes differ, cast via a temp
     src = cont
     if np.dtype(bd.dtype) != np.dtype(out_dtype):
         # pack (handles negative strides) then cast
@@ -564,6 +566,20 @@ def _hb_cumsum_local_phase12(self, bd, off0, stride, n, out_dtype):
         src = tmp.to(self._tdt(out_dtype))
         off0, stride = 0, 1
         self.temps["__cs_src__"] = src
+    return src, off0, stride
+
+
+def _hb_cumsum_local_phase12(self, bd, off0, stride, n, out_dtype):
+    if str(np.dtype(out_dtype)) not in _CS_DT:
+        raise NotImplementedError(f"cumsum dtype {out_dtype}")
+    dt = _CS_DT[str(np.dtype(out_dtype))]
+    src, off0, stride = self._hb_cumsum_src(bd, off0, stride, n, out_dtype)
+    if self.rt.world == 1:
+        # single-pass decoupled lookback runs in phase 3 (the cross-rank
+        # offset is trivially 0); nothing to pre-compute
+        self.temps["__cs_state__"] = (src, off0, stride, n, 0, dt)
+        return np.asarray(0, dtype=out_dtype)[()]
+    nblocks = max(1, (n + _SCAN_CHUNK - 1) // _SCAN_CHUNK)
     bsums = self.torch.empty(nblocks, dtype=self._tdt(out_dtype),
                              device="cuda")
     total = self.torch.empty(1, dtype=self._tdt(out_dtype), device="cuda")
@@ -585,10 +601,26 @@ def _hb_cumsum_local_phase12(self, bd, off0, stride, n, out_dtype):
 def _hb_cumsum_local_phase3(self, bd, off0, stride, n, out_bd, out_off,
                             offset, out_dtype):
     src, off0_, stride_, n_, nblocks, dt = self.temps["__cs_state__"]
-    bsums = self.temps["__cs_bsums__"]
     out = self._cont(out_bd)
     fb = float(offset) if np.dtype(out_dtype).kind == "f" else 0.0
     ib = int(offset) if np.dtype(out_dtype).kind != "f" else 0
+    if self.rt.world == 1:
+        nchunks = max(1, (n_ + _SCAN_CHUNK - 1) // _SCAN_CHUNK)
+        # flags + ticket zeroed every call (G16 "Re-initialise every call")
+        ws = self.torch.zeros(2 * nchunks + nchunks + 1,
+                              dtype=self.torch.int64, device="cuda")
+        agg = ws.data_ptr()
+        inc = agg + 8 * nchunks
+        flag = inc + 8 * nchunks
+        ticket = flag + 4 * nchunks
+        rc = self.lib.rt_cumsum_scan(
+            self._stream(), ctypes.c_void_p(src.data_ptr()), off0_, stride_,
+            n_, ctypes.c_void_p(out.data_ptr()), out_off,
+            ctypes.c_void_p(agg), ctypes.c_void_p(inc),
+            ctypes.c_void_p(flag), ctypes.c_void_p(ticket), fb, ib, dt)
+        self._check(rc, "rt_cumsum_scan")
+        return
+    bsums = self.temps["__cs_bsums__"]
     rc = self.lib.rt_cumsum(
         self._stream(), ctypes.c_void_p(src.data_ptr()), off0_, stride_, n_,
         ctypes.c_void_p(out.data_ptr()), out_off,
