@@ -641,6 +641,7 @@ def _hb_allgather_scalars(self, val, dtype):
     return [np.asarray(o.cpu().numpy()[0], dtype=dtype)[()] for o in outs]
 
 
+HipBackend._hb_cumsum_src = _hb_cumsum_src
 HipBackend.cumsum_local_phase12 = _hb_cumsum_local_phase12
 HipBackend.cumsum_local_phase3 = _hb_cumsum_local_phase3
 HipBackend.allgather_scalars = _hb_allgather_scalars
