@@ -617,12 +617,14 @@ __global__ __launch_bounds__(256) void cumsum_lookback(
     const T *__restrict__ in, int64_t in_off, int64_t in_stride, int64_t n,
     T *__restrict__ out, int64_t out_off, GU64 *agg, GU64 *inc, GU32 *flag,
     GU32 *ticket, T base) {
+    constexpr int ITEMS = 32;                 // 8192-elem chunks halve the
+    constexpr int CHUNK = 256 * ITEMS;        // cross-chunk chain length
 #define LIDX(g) ((g) + ((g) >> 4))
-    __shared__ T lds[SCAN_CHUNK + SCAN_THREADS];
+    __shared__ T lds[CHUNK + CHUNK / 16];
     __shared__ T wsum[4];
     __shared__ unsigned long long tick_s;
     __shared__ T excl_s;
-    const int64_t nchunks = (n + SCAN_CHUNK - 1) / SCAN_CHUNK;
+    const int64_t nchunks = (n + CHUNK - 1) / CHUNK;
     for (;;) {
         if (threadIdx.x == 0)
             tick_s = __hip_atomic_fetch_add(ticket, 1u, __ATOMIC_RELAXED,
@@ -630,19 +632,17 @@ __global__ __launch_bounds__(256) void cumsum_lookback(
         __syncthreads();
         const int64_t t = (int64_t)tick_s;
         if (t >= nchunks) return;
-        const int64_t b0 = t * SCAN_CHUNK;
-        // coalesced load into LDS
-        for (int j = 0; j < SCAN_ITEMS; ++j) {
-            int64_t i = b0 + j * SCAN_THREADS + threadIdx.x;
-            if (i < n) lds[LIDX(j * SCAN_THREADS + threadIdx.x)] =
+        const int64_t b0 = t * (int64_t)CHUNK;
+        for (int j = 0; j < ITEMS; ++j) {
+            int64_t i = b0 + j * 256 + threadIdx.x;
+            if (i < n) lds[LIDX(j * 256 + threadIdx.x)] =
                 in[in_off + i * in_stride];
         }
         __syncthreads();
-        // per-thread partial + block scan (as cumsum_k3)
-        int64_t l0 = (int64_t)threadIdx.x * SCAN_ITEMS;
+        int64_t l0 = (int64_t)threadIdx.x * ITEMS;
         int64_t lmax = n - b0;
         T s = (T)0;
-        for (int j = 0; j < SCAN_ITEMS; ++j)
+        for (int j = 0; j < ITEMS; ++j)
             if (l0 + j < lmax) s += lds[LIDX(l0 + j)];
         T x = s;
         int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
@@ -657,56 +657,74 @@ __global__ __launch_bounds__(256) void cumsum_lookback(
         T thread_excl = wbase + x - s;
         T block_total = (T)0;
         for (int w = 0; w < 4; ++w) block_total += wsum[w];
-        // publish aggregate, then look back for the exclusive prefix
+        // publish aggregate; wave 0 then resolves the exclusive prefix by
+        // a WAVE-PARALLEL lookback: 64 predecessor states per hop
         if (threadIdx.x == 0) {
             pub_u64(agg + t, block_total);
             asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
             __hip_atomic_store(flag + t, 1u, __ATOMIC_RELAXED,
                                __HIP_MEMORY_SCOPE_AGENT);
+        }
+        if (wid == 0) {
             T running = (T)0;
+            int64_t p = t - 1;
             unsigned spins = 0;
-            for (int64_t p = t - 1; p >= 0;) {
-                unsigned st = __hip_atomic_load(flag + p, __ATOMIC_RELAXED,
-                                                __HIP_MEMORY_SCOPE_AGENT);
-                if (st == 0u) {
-                    // progress is guaranteed (ticket p was claimed before
-                    // ours, so its owner is running and publishes its
-                    // aggregate before its own lookback); throttle the
-                    // poll rather than hard-bound it
-                    ++spins;
-                    if (spins > 4096) __builtin_amdgcn_s_sleep(32);
-                    else __builtin_amdgcn_s_sleep(1);
-                    continue;
-                }
-                if (st == 2u) {
-                    running += rd_u64<T>(inc + p);
+            while (p >= 0) {
+                int64_t q = p - lane;
+                unsigned st = q >= 0
+                    ? __hip_atomic_load(flag + q, __ATOMIC_RELAXED,
+                                        __HIP_MEMORY_SCOPE_AGENT)
+                    : 2u;
+                unsigned long long incl = __ballot(st >= 2u);
+                unsigned long long zero = __ballot(st == 0u);
+                int k = incl ? __ffsll((unsigned long long)incl) - 1 : -1;
+                int zk = zero ? __ffsll((unsigned long long)zero) - 1 : 64;
+                if (k >= 0 && zk > k) {
+                    // complete prefix: aggregates for lanes < k, the
+                    // inclusive at lane k
+                    T v = (T)0;
+                    if (lane < k) v = rd_u64<T>(agg + q);
+                    else if (lane == k && q >= 0) v = rd_u64<T>(inc + q);
+                    for (int o = 32; o > 0; o >>= 1) v += __shfl_down(v, o, 64);
+                    running += __shfl(v, 0, 64);
+                    p = -1;
                     break;
                 }
-                running += rd_u64<T>(agg + p);
-                --p;
+                if (zk >= 64) {
+                    // 64 complete partials: take them all, keep walking
+                    T v = q >= 0 ? rd_u64<T>(agg + q) : (T)0;
+                    for (int o = 32; o > 0; o >>= 1) v += __shfl_down(v, o, 64);
+                    running += __shfl(v, 0, 64);
+                    p -= 64;
+                    continue;
+                }
+                ++spins;
+                if (spins > 4096) __builtin_amdgcn_s_sleep(32);
+                else __builtin_amdgcn_s_sleep(1);
             }
-            T excl = base + running;
-            pub_u64(inc + t, excl + block_total);
-            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-            __hip_atomic_store(flag + t, 2u, __ATOMIC_RELAXED,
-                               __HIP_MEMORY_SCOPE_AGENT);
-            excl_s = excl;
+            if (lane == 0) {
+                T excl = base + running;
+                pub_u64(inc + t, excl + block_total);
+                asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+                __hip_atomic_store(flag + t, 2u, __ATOMIC_RELAXED,
+                                   __HIP_MEMORY_SCOPE_AGENT);
+                excl_s = excl;
+            }
         }
         __syncthreads();
         const T bbase = excl_s;
-        // apply + write back through LDS (coalesced stores)
         T run = bbase + thread_excl;
-        for (int j = 0; j < SCAN_ITEMS; ++j) {
+        for (int j = 0; j < ITEMS; ++j) {
             if (l0 + j < lmax) {
                 run += lds[LIDX(l0 + j)];
                 lds[LIDX(l0 + j)] = run;
             }
         }
         __syncthreads();
-        for (int j = 0; j < SCAN_ITEMS; ++j) {
-            int64_t i = b0 + j * SCAN_THREADS + threadIdx.x;
+        for (int j = 0; j < ITEMS; ++j) {
+            int64_t i = b0 + j * 256 + threadIdx.x;
             if (i < n) out[out_off + i] =
-                lds[LIDX(j * SCAN_THREADS + threadIdx.x)];
+                lds[LIDX(j * 256 + threadIdx.x)];
         }
         __syncthreads();
     }
@@ -720,7 +738,7 @@ int cumsum_lookback_launch(uintptr_t stream, const void *in, int64_t in_off,
                            void *ticket, double fbase, int64_t ibase) {
     T base = (T)fbase;
     if ((T)0.5 == 0) base = (T)ibase;  // integer T
-    int64_t nchunks = (n + SCAN_CHUNK - 1) / SCAN_CHUNK;
+    int64_t nchunks = (n + 8191) / 8192;
     int64_t grid = nchunks < 2048 ? nchunks : 2048;
     if (grid < 1) grid = 1;
     hipLaunchKernelGGL((cumsum_lookback<T>), dim3((unsigned)grid),

@@ -605,7 +605,7 @@ def _hb_cumsum_local_phase3(self, bd, off0, stride, n, out_bd, out_off,
     fb = float(offset) if np.dtype(out_dtype).kind == "f" else 0.0
     ib = int(offset) if np.dtype(out_dtype).kind != "f" else 0
     if self.rt.world == 1:
-        nchunks = max(1, (n_ + _SCAN_CHUNK - 1) // _SCAN_CHUNK)
+        nchunks = max(1, (n_ + 8191) // 8192)
         # flags + ticket zeroed every call (G16 "Re-initialise every call")
         ws = self.torch.zeros(2 * nchunks + nchunks + 1,
                               dtype=self.torch.int64, device="cuda")
