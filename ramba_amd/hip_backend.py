@@ -574,7 +574,8 @@ def _hb_cumsum_local_phase12(self, bd, off0, stride, n, out_dtype):
         raise NotImplementedError(f"cumsum dtype {out_dtype}")
     dt = _CS_DT[str(np.dtype(out_dtype))]
     src, off0, stride = self._hb_cumsum_src(bd, off0, stride, n, out_dtype)
-    if self.rt.world == 1:
+    if self.rt.world == 1 and os.environ.get("RAMBA_CUMSUM", "lookback") \
+            == "lookback":
         # single-pass decoupled lookback runs in phase 3 (the cross-rank
         # offset is trivially 0); nothing to pre-compute
         self.temps["__cs_state__"] = (src, off0, stride, n, 0, dt)
@@ -604,7 +605,7 @@ def _hb_cumsum_local_phase3(self, bd, off0, stride, n, out_bd, out_off,
     out = self._cont(out_bd)
     fb = float(offset) if np.dtype(out_dtype).kind == "f" else 0.0
     ib = int(offset) if np.dtype(out_dtype).kind != "f" else 0
-    if self.rt.world == 1:
+    if self.rt.world == 1 and nblocks == 0:
         nchunks = max(1, (n_ + 8191) // 8192)
         # flags + ticket zeroed every call (G16 "Re-initialise every call")
         ws = self.torch.zeros(2 * nchunks + nchunks + 1,
