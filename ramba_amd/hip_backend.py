@@ -574,10 +574,14 @@ def _hb_cumsum_local_phase12(self, bd, off0, stride, n, out_dtype):
         raise NotImplementedError(f"cumsum dtype {out_dtype}")
     dt = _CS_DT[str(np.dtype(out_dtype))]
     src, off0, stride = self._hb_cumsum_src(bd, off0, stride, n, out_dtype)
-    if self.rt.world == 1 and os.environ.get("RAMBA_CUMSUM", "lookback") \
+    if self.rt.world == 1 and os.environ.get("RAMBA_CUMSUM", "3pass") \
             == "lookback":
-        # single-pass decoupled lookback runs in phase 3 (the cross-rank
-        # offset is trivially 0); nothing to pre-compute
+        # REJECTED-by-measurement alternative, kept for reference: the
+        # single-pass decoupled-lookback scan (rt_cumsum_scan) moves
+        # 16 B/elem instead of 24 but its cross-chunk agent-atomic chain
+        # costs more than the saved read pass — same-box A/B at 5e8 i64:
+        # lookback 3.81 ms vs 3-phase 2.93 ms (hand-off pricing, CDNA4
+        # guide §price list).  Default stays 3-phase.
         self.temps["__cs_state__"] = (src, off0, stride, n, 0, dt)
         return np.asarray(0, dtype=out_dtype)[()]
     nblocks = max(1, (n + _SCAN_CHUNK - 1) // _SCAN_CHUNK)
