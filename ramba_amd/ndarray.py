@@ -200,6 +200,65 @@ class ndarray:
         else:
             raise TypeError(f"cannot assign {type(value)}")
 
+    # -- numpy interop (reference __array_ufunc__/__array_function__,
+    #    ramba/ramba.py:6860/6825: real-numpy calls on ramba arrays stay
+    #    distributed instead of silently gathering) ------------------------
+
+    _UFUNC_BIN = {"add": "__add__", "subtract": "__sub__",
+                  "multiply": "__mul__", "true_divide": "__truediv__",
+                  "divide": "__truediv__", "floor_divide": "__floordiv__",
+                  "remainder": "__mod__", "power": "__pow__",
+                  "minimum": "minimum", "maximum": "maximum",
+                  "greater": "__gt__", "less": "__lt__",
+                  "greater_equal": "__ge__", "less_equal": "__le__",
+                  "equal": "__eq__", "not_equal": "__ne__",
+                  "logical_and": "logical_and", "logical_or": "logical_or",
+                  "logical_xor": "logical_xor", "bitwise_and": "__and__",
+                  "bitwise_or": "__or__", "bitwise_xor": "__xor__",
+                  "left_shift": "__lshift__", "right_shift": "__rshift__"}
+    _UFUNC_UN = {"sin": "sin", "cos": "cos", "tan": "tan", "sinh": "sinh",
+                 "cosh": "cosh", "tanh": "tanh", "arcsin": "arcsin",
+                 "arccos": "arccos", "arctan": "arctan", "exp": "exp",
+                 "log": "log", "sqrt": "sqrt", "square": "square",
+                 "absolute": "__abs__", "negative": "__neg__",
+                 "floor": "floor", "ceil": "ceil", "trunc": "trunc",
+                 "rint": "rint", "sign": "sign", "isnan": "isnan",
+                 "isinf": "isinf", "isfinite": "isfinite",
+                 "logical_not": "logical_not", "invert": "__invert__"}
+
+    def __array_ufunc__(self, ufunc, method, *inputs, **kwargs):
+        if kwargs.get("out") is not None:
+            return NotImplemented
+        name = ufunc.__name__
+        if method == "__call__":
+            if len(inputs) == 2 and name in self._UFUNC_BIN:
+                a, b = inputs
+                if a is self:
+                    return getattr(self, self._UFUNC_BIN[name])(b)
+                if isinstance(a, (numbers.Number, np.generic)) \
+                        or isinstance(a, np.ndarray):
+                    # reversed: scalar/ndarray OP self
+                    m = self._UFUNC_BIN[name]
+                    rm = "__r" + m[2:] if m.startswith("__") else None
+                    if rm and hasattr(self, rm):
+                        return getattr(self, rm)(a)
+                    if isinstance(b, ndarray):
+                        return getattr(fromarray(np.asarray(a))
+                                       if isinstance(a, np.ndarray) else a,
+                                       m, None) or NotImplemented
+                return NotImplemented
+            if len(inputs) == 1 and name in self._UFUNC_UN:
+                return getattr(self, self._UFUNC_UN[name])()
+            return NotImplemented
+        if method == "reduce" and not kwargs.get("axis", None) \
+                and len(inputs) == 1 and inputs[0] is self:
+            red = {"add": "sum", "multiply": "prod", "minimum": "min",
+                   "maximum": "max", "logical_and": "all",
+                   "logical_or": "any"}.get(name)
+            if red:
+                return getattr(self, red)()
+        return NotImplemented
+
     # -- conversion ----------------------------------------------------------
 
     def __float__(self):

@@ -576,3 +576,27 @@ def test_fuzz_programs_cpu(ra):
     from fuzz_programs import check_seed
     for seed in range(120):
         check_seed(ra, seed)
+
+
+class TestNumpyInterop:
+    """np.<ufunc>(ramba_array) must stay distributed (reference
+    __array_ufunc__, ramba.py:6860)."""
+
+    def test_real_numpy_ufuncs_dispatch(self, ra):
+        a = ra.arange(500) / 100.0
+        r = np.sin(a)                       # REAL numpy module
+        assert isinstance(r, ra.ndarray), type(r)
+        np.testing.assert_allclose(r.asarray(), np.sin(np.arange(500) / 100.0),
+                                   rtol=1e-12, atol=1e-12)
+        r2 = np.add(a, 2.0)
+        assert isinstance(r2, ra.ndarray)
+        r3 = np.maximum(a, 3.0)
+        assert isinstance(r3, ra.ndarray)
+        r4 = np.add.reduce(a)
+        assert abs(float(r4) - (np.arange(500) / 100.0).sum()) < 1e-9
+
+    def test_scalar_op_array_reversed(self, ra):
+        a = ra.arange(100) * 1.0
+        r = np.subtract(5.0, a)
+        assert isinstance(r, ra.ndarray)
+        np.testing.assert_allclose(r.asarray(), 5.0 - np.arange(100) * 1.0)
