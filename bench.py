@@ -231,6 +231,16 @@ def main():
                                      capture_output=True, timeout=300,
                                      check=True)
                 r = json.loads(out.stdout.decode().strip().splitlines()[-1])
+                # BASELINE.md substitute (i): NumPy eager, single-threaded,
+                # the reference README's methodology row 1
+                ne = 20_000_000
+                xs = np.arange(ne, dtype=np.float64) * 0.001
+                t0 = time.perf_counter()
+                bs = np.sin(xs)
+                cs = np.cos(xs)
+                ds = bs * bs + cs ** 2
+                t1 = time.perf_counter()
+                del bs, cs, ds
                 cpu_baseline = {
                     "value": r["elems"] / r["last_iter_secs"] / 1e9,
                     "unit": "GElem/s",
@@ -240,6 +250,7 @@ def main():
                               f"loop (C/OpenMP restatement; the reference's "
                               f"Numba+MPI path is not runnable here — "
                               f"BASELINE.md)",
+                    "numpy_eager_gelem_s": ne / (t1 - t0) / 1e9,
                 }
             except Exception as e:  # noqa: BLE001
                 print(f"cpu_baseline failed: {e}", file=sys.stderr)

@@ -606,6 +606,19 @@ def linspace(start, stop, num=50, endpoint=True, retstep=False, dtype=None):
     return res
 
 
+def eye(n, m=None, k=0, dtype=None, **kw):
+    """Identity / shifted-diagonal matrix (reference creation routines,
+    ramba.py:8688-8991)."""
+    m = n if m is None else m
+    dt = np.dtype(np.float64 if dtype is None else dtype)
+    out = _new_result((int(n), int(m)), dt)
+    cond = ir.Bin("eq", ir.Bin("add", ir.Iota(0), int(k), ir.I64),
+                  ir.Iota(1), ir.BOOL)
+    deferred.add_op(out, "=", ir.Where(cond, 1, 0, dt),
+                    empty_like=empty_like)
+    return out
+
+
 def fromfunction(function, shape, dtype=None, **kw):
     """Iota-expression fill (analog of the reference's string fillers,
     create_array_executor ramba.py:8563).  `function` receives symbolic

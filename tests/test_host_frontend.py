@@ -600,3 +600,9 @@ class TestNumpyInterop:
         r = np.subtract(5.0, a)
         assert isinstance(r, ra.ndarray)
         np.testing.assert_allclose(r.asarray(), 5.0 - np.arange(100) * 1.0)
+
+
+def test_eye(ra):
+    run_both(lambda np_: np_.eye(17), ra)
+    run_both(lambda np_: np_.eye(9, 13, 2), ra)
+    run_both(lambda np_: np_.eye(8, 8, -3, dtype=np.int64), ra)
