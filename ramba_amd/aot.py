@@ -128,7 +128,6 @@ class AotCompileBackend:
 def seed():
     """Compile the kernels the bench/smoke programs need."""
     import ramba_amd as ra
-    from . import deferred
     if ra._initialized["done"]:
         ra.shutdown()
     be = AotCompileBackend()

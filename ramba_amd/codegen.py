@@ -39,15 +39,8 @@ CTYPE = {
     np.dtype(np.float64): "double",
 }
 
-VECN = {"double": "d2", "float": "f4", "long long": "l2"}
-
-
 def ctype(dt):
     return CTYPE[np.dtype(dt)]
-
-
-def is_float(dt):
-    return np.dtype(dt).kind == "f"
 
 
 # ---------------------------------------------------------------------------

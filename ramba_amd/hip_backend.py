@@ -171,9 +171,6 @@ class HipBackend:
         assert c is not None, f"shard for gid {bd.gid} not allocated"
         return c
 
-    def _strides_of(self, t):
-        return tuple(t.stride())
-
     def _copy(self, dst_t, dst_strides, dst_off, src_t, src_strides, src_off,
               shape, elemsize):
         nd = len(shape)
