@@ -134,21 +134,30 @@ class View:
         """
         if not isinstance(index, tuple):
             index = (index,)
-        # expand Ellipsis
+        # expand Ellipsis (None/np.newaxis consumes no view axis)
+        n_consume = sum(1 for ix in index
+                        if ix is not Ellipsis and ix is not None)
         if any(ix is Ellipsis for ix in index):
             k = index.index(Ellipsis)
-            n_given = sum(1 for ix in index if ix is not Ellipsis)
-            fill = (slice(None),) * (self.ndim - n_given)
+            fill = (slice(None),) * (self.ndim - n_consume)
             index = index[:k] + fill + index[k + 1:]
-        if len(index) > self.ndim:
+            n_consume = self.ndim
+        if n_consume > self.ndim:
             raise IndexError(
                 f"too many indices for array: array is {self.ndim}-dimensional,"
-                f" but {len(index)} were indexed")
-        index = index + (slice(None),) * (self.ndim - len(index))
+                f" but {n_consume} were indexed")
+        index = index + (slice(None),) * (self.ndim - n_consume)
 
         new_shape, new_map, new_steps = [], [], []
         offset = list(self.offset)
-        for v, ix in enumerate(index):
+        v = -1
+        for ix in index:
+            if ix is None:   # np.newaxis: insert a broadcast axis
+                new_shape.append(1)
+                new_map.append(-1)
+                new_steps.append(0)
+                continue
+            v += 1
             b, st = self.axis_map[v], self.steps[v]
             if isinstance(ix, (int, np.integer)):
                 i = int(ix)

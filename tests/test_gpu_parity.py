@@ -320,3 +320,21 @@ def test_large_axis0_max_chunked_float(ra_gpu):
                              (4096, 3001), dtype=np.float64)
         return a.max(axis=0)
     run_both(impl, ra_gpu)
+
+
+def test_broadcast_matmul_gpu(ra_gpu):
+    """matmul via broadcast+axis-reduce (reference TestApps pattern) on
+    the HIP path, checked against a real matmul."""
+    n, k, m = 64, 48, 80
+
+    def impl(np_):
+        A = np_.fromfunction(lambda i, j: (i * k + j) % 31 + 1.0, (n, k),
+                             dtype=np.float64)
+        B = np_.fromfunction(lambda i, j: (i * m + j) % 17 - 3.0, (k, m),
+                             dtype=np.float64)
+        prod = A[:, :, None] * B[None, :, :]
+        return prod.sum(axis=1)
+    r, ref = run_both(impl, ra_gpu)
+    Ae = np.fromfunction(lambda i, j: (i * k + j) % 31 + 1.0, (n, k))
+    Be = np.fromfunction(lambda i, j: (i * m + j) % 17 - 3.0, (k, m))
+    np.testing.assert_allclose(r, Ae @ Be, rtol=1e-12)

@@ -606,3 +606,46 @@ def test_eye(ra):
     run_both(lambda np_: np_.eye(17), ra)
     run_both(lambda np_: np_.eye(9, 13, 2), ra)
     run_both(lambda np_: np_.eye(8, 8, -3, dtype=np.int64), ra)
+
+
+class TestNewaxisMatmul:
+    """np.newaxis views + the reference's matmul-via-broadcast+reduce
+    pattern (TestApps, test_distributed_array.py:58-87)."""
+
+    def test_newaxis_views(self, ra):
+        def impl(np_):
+            a = np_.arange(12)
+            b = np_.arange(12)
+            return a[:, None] * b[None, :]
+        run_both(impl, ra)
+
+    def test_broadcast_matmul_kat(self, ra):
+        n, k, m = 9, 7, 11
+
+        def impl(np_):
+            A = np_.fromfunction(lambda i, j: i * k + j + 1, (n, k),
+                                 dtype=np.int64)
+            B = np_.fromfunction(lambda i, j: (i * m + j) % 13, (k, m),
+                                 dtype=np.int64)
+            prod = A[:, :, None] * B[None, :, :]
+            return prod.sum(axis=1)
+        r, ref = run_both(impl, ra)
+        # cross-check against an actual matmul
+        Ae = np.fromfunction(lambda i, j: i * k + j + 1, (n, k), dtype=np.int64)
+        Be = np.fromfunction(lambda i, j: (i * m + j) % 13, (k, m), dtype=np.int64)
+        assert np.array_equal(r, Ae @ Be)
+
+    def test_reduction_fusion_consistency_kat(self, ra):
+        # reference TestStencil :47-56: the same sum computed four ways
+        # must agree exactly
+        A = ra.arange(4000) / 250.0
+        s1 = float(A.sum())
+        B = A * 1.0
+        s2 = float(B.sum())
+        C = ra.sin(A)
+        s3 = float((C * C + ra.cos(A) ** 2).sum())
+        D = ra.sin(A) ** 2 + ra.cos(A) ** 2
+        ra.sync()
+        s4 = float(D.sum())
+        assert s1 == s2
+        assert abs(s3 - s4) < 1e-9
