@@ -865,10 +865,17 @@ class KernelGen:
                        and any(c.krow for c in self.class_list)) else 1
         self.yblock = yblock
         ind = 2
-        if nd >= 3:
+        if nd == 3:
             L.append(f"{' '*ind}for (i64 i0 = blockIdx.z; i0 < a.n0; "
                      f"i0 += gridDim.z) {{")
             ind += 2
+        elif nd == 4:
+            # leading axes collapsed onto blockIdx.z
+            L.append(f"{' '*ind}for (i64 zz = blockIdx.z; zz < a.n0 * a.n1;"
+                     f" zz += gridDim.z) {{")
+            ind += 2
+            L.append(f"{' '*ind}const i64 i0 = zz / a.n1;")
+            L.append(f"{' '*ind}const i64 i1 = zz % a.n1;")
         if nd >= 2 and yblock == 1:
             oy = nd - 2
             L.append(f"{' '*ind}for (i64 i{oy} = blockIdx.y; i{oy} < a.n{oy};"
@@ -1060,8 +1067,9 @@ def decide_vec(plan):
 def generate(plan):
     """plan -> GeneratedKernel (source + packing recipe)."""
     nd = len(plan.itershape)
-    if nd < 1 or nd > 3:
-        raise NotImplementedError(f"{nd}-d iteration spaces")
+    if nd < 1 or nd > 4:
+        raise NotImplementedError(f"{nd}-d iteration spaces (shardview is "
+                                  "<=4-D, SURVEY §8 a8)")
     vec, classes, anchor, key = plan_structure(plan)
     gen = KernelGen(plan, classes, vec)
     source, kmain, kfinish, finish_source, fields = gen.generate(key)

@@ -338,7 +338,12 @@ class HipBackend:
         yb = getattr(gk, "yblock", 1)
         gy = max(1, min(8192, (shape[nd - 2] + yb - 1) // yb)) \
             if nd >= 2 else 1
-        gz = max(1, min(64, shape[0])) if nd >= 3 else 1
+        if nd == 3:
+            gz = max(1, min(64, shape[0]))
+        elif nd >= 4:
+            gz = max(1, min(64, shape[0] * shape[1]))
+        else:
+            gz = 1
         if gk.nred:
             # bound the partials array: the finish kernel is one block
             gy = max(1, min(gy, 8192 // gx))

@@ -159,3 +159,12 @@ def test_axis_reduce_kernels_compile():
             nd, axes, np.dtype(idt), np.dtype(odt), kind, chunked=True)
         rc = lib.rt_compile_check(src.encode())
         assert rc == 0, lib.rt_last_error().decode() + "\n" + src
+
+
+def test_4d_kernel_compiles(rac):
+    def impl(np_):
+        a = np_.fromfunction(
+            lambda w, x, y, z: w + x * 2 + y * 3 + z, (3, 4, 5, 8),
+            dtype=np.int64)
+        return a * 3 + a[:, :, :, ::-1]
+    run_both(impl, rac)

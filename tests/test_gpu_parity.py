@@ -338,3 +338,13 @@ def test_broadcast_matmul_gpu(ra_gpu):
     Ae = np.fromfunction(lambda i, j: (i * k + j) % 31 + 1.0, (n, k))
     Be = np.fromfunction(lambda i, j: (i * m + j) % 17 - 3.0, (k, m))
     np.testing.assert_allclose(r, Ae @ Be, rtol=1e-12)
+
+
+def test_4d_gpu(ra_gpu):
+    def impl(np_):
+        a = np_.fromfunction(
+            lambda w, x, y, z: (w * 31 + x * 7 + y * 3 + z) % 101,
+            (6, 7, 33, 40), dtype=np.int64)
+        b = a + a[:, :, :, ::-1] * 2
+        return np.array([int(b.sum()), int(b.max())])
+    run_both(impl, ra_gpu)

@@ -649,3 +649,21 @@ class TestNewaxisMatmul:
         s4 = float(D.sum())
         assert s1 == s2
         assert abs(s3 - s4) < 1e-9
+
+
+class Test4D:
+    def test_4d_elementwise(self, ra):
+        def impl(np_):
+            a = np_.fromfunction(
+                lambda w, x, y, z: w * 1000 + x * 100 + y * 10 + z,
+                (5, 6, 7, 8), dtype=np.int64)
+            return a * 2 + a[:, :, :, ::-1]
+        run_both(impl, ra)
+
+    def test_4d_slice_reduce(self, ra):
+        def impl(np_):
+            a = np_.fromfunction(
+                lambda w, x, y, z: (w + x * 2 + y * 3 + z * 5) % 11,
+                (4, 5, 6, 7), dtype=np.int64)
+            return np.array([int(a[1:3, :, 2:5, ::2].sum())])
+        run_both(impl, ra)
