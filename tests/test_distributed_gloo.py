@@ -163,6 +163,26 @@ def test_broadcast_2d_spmd():
     """, world=2)
 
 
+@pytest.mark.parametrize("world", [3, 8])
+def test_odd_and_full_world_spmd(world):
+    """world sizes the 8-GPU node will see (8) and an odd split (3)."""
+    run_spmd("""
+        A = np_.arange(100001) / 1000.0
+        D = np_.sin(A) ** 2 + np_.cos(A) ** 2
+        s = D.sum()
+        B = np_.zeros(100001)
+        B[2:-2] = 0.25 * (A[:-4] + A[1:-3] + A[3:-1] + A[4:])
+        c = np_.arange(10007).cumsum()
+        a2 = np_.fromfunction(lambda x, y: x * 53 + y, (53, 71),
+                              dtype=np.int64).sum(axis=0)
+        if np_ is np:
+            return np.concatenate([[s], B, c * 1.0, a2 * 1.0])
+        import numpy as _np
+        return _np.concatenate([[float(s)], B.asarray(),
+                                c.asarray() * 1.0, a2.asarray() * 1.0])
+    """, world=world, tol=1e-12)
+
+
 @pytest.mark.parametrize("world", [2, 4])
 def test_axis_reduction_spmd(world):
     run_spmd("""
