@@ -43,11 +43,15 @@ class NumpyBackend:
     # -- memory -------------------------------------------------------------
 
     def alloc_container(self, bd, rt):
+        from ramba_amd.common import debug_poison
         _, cshape, _, _ = rt.shard_geometry(bd)
         if cshape is None:
             self.containers[bd.gid] = None
             return
-        self.containers[bd.gid] = np.empty(cshape, dtype=bd.dtype)
+        c = np.empty(cshape, dtype=bd.dtype)
+        if debug_poison:
+            c.view(np.uint8)[...] = 0xCC   # NaN-ish for floats, junk ints
+        self.containers[bd.gid] = c
 
     def free_container(self, bd):
         self.containers.pop(bd.gid, None)

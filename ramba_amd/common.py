@@ -23,6 +23,10 @@ import numpy as np
 ndebug = int(os.environ.get("RAMBA_DEBUG", "0"))
 default_border = int(os.environ.get("RAMBA_BORDER", "4"))
 ntiming = int(os.environ.get("RAMBA_TIMING", "0"))
+# debug sanitizer: poison fresh shard containers (NaN / 0xCC) so any read
+# of an unfilled border or uninitialised cell corrupts results visibly —
+# the analog of the reference build plan's hipMemsetD poisoning
+debug_poison = int(os.environ.get("RAMBA_DEBUG_POISON", "0"))
 
 
 def dprint(level, *args):

@@ -147,12 +147,16 @@ class HipBackend:
         return TORCH_DTYPE[np.dtype(dtype)]
 
     def alloc_container(self, bd, rt):
+        from .common import debug_poison
         _, cshape, _, _ = rt.shard_geometry(bd)
         if cshape is None:
             self.containers[bd.gid] = None
             return
-        self.containers[bd.gid] = self.torch.empty(
-            cshape, dtype=self._tdt(bd.dtype), device="cuda")
+        t = self.torch.empty(cshape, dtype=self._tdt(bd.dtype),
+                             device="cuda")
+        if debug_poison:
+            t.view(self.torch.uint8).fill_(0xCC)
+        self.containers[bd.gid] = t
 
     def free_container(self, bd):
         self.containers.pop(bd.gid, None)

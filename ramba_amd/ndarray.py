@@ -259,6 +259,17 @@ class ndarray:
                 return getattr(self, red)()
         return NotImplemented
 
+    # np.where / np.sum / … on ramba arrays: dispatch through the deferred
+    # path instead of silently gathering (reference __array_function__,
+    # ramba/ramba.py:6825).
+    _ARRAY_FUNC = {}   # filled after the class body (needs module fns)
+
+    def __array_function__(self, func, types, args, kwargs):
+        handler = self._ARRAY_FUNC.get(func.__name__)
+        if handler is None:
+            return NotImplemented
+        return handler(*args, **kwargs)
+
     # -- conversion ----------------------------------------------------------
 
     def __float__(self):
@@ -817,3 +828,35 @@ pi = np.pi
 e = np.e
 inf = np.inf
 nan = np.nan
+
+
+# __array_function__ dispatch table: real-numpy module calls on ramba
+# arrays route to the deferred implementations above.
+def _af_reduce(meth):
+    def h(a, axis=None, dtype=None, out=None, keepdims=False, **kw):
+        if out is not None:
+            raise NotImplementedError("out= is not supported")
+        if meth in ("min", "max", "any", "all"):
+            return getattr(a, meth)(axis=axis, keepdims=keepdims)
+        return getattr(a, meth)(axis=axis, dtype=dtype, keepdims=keepdims)
+    return h
+
+
+ndarray._ARRAY_FUNC.update({
+    "where": lambda c, x=None, y=None: where(c, x, y),
+    "sum": _af_reduce("sum"),
+    "prod": _af_reduce("prod"),
+    "amin": _af_reduce("min"), "min": _af_reduce("min"),
+    "amax": _af_reduce("max"), "max": _af_reduce("max"),
+    "any": _af_reduce("any"), "all": _af_reduce("all"),
+    "mean": lambda a, axis=None, **kw: a.mean(
+        axis=axis, keepdims=kw.get("keepdims", False)),
+    "cumsum": lambda a, axis=None, dtype=None, **kw: a.cumsum(
+        axis=axis, dtype=dtype),
+    "clip": lambda a, lo=None, hi=None, **kw: a.clip(lo, hi),
+    "transpose": lambda a, axes=None: transpose(a, axes),
+    "broadcast_to": lambda a, shape, **kw: broadcast_to(a, shape),
+    "shape": lambda a: a.shape,
+    "ndim": lambda a: a.ndim,
+    "size": lambda a: a.size,
+})

@@ -409,6 +409,10 @@ class Runtime:
                     str(tgt[1]))
         msgs = sorted(msgs, key=key)
         tc0 = time.perf_counter()
+        nbytes = sum(box_shape(m[3]) and
+                     int(np.prod(box_shape(m[3]))) * m[2].dtype.itemsize
+                     for m in msgs)
+        add_time("exchange_bytes", nbytes)
         sends, recvs = [], []
         for m in msgs:
             dst, src, bd, bx, tgt = m

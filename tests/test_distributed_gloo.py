@@ -220,3 +220,34 @@ def test_tiny_array_uni_dist():
         z = np_.arange(0)
         return b
     """, world=4)
+
+
+def test_poison_sanitizer_halo_correct():
+    """With RAMBA_DEBUG_POISON=1 every fresh container is junk-filled;
+    correct results prove no unfilled border/cell leaks into outputs
+    (the hipMemsetD-poisoning analog)."""
+    old = os.environ.get("RAMBA_DEBUG_POISON")
+    os.environ["RAMBA_DEBUG_POISON"] = "1"
+    try:
+        run_spmd("""
+            A = np_.arange(2001) * 1.0
+            B = np_.zeros(2001)
+            B[2:-2] = (0.1 * A[:-4] + 0.2 * A[1:-3] + 0.4 * A[2:-2]
+                       + 0.2 * A[3:-1] + 0.1 * A[4:])
+            s = B.sum()
+            c = np_.fromfunction(lambda x, y: x + y, (33, 35),
+                                 dtype=np.float32)
+            d = np_.zeros((33, 35), dtype=np.float32)
+            d[1:-1, 1:-1] = (c[:-2, 1:-1] + c[2:, 1:-1] + c[1:-1, :-2]
+                             + c[1:-1, 2:])
+            if np_ is np:
+                return np.concatenate([[s], B, d.reshape(-1)])
+            import numpy as _np
+            return _np.concatenate([[float(s)], B.asarray(),
+                                    d.asarray().reshape(-1)])
+        """, world=2, tol=1e-12)
+    finally:
+        if old is None:
+            os.environ.pop("RAMBA_DEBUG_POISON", None)
+        else:
+            os.environ["RAMBA_DEBUG_POISON"] = old

@@ -667,3 +667,39 @@ class Test4D:
                 (4, 5, 6, 7), dtype=np.int64)
             return np.array([int(a[1:3, :, 2:5, ::2].sum())])
         run_both(impl, ra)
+
+
+class TestArrayFunctionProtocol:
+    """Real-numpy module calls on ramba arrays stay deferred/distributed
+    (__array_function__; reference ramba/ramba.py:6825)."""
+
+    def test_np_where(self, ra):
+        a = ra.arange(100) * 1.0
+        r = np.where(a > 50.0, a, -a)
+        assert type(r).__module__.startswith("ramba_amd")
+        i = np.arange(100) * 1.0
+        np.testing.assert_allclose(r.asarray(), np.where(i > 50.0, i, -i))
+
+    def test_np_reductions(self, ra):
+        a = ra.fromfunction(lambda x, y: x * 11 + y, (13, 7),
+                            dtype=np.float64)
+        n = np.fromfunction(lambda x, y: x * 11 + y, (13, 7))
+        assert abs(float(np.sum(a)) - n.sum()) < 1e-9
+        assert float(np.max(a)) == n.max()
+        assert float(np.min(a)) == n.min()
+        np.testing.assert_allclose(np.sum(a, axis=0).asarray(),
+                                   n.sum(axis=0))
+        np.testing.assert_allclose(np.mean(a, axis=1).asarray(),
+                                   n.mean(axis=1))
+
+    def test_np_cumsum_clip_transpose(self, ra):
+        a = ra.arange(64) * 1.0
+        n = np.arange(64) * 1.0
+        np.testing.assert_allclose(np.cumsum(a).asarray(), n.cumsum())
+        np.testing.assert_allclose(np.clip(a, 5.0, 40.0).asarray(),
+                                   n.clip(5.0, 40.0))
+        b = ra.fromfunction(lambda x, y: x * 9 + y, (8, 9))
+        np.testing.assert_allclose(
+            np.transpose(b).asarray(),
+            np.fromfunction(lambda x, y: x * 9 + y, (8, 9)).T)
+        assert np.shape(b) == (8, 9) and np.ndim(b) == 2
