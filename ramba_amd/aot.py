@@ -125,8 +125,11 @@ class AotCompileBackend:
         pass
 
 
-def seed():
-    """Compile the kernels the bench/smoke programs need."""
+def seed(fuzz_seeds=None):
+    """Compile the kernels the bench/smoke programs need; optionally also
+    every kernel the fuzz-parity sweep (tests/fuzz_programs.py) generates,
+    so `pytest -m gpu` on a box loads code objects instead of invoking
+    hiprtc (`RAMBA_AOT_FUZZ` seeds, default 240; 0 disables)."""
     import ramba_amd as ra
     if ra._initialized["done"]:
         ra.shutdown()
@@ -155,6 +158,18 @@ def seed():
                      + ss[1:-1, 2:] - 4.0 * ss[1:-1, 1:-1])
     Z.sum()
     ra.sync()
+    if fuzz_seeds is None:
+        fuzz_seeds = int(os.environ.get("RAMBA_AOT_FUZZ", "240"))
+    if fuzz_seeds:
+        import sys
+        sys.path.insert(0, os.path.join(
+            os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+            "tests"))
+        from fuzz_programs import build_program
+        with np.errstate(all="ignore"):
+            for sd in range(fuzz_seeds):
+                impl, _ = build_program(sd)
+                impl(ra)
     n = be.compiled
     ra.shutdown()
     return n

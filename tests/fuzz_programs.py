@@ -113,7 +113,8 @@ def build_program(seed):
         for _ in range(nsteps):
             action = rng.choice(
                 ["un", "bin", "view", "setitem", "mask", "reduce",
-                 "axred", "cumsum", "astype"])
+                 "axred", "cumsum", "astype", "where", "clip", "mean",
+                 "outer"])
             i = int(rng.integers(0, len(pool)))
             val, kind = pool[i]
             if action == "un":
@@ -191,6 +192,27 @@ def build_program(seed):
                 dt = np.float32 if kind == "i" else np.int64
                 pool.append((val.astype(dt), "g" if dt == np.float32
                              else "i"))
+            elif action == "where":
+                c = float(rng.uniform(-5.0, 50.0))
+                r = np_.where(val > c, val, -val)
+                pool.append((r, kind))
+            elif action == "clip":
+                lo = float(rng.uniform(-10.0, 0.0))
+                hi = float(rng.uniform(1.0, 100.0))
+                if kind == "i":
+                    r = val.clip(int(lo), int(hi))
+                else:
+                    r = val.clip(lo, hi)
+                pool.append((r, kind))
+            elif action == "mean":
+                if kind != "i":
+                    scalars.append(float(val.mean()))
+                else:
+                    scalars.append(float(val.sum()))
+            elif action == "outer":
+                if val.ndim == 1 and val.shape[0] <= 200:
+                    r = val[:, None] * (val[None, :] + 1)
+                    pool.append((r, kind))
 
         # result: flattened concat of the last few pool values + scalars
         outs = []
