@@ -95,6 +95,18 @@ int rt_cumsum_scan(uintptr_t stream, const void *in, int64_t in_off,
                    void *agg, void *inc, void *flag, void *ticket,
                    double fbase, int64_t ibase, int dtype);
 
+/* Ordered boolean-mask compaction over the rank's local core box, C
+ * iteration order (reference: the compressing boolean getitem of
+ * ramba.ndarray, ramba/ramba.py maskarray path).  phase 1: per-4096-chunk
+ * selected counts into bcounts (int64[nchunks]); phase 2 is
+ * rt_cumsum(phase=2, dtype=2) on bcounts (exclusive scan + total);
+ * phase 3: write selected elements of `a` densely into `out`.
+ * dtype: 0=f64 1=f32 2=i64 3=i32 4=i16 5=i8 6=u8. */
+int rt_mask_compact(uintptr_t stream, const void *a, const void *m,
+                    void *out, int nd, const int64_t *shape,
+                    const int64_t *a_strides, const int64_t *m_strides,
+                    void *bcounts, int64_t nchunks, int dtype, int phase);
+
 int rt_stream_sync(uintptr_t stream);
 int rt_device_sync(void);
 

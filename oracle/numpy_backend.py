@@ -238,3 +238,29 @@ class NumpyBackend:
         outs = [torch.empty_like(t) for _ in range(self.rt.world)]
         dist.all_gather(outs, t)
         return [o.numpy()[0] for o in outs]
+
+
+def _nb_mask_compact(self, bd_a, bd_m, rt):
+    d, _, _, pads_a = rt.shard_geometry(bd_a)
+    if d is None:
+        return None, 0
+    _, _, _, pads_m = rt.shard_geometry(bd_m)
+    nd = len(bd_a.shape)
+    sl_a = tuple(slice(pads_a[i], pads_a[i] + int(d[1, i] - d[0, i] + 1))
+                 for i in range(nd))
+    sl_m = tuple(slice(pads_m[i], pads_m[i] + int(d[1, i] - d[0, i] + 1))
+                 for i in range(nd))
+    a = self.containers[bd_a.gid][sl_a]
+    m = self.containers[bd_m.gid][sl_m]
+    sel = a[m != 0]          # C order
+    return sel.copy(), int(sel.size)
+
+
+def _nb_write_local_dense(self, out_bd, rt, local):
+    _, _, _, pads = rt.shard_geometry(out_bd)
+    cont = self.containers[out_bd.gid]
+    cont[pads[0]:pads[0] + local.shape[0]] = local
+
+
+NumpyBackend.mask_compact = _nb_mask_compact
+NumpyBackend.write_local_dense = _nb_write_local_dense

@@ -108,6 +108,12 @@ class AotCompileBackend:
     def allgather_scalars(self, val, dtype):
         return [val]
 
+    def mask_compact(self, bd_a, bd_m, rt):
+        return None, 0   # kernels live in libramba_rt.so (hipcc-built)
+
+    def write_local_dense(self, out_bd, rt, local):
+        pass
+
     def box_to_numpy(self, bd, rt, box):
         from .shardview import box_shape
         return np.zeros(box_shape(box), dtype=bd.dtype)

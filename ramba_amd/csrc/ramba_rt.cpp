@@ -841,3 +841,173 @@ extern "C" int rt_cumsum(uintptr_t stream, const void *in, int64_t in_off,
             return 1;
     }
 }
+
+// ---------------------------------------------------------------------------
+// ordered boolean-mask compaction (SURVEY §8f n3 second half; the
+// reference's compressing boolean getitem `a[mask]`, ramba.py maskarray
+// getitem path).  Three phases over the rank's local box, C iteration
+// order: (1) per-chunk selected counts, (2) exclusive scan of the chunk
+// counts (reuses rt_cumsum phase 2 with dtype i64), (3) ordered write of
+// the selected elements into a dense output at chunk_base + intra-chunk
+// exclusive prefix.  Cross-rank ordering is the host runtime's job
+// (allgather of local counts -> uneven result divisions).
+// ---------------------------------------------------------------------------
+
+namespace {
+
+struct MCArgs {
+    int64_t n;
+    int64_t shape[4];
+    int64_t astr[4];   // element strides of a over the box
+    int64_t mstr[4];   // element strides of the mask over the box
+    int nd;
+};
+
+__device__ __forceinline__ int64_t mc_addr(const MCArgs &g, int64_t i,
+                                           const int64_t *str) {
+    int64_t rem = i, off = 0;
+    for (int d = g.nd - 1; d >= 0; --d) {
+        int64_t idx = rem % g.shape[d];
+        rem /= g.shape[d];
+        off += idx * str[d];
+    }
+    return off;
+}
+
+__global__ __launch_bounds__(256) void mask_count_k(
+    const uint8_t *__restrict__ m, MCArgs g, int64_t *__restrict__ bcounts) {
+    int64_t blk = blockIdx.x;
+    int64_t nblocks = gridDim.x;
+    __shared__ int64_t lds[4];
+    for (; blk * SCAN_CHUNK < g.n; blk += nblocks) {
+        int64_t base = blk * SCAN_CHUNK;
+        int64_t acc = 0;
+        for (int j = 0; j < SCAN_ITEMS; ++j) {
+            int64_t i = base + j * SCAN_THREADS + threadIdx.x;
+            if (i < g.n && m[mc_addr(g, i, g.mstr)] != 0) ++acc;
+        }
+        for (int off = 32; off > 0; off >>= 1) acc += __shfl_down(acc, off, 64);
+        if ((threadIdx.x & 63) == 0) lds[threadIdx.x >> 6] = acc;
+        __syncthreads();
+        if (threadIdx.x == 0)
+            bcounts[blk] = lds[0] + lds[1] + lds[2] + lds[3];
+        __syncthreads();
+    }
+}
+
+template <typename T>
+__global__ __launch_bounds__(256) void mask_write_k(
+    const T *__restrict__ a, const uint8_t *__restrict__ m,
+    T *__restrict__ out, MCArgs g, const int64_t *__restrict__ bexcl) {
+    // thread t owns elements [t*ITEMS, t*ITEMS+ITEMS) of the chunk, so the
+    // intra-chunk exclusive scan of per-thread counts preserves C order
+    __shared__ int64_t wsum[4];
+    int64_t blk = blockIdx.x;
+    int64_t nblocks = gridDim.x;
+    for (; blk * SCAN_CHUNK < g.n; blk += nblocks) {
+        int64_t b0 = blk * SCAN_CHUNK;
+        int64_t l0 = b0 + (int64_t)threadIdx.x * SCAN_ITEMS;
+        unsigned char sel[SCAN_ITEMS];
+        int64_t s = 0;
+        for (int j = 0; j < SCAN_ITEMS; ++j) {
+            int64_t i = l0 + j;
+            sel[j] = (i < g.n) && m[mc_addr(g, i, g.mstr)] != 0;
+            s += sel[j];
+        }
+        int64_t x = s;
+        int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
+        for (int off = 1; off < 64; off <<= 1) {
+            int64_t y = __shfl_up(x, off, 64);
+            if (lane >= off) x += y;
+        }
+        if (lane == 63) wsum[wid] = x;
+        __syncthreads();
+        int64_t wbase = 0;
+        for (int w = 0; w < wid; ++w) wbase += wsum[w];
+        int64_t pos = bexcl[blk] + wbase + x - s;
+        for (int j = 0; j < SCAN_ITEMS; ++j)
+            if (sel[j]) out[pos++] = a[mc_addr(g, l0 + j, g.astr)];
+        __syncthreads();
+    }
+}
+
+template <typename T>
+int mask_compact_launch(uintptr_t stream, const void *a, const void *m,
+                        void *out, const MCArgs &g, void *bcounts,
+                        int64_t nchunks, int phase) {
+    hipStream_t st = reinterpret_cast<hipStream_t>(stream);
+    int64_t grid = nchunks < 2048 ? nchunks : 2048;
+    if (grid < 1) grid = 1;
+    if (phase == 1) {
+        hipLaunchKernelGGL(mask_count_k, dim3((unsigned)grid),
+                           dim3(SCAN_THREADS), 0, st,
+                           static_cast<const uint8_t *>(m), g,
+                           static_cast<int64_t *>(bcounts));
+    } else {
+        hipLaunchKernelGGL((mask_write_k<T>), dim3((unsigned)grid),
+                           dim3(SCAN_THREADS), 0, st,
+                           static_cast<const T *>(a),
+                           static_cast<const uint8_t *>(m),
+                           static_cast<T *>(out), g,
+                           static_cast<const int64_t *>(bcounts));
+    }
+    hipError_t e = hipGetLastError();
+    if (e != hipSuccess) {
+        set_error(std::string("mask_compact launch: ") + hipGetErrorString(e));
+        return 1;
+    }
+    return 0;
+}
+
+}  // namespace
+
+// phase 1: bcounts[c] = selected count of chunk c (nchunks =
+// ceil(n/4096) int64 slots).  Phase 2 is rt_cumsum(phase=2, dtype=2) on
+// bcounts.  Phase 3: ordered write into dense `out`.  a/m pointers are
+// pre-offset by the caller; strides/shape describe the local box.
+// dtype: 0=f64 1=f32 2=i64 3=i32 4=i16 5=i8 6=u8.
+extern "C" int rt_mask_compact(uintptr_t stream, const void *a,
+                               const void *m, void *out, int nd,
+                               const int64_t *shape,
+                               const int64_t *a_strides,
+                               const int64_t *m_strides, void *bcounts,
+                               int64_t nchunks, int dtype, int phase) {
+    if (nd < 1 || nd > 4) {
+        set_error("rt_mask_compact: nd out of range");
+        return 1;
+    }
+    MCArgs g;
+    g.nd = nd;
+    g.n = 1;
+    for (int d = 0; d < nd; ++d) {
+        g.shape[d] = shape[d];
+        g.astr[d] = a_strides[d];
+        g.mstr[d] = m_strides[d];
+        g.n *= shape[d];
+    }
+    for (int d = nd; d < 4; ++d) {
+        g.shape[d] = 1;
+        g.astr[d] = 0;
+        g.mstr[d] = 0;
+    }
+    if (g.n == 0) return 0;
+    switch (dtype) {
+        case 0: return mask_compact_launch<double>(stream, a, m, out, g,
+                                                   bcounts, nchunks, phase);
+        case 1: return mask_compact_launch<float>(stream, a, m, out, g,
+                                                  bcounts, nchunks, phase);
+        case 2: return mask_compact_launch<int64_t>(stream, a, m, out, g,
+                                                    bcounts, nchunks, phase);
+        case 3: return mask_compact_launch<int32_t>(stream, a, m, out, g,
+                                                    bcounts, nchunks, phase);
+        case 4: return mask_compact_launch<int16_t>(stream, a, m, out, g,
+                                                    bcounts, nchunks, phase);
+        case 5: return mask_compact_launch<int8_t>(stream, a, m, out, g,
+                                                   bcounts, nchunks, phase);
+        case 6: return mask_compact_launch<uint8_t>(stream, a, m, out, g,
+                                                    bcounts, nchunks, phase);
+        default:
+            set_error("rt_mask_compact: bad dtype");
+            return 1;
+    }
+}

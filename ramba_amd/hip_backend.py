@@ -68,6 +68,14 @@ def _load_lib():
                                    ctypes.c_void_p, ctypes.c_void_p,
                                    ctypes.c_void_p, ctypes.c_double,
                                    ctypes.c_int64, ctypes.c_int]
+    lib.rt_mask_compact.argtypes = [ctypes.c_size_t, ctypes.c_void_p,
+                                    ctypes.c_void_p, ctypes.c_void_p,
+                                    ctypes.c_int,
+                                    ctypes.POINTER(ctypes.c_int64),
+                                    ctypes.POINTER(ctypes.c_int64),
+                                    ctypes.POINTER(ctypes.c_int64),
+                                    ctypes.c_void_p, ctypes.c_int64,
+                                    ctypes.c_int, ctypes.c_int]
     lib.rt_stream_sync.argtypes = [ctypes.c_size_t]
     return lib
 
@@ -656,3 +664,67 @@ HipBackend._hb_cumsum_src = _hb_cumsum_src
 HipBackend.cumsum_local_phase12 = _hb_cumsum_local_phase12
 HipBackend.cumsum_local_phase3 = _hb_cumsum_local_phase3
 HipBackend.allgather_scalars = _hb_allgather_scalars
+
+
+# -- boolean-mask compaction (a[mask]) ------------------------
+
+_MC_DT = {"float64": 0, "float32": 1, "int64": 2, "int32": 3,
+          "int16": 4, "int8": 5, "uint8": 6, "bool": 6}
+
+
+def _hb_mask_compact(self, bd_a, bd_m, rt):
+    """Local ordered compaction of the rank's core box: returns
+    (dense device tensor of selected elements, count)."""
+    d, _, cstr_a, pads_a = rt.shard_geometry(bd_a)
+    if d is None:
+        return None, 0
+    _, _, cstr_m, pads_m = rt.shard_geometry(bd_m)
+    nd = len(bd_a.shape)
+    shape = [int(d[1, i] - d[0, i] + 1) for i in range(nd)]
+    n = 1
+    for sz in shape:
+        n *= sz
+    if n == 0:
+        return None, 0
+    ca, cm = self._cont(bd_a), self._cont(bd_m)
+    off_a = sum(pads_a[i] * cstr_a[i] for i in range(nd))
+    off_m = sum(pads_m[i] * cstr_m[i] for i in range(nd))
+    a_ptr = ca.data_ptr() + off_a * ca.element_size()
+    m_ptr = cm.data_ptr() + off_m * cm.element_size()
+    dtc = _MC_DT[str(np.dtype(bd_a.dtype))]
+    nchunks = max(1, (n + _SCAN_CHUNK - 1) // _SCAN_CHUNK)
+    bcounts = self.torch.empty(nchunks, dtype=self.torch.int64,
+                               device="cuda")
+    total = self.torch.empty(1, dtype=self.torch.int64, device="cuda")
+    st = self._stream()
+    rc = self.lib.rt_mask_compact(
+        st, ctypes.c_void_p(a_ptr), ctypes.c_void_p(m_ptr), None, nd,
+        _i64arr(shape), _i64arr(cstr_a), _i64arr(cstr_m),
+        ctypes.c_void_p(bcounts.data_ptr()), nchunks, dtc, 1)
+    self._check(rc, "rt_mask_compact(1)")
+    rc = self.lib.rt_cumsum(
+        st, None, 0, 0, 0, None, 0, ctypes.c_void_p(bcounts.data_ptr()),
+        nchunks, ctypes.c_void_p(total.data_ptr()), 0.0, 0, 2, 2)
+    self._check(rc, "rt_mask_compact(2)")
+    count = int(total.cpu().numpy()[0])   # D2H syncs the stream
+    if count == 0:
+        return None, 0
+    out = self.torch.empty(count, dtype=ca.dtype, device="cuda")
+    rc = self.lib.rt_mask_compact(
+        st, ctypes.c_void_p(a_ptr), ctypes.c_void_p(m_ptr),
+        ctypes.c_void_p(out.data_ptr()), nd, _i64arr(shape),
+        _i64arr(cstr_a), _i64arr(cstr_m),
+        ctypes.c_void_p(bcounts.data_ptr()), nchunks, dtc, 3)
+    self._check(rc, "rt_mask_compact(3)")
+    return out, count
+
+
+def _hb_write_local_dense(self, out_bd, rt, local):
+    """Write a dense 1-D local tensor into the rank's container core."""
+    _, _, _, pads = rt.shard_geometry(out_bd)
+    cont = self._cont(out_bd)
+    cont[pads[0]:pads[0] + local.shape[0]] = local
+
+
+HipBackend.mask_compact = _hb_mask_compact
+HipBackend.write_local_dense = _hb_write_local_dense

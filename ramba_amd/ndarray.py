@@ -139,6 +139,24 @@ class ndarray:
     # -- indexing ------------------------------------------------------------
 
     def __getitem__(self, index):
+        # compressing boolean getitem: a[mask] -> 1-D array of the selected
+        # elements in C order (reference maskarray getitem; dynamic-shape
+        # sync point).  Value and mask are first copied onto one fresh
+        # common partition through the fused engine, then each rank
+        # compacts its core box (runtime.mask_compact_op).
+        if isinstance(index, np.ndarray) and index.dtype == np.bool_:
+            index = fromarray(index)
+        if isinstance(index, ndarray) and index.dtype == np.bool_:
+            if index.shape != self.shape:
+                raise IndexError(
+                    f"boolean index shape {index.shape} does not match "
+                    f"array shape {self.shape}")
+            ta = self.copy()
+            tm = index.astype(np.uint8)
+            deferred.flush()
+            rt = deferred.get_runtime()
+            out_bd = rt.mask_compact_op(ta.bdarray, tm.bdarray)
+            return ndarray(out_bd, View.identity(out_bd.shape))
         if not isinstance(index, tuple):
             index = (index,)
         if all(isinstance(i, (int, np.integer)) for i in index) \

@@ -656,3 +656,60 @@ def cumsum_op(self, arr, out_dtype):
 
 
 Runtime.cumsum_op = cumsum_op
+
+# ---------------------------------------------------------------------------
+# boolean-mask compaction getitem (SURVEY §8f n3 second half; the
+# reference's compressing `a[mask]`, ramba.py maskarray getitem).  The
+# frontend first materialises value and mask onto ONE common partition
+# (fresh default-partition copies through the fused engine), so here both
+# arrays share divisions; each rank compacts its core box in C order and
+# the result is partitioned unevenly by the per-rank counts — no data
+# exchange, just an allgather of counts.
+# ---------------------------------------------------------------------------
+
+def mask_compact_op(self, bd_a, bd_m):
+    assert np.array_equal(bd_a.divisions, bd_m.divisions), \
+        "mask_compact_op requires co-partitioned value/mask"
+    nd = len(bd_a.shape)
+    order = []
+    for r in range(self.world):
+        cb = self.core_box(bd_a, r)
+        if cb is None:
+            continue
+        # C-order concatenation across ranks needs each rank's box to
+        # span the full extent of every axis but the first
+        for dax in range(1, nd):
+            if not (cb[0, dax] == 0
+                    and cb[1, dax] == bd_a.shape[dax] - 1):
+                raise NotImplementedError(
+                    "a[mask] needs C-contiguous (axis-0 split) partitions "
+                    f"for nd>={nd}; got division {cb.tolist()} of rank {r}")
+        order.append((int(cb[0, 0]), r))
+    order.sort()
+
+    local, count = self.backend.mask_compact(bd_a, bd_m, self)
+    counts = self.backend.allgather_scalars(np.int64(count), np.int64)
+    prefix, run = {}, 0
+    for (_, r) in order:
+        prefix[r] = run
+        run += int(counts[r])
+    total = run
+    divs = np.zeros((self.world, 2, 1), dtype=np.int64)
+    for r in range(self.world):
+        c = int(counts[r])
+        if c == 0:
+            divs[r, 1, 0] = -1
+        else:
+            divs[r, 0, 0] = prefix[r]
+            divs[r, 1, 0] = prefix[r] + c - 1
+    out_bd = deferred.bdarray((total,), bd_a.dtype, divs, default_border,
+                              flex=False)
+    self.backend.alloc_container(out_bd, self)
+    out_bd.constructed = True
+    if count:
+        self.backend.write_local_dense(out_bd, self, local)
+    self.backend.free_temps()
+    return out_bd
+
+
+Runtime.mask_compact_op = mask_compact_op

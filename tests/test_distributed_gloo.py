@@ -251,3 +251,23 @@ def test_poison_sanitizer_halo_correct():
             os.environ.pop("RAMBA_DEBUG_POISON", None)
         else:
             os.environ["RAMBA_DEBUG_POISON"] = old
+
+
+def test_mask_getitem_spmd():
+    """Compressing a[mask] across ranks: uneven result divisions ordered
+    by core-box origin; no data-path exchange (runtime.mask_compact_op)."""
+    for world in (2, 3):
+        run_spmd("""
+            a = np_.arange(1000) * 1.0
+            b = np_.sin(a * 0.01)
+            sel = b[b > 0.3]
+            t = np_.fromfunction(lambda x, y: x * 3 + y, (401, 3))
+            sel2 = t[(t % 7.0) == 0.0]
+            if np_ is np:
+                return np.concatenate([sel, sel2,
+                                       [float(sel.size), float(sel2.size)]])
+            import numpy as _np
+            return _np.concatenate([sel.asarray(), sel2.asarray(),
+                                    [float(sel.shape[0]),
+                                     float(sel2.shape[0])]])
+        """, world=world, tol=1e-12)

@@ -348,3 +348,34 @@ def test_4d_gpu(ra_gpu):
         b = a + a[:, :, :, ::-1] * 2
         return np.array([int(b.sum()), int(b.max())])
     run_both(impl, ra_gpu)
+
+
+def test_mask_getitem_gpu(ra_gpu):
+    """a[mask] compaction through rt_mask_compact (ordered 3-phase)."""
+    def impl(np_):
+        a = np_.arange(1_000_000) * 1.0
+        b = np_.sin(a * 1e-5)
+        sel = b[b > 0.7]
+        t = np_.fromfunction(lambda x, y: x * 31 + y, (300, 31))
+        sel2 = t[(t % 13.0) == 0.0]
+        e = a[a < -1.0]
+        if np_ is np:
+            return np.concatenate([sel, sel2,
+                                   [float(sel.size), float(sel2.size),
+                                    float(e.size)]])
+        import numpy as _np
+        return _np.concatenate([sel.asarray(), sel2.asarray(),
+                                [float(sel.shape[0]), float(sel2.shape[0]),
+                                 float(e.shape[0])]])
+    run_both(impl, ra_gpu)
+
+
+def test_mask_getitem_gpu_int(ra_gpu):
+    def impl(np_):
+        d = np_.arange(100_000)
+        r = d[(d % 97) == 0]
+        if np_ is np:
+            return np.asarray(r, dtype=np.float64)
+        import numpy as _np
+        return _np.asarray(r.asarray(), dtype=np.float64)
+    run_both(impl, ra_gpu)

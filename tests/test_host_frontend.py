@@ -703,3 +703,55 @@ class TestArrayFunctionProtocol:
             np.transpose(b).asarray(),
             np.fromfunction(lambda x, y: x * 9 + y, (8, 9)).T)
         assert np.shape(b) == (8, 9) and np.ndim(b) == 2
+
+
+class TestMaskGetitem:
+    """Compressing boolean getitem a[mask] (reference maskarray getitem;
+    runtime.mask_compact_op)."""
+
+    def test_1d_basic(self, ra):
+        a = ra.arange(100) * 1.0
+        i = np.arange(100) * 1.0
+        np.testing.assert_allclose(a[(a % 7.0) == 0.0].asarray(),
+                                   i[(i % 7) == 0])
+
+    def test_deferred_mask_and_values(self, ra):
+        a = ra.arange(500) * 1.0
+        b = ra.sin(a * 0.01)
+        i = np.arange(500) * 1.0
+        nb = np.sin(i * 0.01)
+        np.testing.assert_allclose(b[b > 0.5].asarray(), nb[nb > 0.5],
+                                   rtol=1e-12, atol=1e-12)
+
+    def test_empty_and_full(self, ra):
+        a = ra.arange(64) * 1.0
+        i = np.arange(64) * 1.0
+        assert a[a < -1.0].shape == (0,)
+        np.testing.assert_allclose(a[a > -1.0].asarray(), i)
+
+    def test_2d(self, ra):
+        c = ra.fromfunction(lambda x, y: x * 13 + y, (17, 13))
+        nc = np.fromfunction(lambda x, y: x * 13 + y, (17, 13))
+        np.testing.assert_allclose(c[(c % 5.0) == 0.0].asarray(),
+                                   nc[(nc % 5) == 0])
+
+    def test_numpy_mask_int_dtype(self, ra):
+        d = ra.arange(50)
+        nd_ = np.arange(50)
+        nm = (nd_ % 9) == 0
+        np.testing.assert_array_equal(d[nm].asarray(), nd_[nm])
+        np.testing.assert_array_equal(d[(d % 3) == 0].asarray(),
+                                      nd_[nd_ % 3 == 0])
+
+    def test_shape_mismatch_raises(self, ra):
+        a = ra.arange(10) * 1.0
+        with pytest.raises(IndexError):
+            a[(ra.arange(9) * 1.0) > 4.0]
+
+    def test_view_source(self, ra):
+        a = ra.arange(200) * 1.0
+        v = a[20:180:2]
+        i = np.arange(200) * 1.0
+        vi = i[20:180:2]
+        np.testing.assert_allclose(v[(v % 4.0) == 0.0].asarray(),
+                                   vi[(vi % 4) == 0])
