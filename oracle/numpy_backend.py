@@ -264,3 +264,49 @@ def _nb_write_local_dense(self, out_bd, rt, local):
 
 NumpyBackend.mask_compact = _nb_mask_compact
 NumpyBackend.write_local_dense = _nb_write_local_dense
+
+
+# -- axis-wise cumsum (N-D scumulative) -----------------------
+
+
+def _nb_axis_scan_local(self, bd_in, off0, strides, lshape, axis, out_bd,
+                        out_off, out_strides):
+    base = self._cont(bd_in).reshape(-1)
+    isz = base.itemsize
+    view = np.lib.stride_tricks.as_strided(
+        base[off0:] if off0 else base, shape=tuple(lshape),
+        strides=tuple(s * isz for s in strides))
+    cs = np.cumsum(view.astype(out_bd.dtype, copy=False), axis=axis)
+    out = self.containers[out_bd.gid].reshape(-1)
+    osz = out.itemsize
+    oview = np.lib.stride_tricks.as_strided(
+        out[out_off:] if out_off else out, shape=tuple(lshape),
+        strides=tuple(s * osz for s in out_strides))
+    oview[...] = cs
+    sel = tuple(slice(None) if d != axis else -1
+                for d in range(len(lshape)))
+    self.temps["__axcs_tot__"] = np.ascontiguousarray(cs[sel])
+
+
+def _nb_axcs_init_offsets(self, lines_shape, dtype):
+    self.temps["__axcs_off__"] = np.zeros(
+        lines_shape if lines_shape else (1,), dtype=dtype)
+
+
+def _nb_axcs_accumulate(self, rel_box, buf):
+    offs = self.temps["__axcs_off__"]
+    sl = tuple(slice(int(rel_box[0, i]), int(rel_box[1, i]) + 1)
+               for i in range(offs.ndim))
+    offs[sl] += buf.numpy().reshape(offs[sl].shape)
+
+
+def _nb_axcs_apply(self, out_bd, rt, box, axis):
+    offs = self.temps["__axcs_off__"]
+    dst = self._cont(out_bd)[rt.container_slice(out_bd, box)]
+    dst += np.expand_dims(offs, axis)
+
+
+NumpyBackend.axis_scan_local = _nb_axis_scan_local
+NumpyBackend.axcs_init_offsets = _nb_axcs_init_offsets
+NumpyBackend.axcs_accumulate = _nb_axcs_accumulate
+NumpyBackend.axcs_apply = _nb_axcs_apply

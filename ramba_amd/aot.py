@@ -111,6 +111,18 @@ class AotCompileBackend:
     def mask_compact(self, bd_a, bd_m, rt):
         return None, 0   # kernels live in libramba_rt.so (hipcc-built)
 
+    def axis_scan_local(self, *a):
+        pass
+
+    def axcs_init_offsets(self, *a):
+        pass
+
+    def axcs_accumulate(self, *a):
+        pass
+
+    def axcs_apply(self, *a):
+        pass
+
     def write_local_dense(self, out_bd, rt, local):
         pass
 

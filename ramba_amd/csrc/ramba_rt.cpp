@@ -1011,3 +1011,160 @@ extern "C" int rt_mask_compact(uintptr_t stream, const void *a,
             return 1;
     }
 }
+
+// ---------------------------------------------------------------------------
+// axis-wise cumulative sum over the rank's local box (the
+// N-D/axis half of the reference's scumulative, ramba/ramba.py:10057-10171 +
+// scumulative_worker 3378-3440).  Local inclusive scan along `axis`, line
+// totals into a dense slab; the cross-rank fixup is the host runtime's job
+// (slab exchange + broadcast add via rt_combine_box with stride 0 on the
+// scan axis — replaces the reference's sequential worker relay chain).
+// Two mappings: axis != last -> one thread per line, serial walk, loads
+// coalesced across threads; axis == last -> one wave per line, shfl_up
+// segment scans with a carried prefix.
+// ---------------------------------------------------------------------------
+
+namespace {
+
+struct ASArgs {
+    int64_t len;        // shape[axis]
+    int64_t nlines;     // product of the other axes
+    int64_t lshape[3];  // line-index space (C order, axis removed)
+    int64_t istr[3];    // in strides over the line space
+    int64_t ostr[3];    // out strides over the line space
+    int64_t istr_ax, ostr_ax;
+    int nld;            // line-space ndim (nd - 1)
+};
+
+__device__ __forceinline__ void as_addr(const ASArgs &g, int64_t line,
+                                        int64_t &ioff, int64_t &ooff) {
+    int64_t rem = line;
+    ioff = 0;
+    ooff = 0;
+    for (int d = g.nld - 1; d >= 0; --d) {
+        int64_t idx = rem % g.lshape[d];
+        rem /= g.lshape[d];
+        ioff += idx * g.istr[d];
+        ooff += idx * g.ostr[d];
+    }
+}
+
+template <typename T>
+__global__ __launch_bounds__(256) void axis_scan_lines_k(
+    const T *__restrict__ in, T *__restrict__ out,
+    T *__restrict__ totals, ASArgs g) {
+    int64_t line = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; line < g.nlines; line += stride) {
+        int64_t ioff, ooff;
+        as_addr(g, line, ioff, ooff);
+        T acc = (T)0;
+        for (int64_t k = 0; k < g.len; ++k) {
+            acc += in[ioff + k * g.istr_ax];
+            out[ooff + k * g.ostr_ax] = acc;
+        }
+        totals[line] = acc;
+    }
+}
+
+template <typename T>
+__global__ __launch_bounds__(256) void axis_scan_waves_k(
+    const T *__restrict__ in, T *__restrict__ out,
+    T *__restrict__ totals, ASArgs g) {
+    int64_t wave = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+    int64_t nwaves = ((int64_t)gridDim.x * blockDim.x) >> 6;
+    int lane = threadIdx.x & 63;
+    for (int64_t line = wave; line < g.nlines; line += nwaves) {
+        int64_t ioff, ooff;
+        as_addr(g, line, ioff, ooff);
+        T carry = (T)0;
+        for (int64_t base = 0; base < g.len; base += 64) {
+            int64_t k = base + lane;
+            T x = k < g.len ? in[ioff + k * g.istr_ax] : (T)0;
+            for (int off = 1; off < 64; off <<= 1) {
+                T y = __shfl_up(x, off, 64);
+                if (lane >= off) x += y;
+            }
+            x += carry;
+            if (k < g.len) out[ooff + k * g.ostr_ax] = x;
+            carry = __shfl(x, 63, 64);
+        }
+        if (lane == 0) totals[line] = carry;
+    }
+}
+
+template <typename T>
+int axis_scan_launch(uintptr_t stream, const void *in, void *out,
+                     void *totals, const ASArgs &g, int waves) {
+    hipStream_t st = reinterpret_cast<hipStream_t>(stream);
+    int64_t work = waves ? g.nlines * 64 : g.nlines;
+    int64_t blocks = (work + 255) / 256;
+    if (blocks > 4096) blocks = 4096;
+    if (blocks < 1) blocks = 1;
+    if (waves)
+        hipLaunchKernelGGL((axis_scan_waves_k<T>), dim3((unsigned)blocks),
+                           dim3(256), 0, st, static_cast<const T *>(in),
+                           static_cast<T *>(out), static_cast<T *>(totals),
+                           g);
+    else
+        hipLaunchKernelGGL((axis_scan_lines_k<T>), dim3((unsigned)blocks),
+                           dim3(256), 0, st, static_cast<const T *>(in),
+                           static_cast<T *>(out), static_cast<T *>(totals),
+                           g);
+    hipError_t e = hipGetLastError();
+    if (e != hipSuccess) {
+        set_error(std::string("axis_scan launch: ") + hipGetErrorString(e));
+        return 1;
+    }
+    return 0;
+}
+
+}  // namespace
+
+// in/out pre-offset to the box origin; shape/strides over the FULL local
+// box (nd dims, nd 2..4); totals: dense C-order slab of the line space.
+// dtype: 0=f64 1=f32 2=i64 3=i32.
+extern "C" int rt_axis_scan(uintptr_t stream, const void *in, void *out,
+                            void *totals, int nd, const int64_t *shape,
+                            const int64_t *in_strides,
+                            const int64_t *out_strides, int axis,
+                            int dtype) {
+    if (nd < 2 || nd > 4 || axis < 0 || axis >= nd) {
+        set_error("rt_axis_scan: bad nd/axis");
+        return 1;
+    }
+    ASArgs g;
+    g.len = shape[axis];
+    g.istr_ax = in_strides[axis];
+    g.ostr_ax = out_strides[axis];
+    g.nld = 0;
+    g.nlines = 1;
+    for (int d = 0; d < nd; ++d) {
+        if (d == axis) continue;
+        g.lshape[g.nld] = shape[d];
+        g.istr[g.nld] = in_strides[d];
+        g.ostr[g.nld] = out_strides[d];
+        g.nlines *= shape[d];
+        ++g.nld;
+    }
+    for (int d = g.nld; d < 3; ++d) {
+        g.lshape[d] = 1;
+        g.istr[d] = 0;
+        g.ostr[d] = 0;
+    }
+    if (g.nlines == 0 || g.len == 0) return 0;
+    int waves = (axis == nd - 1) ? 1 : 0;
+    switch (dtype) {
+        case 0: return axis_scan_launch<double>(stream, in, out, totals, g,
+                                                waves);
+        case 1: return axis_scan_launch<float>(stream, in, out, totals, g,
+                                               waves);
+        case 2: return axis_scan_launch<int64_t>(stream, in, out, totals, g,
+                                                 waves);
+        case 3: return axis_scan_launch<int32_t>(stream, in, out, totals, g,
+                                                 waves);
+        default:
+            set_error("rt_axis_scan: bad dtype");
+            return 1;
+    }
+}

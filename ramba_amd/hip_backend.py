@@ -76,6 +76,13 @@ def _load_lib():
                                     ctypes.POINTER(ctypes.c_int64),
                                     ctypes.c_void_p, ctypes.c_int64,
                                     ctypes.c_int, ctypes.c_int]
+    lib.rt_axis_scan.argtypes = [ctypes.c_size_t, ctypes.c_void_p,
+                                 ctypes.c_void_p, ctypes.c_void_p,
+                                 ctypes.c_int,
+                                 ctypes.POINTER(ctypes.c_int64),
+                                 ctypes.POINTER(ctypes.c_int64),
+                                 ctypes.POINTER(ctypes.c_int64),
+                                 ctypes.c_int, ctypes.c_int]
     lib.rt_stream_sync.argtypes = [ctypes.c_size_t]
     return lib
 
@@ -728,3 +735,76 @@ def _hb_write_local_dense(self, out_bd, rt, local):
 
 HipBackend.mask_compact = _hb_mask_compact
 HipBackend.write_local_dense = _hb_write_local_dense
+
+
+# -- axis-wise cumsum (N-D scumulative) -----------------------
+
+
+def _hb_axis_scan_local(self, bd_in, off0, strides, lshape, axis, out_bd,
+                        out_off, out_strides):
+    """Local inclusive scan along `axis` over the rank's box; line totals
+    land in temps["__axcs_tot__"] (dense C-order line-space tensor)."""
+    dt = _CS_DT[str(np.dtype(out_bd.dtype))]
+    ca, co = self._cont(bd_in), self._cont(out_bd)
+    nd = len(lshape)
+    lines_shape = tuple(lshape[d] for d in range(nd) if d != axis)
+    tot = self.torch.empty(lines_shape, dtype=co.dtype, device="cuda")
+    self.temps["__axcs_tot__"] = tot
+    in_ptr = ca.data_ptr() + off0 * ca.element_size()
+    out_ptr = co.data_ptr() + out_off * co.element_size()
+    rc = self.lib.rt_axis_scan(
+        self._stream(), ctypes.c_void_p(in_ptr), ctypes.c_void_p(out_ptr),
+        ctypes.c_void_p(tot.data_ptr()), nd, _i64arr(lshape),
+        _i64arr(strides), _i64arr(out_strides), axis, dt)
+    self._check(rc, "rt_axis_scan")
+
+
+def _hb_axcs_init_offsets(self, lines_shape, dtype):
+    self._axcs_np_dtype = np.dtype(dtype)
+    self.temps["__axcs_off__"] = self.torch.zeros(
+        lines_shape if lines_shape else (1,), dtype=self._tdt(dtype),
+        device="cuda")
+
+
+def _hb_axcs_accumulate(self, rel_box, buf):
+    offs = self.temps["__axcs_off__"]
+    from .shardview import box_shape as _bs
+    shape = _bs(rel_box)
+    ts = offs.stride()
+    dst_off = sum(int(rel_box[0, i]) * ts[i] for i in range(len(shape)))
+    self._check(self.lib.rt_combine_box(
+        self._stream(), ctypes.c_void_p(offs.data_ptr()),
+        ctypes.c_void_p(buf.data_ptr()), len(shape), _i64arr(shape),
+        _i64arr(ts), _i64arr(buf.stride()), dst_off, 0,
+        _DT_ENUM[str(self._axcs_np_dtype)], _OP_ENUM["sum"]),
+        "axcs_accumulate")
+
+
+def _hb_axcs_apply(self, out_bd, rt, box, axis):
+    """out[box] += offsets, broadcast along the scan axis (src stride 0)."""
+    offs = self.temps["__axcs_off__"]
+    from .shardview import box_shape as _bs
+    cont = self._cont(out_bd)
+    dst_off, cs = self._box_off(out_bd, rt, box)
+    shape = _bs(box)
+    ostr = list(offs.stride())
+    src_strides = []
+    j = 0
+    for d in range(len(shape)):
+        if d == axis:
+            src_strides.append(0)
+        else:
+            src_strides.append(ostr[j])
+            j += 1
+    self._check(self.lib.rt_combine_box(
+        self._stream(), ctypes.c_void_p(cont.data_ptr()),
+        ctypes.c_void_p(offs.data_ptr()), len(shape), _i64arr(shape),
+        _i64arr(cs), _i64arr(src_strides), dst_off, 0,
+        _DT_ENUM[str(np.dtype(out_bd.dtype))], _OP_ENUM["sum"]),
+        "axcs_apply")
+
+
+HipBackend.axis_scan_local = _hb_axis_scan_local
+HipBackend.axcs_init_offsets = _hb_axcs_init_offsets
+HipBackend.axcs_accumulate = _hb_axcs_accumulate
+HipBackend.axcs_apply = _hb_axcs_apply

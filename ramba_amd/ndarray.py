@@ -452,16 +452,29 @@ class ndarray:
         return out
 
     def cumsum(self, axis=None, dtype=None, **kw):
-        """1-D cumulative sum (reference scumulative/cumsum,
-        ramba.py:10057-10171, 9675)."""
-        assert self.ndim == 1 and axis in (None, 0), \
-            "cumsum: 1-D only (reference scumulative)"
+        """Cumulative sum (reference scumulative/cumsum,
+        ramba.py:10057-10171, 9675: 1-D, or N-D along an explicit axis —
+        the reference's scumulative asserts a concrete axis for N-D)."""
         dt = self.dtype if dtype is None else np.dtype(dtype)
         if dtype is None and dt.kind in "bi" and dt.itemsize < 8:
             dt = np.dtype(np.int64)   # NumPy platform-int promotion
         deferred.flush()
         rt = deferred.get_runtime()
-        out_bd = rt.cumsum_op(self, dt)
+        if self.ndim == 1 and axis in (None, 0, -1):
+            out_bd = rt.cumsum_op(self, dt)
+            return ndarray(out_bd, View.identity(self.shape))
+        if axis is None:
+            raise ValueError(
+                "cumsum on an N-D array needs an explicit axis "
+                "(reference scumulative, ramba.py:10061)")
+        ax = int(axis)
+        if ax < 0:
+            ax += self.ndim
+        if not (0 <= ax < self.ndim):
+            raise ValueError(f"cumsum: axis {axis} out of range")
+        src = self if np.dtype(self.dtype) == dt else self.astype(dt)
+        deferred.flush()
+        out_bd = rt.cumsum_axis_op(src, ax, dt)
         return ndarray(out_bd, View.identity(self.shape))
 
     def mean(self, axis=None, keepdims=False, **kw):

@@ -713,3 +713,93 @@ def mask_compact_op(self, bd_a, bd_m):
 
 
 Runtime.mask_compact_op = mask_compact_op
+
+# ---------------------------------------------------------------------------
+# axis-wise cumsum on N-D arrays (the axis half of the
+# reference's scumulative, ramba.py:10057-10171 + scumulative_worker
+# 3378-3440).  Local inclusive scan along the axis per rank; the
+# reference's sequential cross-worker relay chain becomes ONE deterministic
+# pairwise slab exchange: every rank that precedes another along the scan
+# axis sends the overlap of its line-totals slab, receivers sum incoming
+# slabs and broadcast-add them over their local box (rt_combine_box with
+# stride 0 on the axis).
+# ---------------------------------------------------------------------------
+
+def cumsum_axis_op(self, arr, axis, out_dtype):
+    from .shardview import exec_boxes as _eb
+    bd, v = arr.bdarray, arr.view
+    nd = v.ndim
+    assert 0 <= axis < nd and nd >= 2
+    assert np.dtype(bd.dtype) == np.dtype(out_dtype)  # frontend casts
+    lbs = _eb(v, bd.divisions)
+    divs = np.zeros((self.world, 2, nd), dtype=np.int64)
+    for r, b in enumerate(lbs):
+        if b is None:
+            divs[r, 1, :] = -1
+        else:
+            divs[r] = b
+    out_bd = deferred.bdarray(v.shape, out_dtype, divs, default_border,
+                              flex=False)
+    self.backend.alloc_container(out_bd, self)
+    out_bd.constructed = True
+
+    lb = lbs[self.rank]
+    if lb is not None:
+        d_, _, cstrides, pads = self.shard_geometry(bd)
+        off0, strides = v.operand_addressing(lb[0], cstrides, d_[0], pads)
+        od, _, ocs, opads = self.shard_geometry(out_bd)
+        out_off = sum((int(lb[0, i]) - int(od[0, i]) + opads[i]) * ocs[i]
+                      for i in range(nd))
+        self.backend.axis_scan_local(bd, off0, strides, box_shape(lb),
+                                     axis, out_bd, out_off, ocs)
+
+    def _proj(b):
+        keep = [i for i in range(nd) if i != axis]
+        return np.array([[int(b[0, i]) for i in keep],
+                         [int(b[1, i]) for i in keep]], dtype=np.int64)
+
+    # deterministic pair plan on every rank
+    msgs = []
+    for s_ in range(self.world):          # receiver
+        bs = lbs[s_]
+        if bs is None:
+            continue
+        for r_ in range(self.world):      # sender (strictly precedes s_)
+            br = lbs[r_]
+            if br is None or r_ == s_:
+                continue
+            if int(br[1, axis]) < int(bs[0, axis]):
+                inter = box_intersect(_proj(br), _proj(bs))
+                if inter is not None:
+                    msgs.append((s_, r_, inter))
+    msgs.sort(key=lambda m: (m[0], m[1], tuple(m[2][0]), tuple(m[2][1])))
+
+    my_lines = _proj(lb) if lb is not None else None
+    sends, recvs = [], []
+    for (dst, src, bx) in msgs:
+        if src == self.rank:
+            rel = bx.copy()
+            rel[0] -= my_lines[0]
+            rel[1] -= my_lines[0]
+            sends.append((dst,
+                          self.backend.pack_temp_box("__axcs_tot__", rel)))
+        if dst == self.rank:
+            recvs.append((src,
+                          self.backend.new_message_buffer(box_shape(bx),
+                                                          out_dtype), bx))
+    if recvs:
+        self.backend.axcs_init_offsets(box_shape(my_lines), out_dtype)
+    if sends or recvs:
+        self.backend.exchange(sends, [(s, b) for (s, b, _) in recvs])
+    for (src, buf, bx) in recvs:
+        rel = bx.copy()
+        rel[0] -= my_lines[0]
+        rel[1] -= my_lines[0]
+        self.backend.axcs_accumulate(rel, buf)
+    if recvs:
+        self.backend.axcs_apply(out_bd, self, lb, axis)
+    self.backend.free_temps()
+    return out_bd
+
+
+Runtime.cumsum_axis_op = cumsum_axis_op

@@ -271,3 +271,25 @@ def test_mask_getitem_spmd():
                                     [float(sel.shape[0]),
                                      float(sel2.shape[0])]])
         """, world=world, tol=1e-12)
+
+
+def test_axis_cumsum_spmd():
+    """N-D cumsum along an axis: local scan + pairwise line-totals slab
+    exchange (runtime.cumsum_axis_op).  world=4 gives a 2x2 partition
+    grid, exercising partial line overlaps and multi-predecessor sums."""
+    for world in (2, 3, 4):
+        run_spmd("""
+            c = np_.fromfunction(lambda x, y: x * 97 + y, (37, 29))
+            r0 = c.cumsum(axis=0)
+            r1 = c.cumsum(axis=1)
+            v = c[3:33:2, 1:25]
+            r2 = v.cumsum(axis=1)
+            e = np_.fromfunction(lambda x, y, z: x * 100 + y * 10 + z,
+                                 (12, 10, 8))
+            r3 = e.cumsum(axis=0)
+            outs = [r0, r1, r2, r3]
+            if np_ is np:
+                return np.concatenate([o.reshape(-1) for o in outs])
+            import numpy as _np
+            return _np.concatenate([o.asarray().reshape(-1) for o in outs])
+        """, world=world, tol=1e-12)

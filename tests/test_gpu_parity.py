@@ -382,3 +382,28 @@ def test_mask_getitem_gpu_int(ra_gpu):
         import numpy as _np
         return _np.asarray(r.asarray(), dtype=np.float64)
     run_both(impl, ra_gpu)
+
+
+def test_axis_cumsum_gpu(ra_gpu):
+    """N-D axis cumsum through rt_axis_scan (thread-per-line for axis<last,
+    wave-per-line shfl scan for the last axis)."""
+    def impl(np_):
+        c = np_.fromfunction(lambda x, y: x * 997 + y, (513, 259))
+        r0 = c.cumsum(axis=0)
+        r1 = c.cumsum(axis=1)
+        v = c[3:500:2, 1:250]
+        r2 = v.cumsum(axis=1)
+        e = np_.fromfunction(lambda x, y, z: x * 100 + y * 10 + z,
+                             (40, 30, 70))
+        r3 = e.cumsum(axis=2)
+        i = np_.fromfunction(lambda x, y: x + y, (100, 65),
+                             dtype=np.int32).cumsum(axis=0)
+        outs = [r0, r1, r2, r3, i]
+        if np_ is np:
+            return np.concatenate([np.asarray(o, dtype=np.float64).reshape(-1)
+                                   for o in outs])
+        import numpy as _np
+        return _np.concatenate([_np.asarray(o.asarray(),
+                                            dtype=_np.float64).reshape(-1)
+                                for o in outs])
+    run_both(impl, ra_gpu)

@@ -755,3 +755,45 @@ class TestMaskGetitem:
         vi = i[20:180:2]
         np.testing.assert_allclose(v[(v % 4.0) == 0.0].asarray(),
                                    vi[(vi % 4) == 0])
+
+
+class TestAxisCumsum:
+    """N-D cumsum along an explicit axis (reference scumulative,
+    ramba.py:10057; runtime.cumsum_axis_op)."""
+
+    def test_2d_both_axes(self, ra):
+        c = ra.fromfunction(lambda x, y: x * 13 + y, (17, 13))
+        nc = np.fromfunction(lambda x, y: x * 13 + y, (17, 13))
+        np.testing.assert_allclose(c.cumsum(axis=0).asarray(),
+                                   nc.cumsum(axis=0))
+        np.testing.assert_allclose(c.cumsum(axis=1).asarray(),
+                                   nc.cumsum(axis=1))
+        np.testing.assert_allclose(c.cumsum(axis=-1).asarray(),
+                                   nc.cumsum(axis=-1))
+
+    def test_int_promotion(self, ra):
+        d = ra.fromfunction(lambda x, y: x + y, (9, 7), dtype=np.int32)
+        nd_ = np.fromfunction(lambda x, y: x + y, (9, 7)).astype(np.int32)
+        r = d.cumsum(axis=0)
+        assert r.dtype == np.int64
+        np.testing.assert_array_equal(r.asarray(), nd_.cumsum(axis=0))
+
+    def test_view_and_transpose_sources(self, ra):
+        c = ra.fromfunction(lambda x, y: x * 13 + y, (17, 13))
+        nc = np.fromfunction(lambda x, y: x * 13 + y, (17, 13))
+        np.testing.assert_allclose(c[2:15:2, 1:12].cumsum(axis=1).asarray(),
+                                   nc[2:15:2, 1:12].cumsum(axis=1))
+        np.testing.assert_allclose(c.T.cumsum(axis=0).asarray(),
+                                   nc.T.cumsum(axis=0))
+
+    def test_3d(self, ra):
+        e = ra.fromfunction(lambda x, y, z: x * 100 + y * 10 + z, (6, 5, 4))
+        ne = np.fromfunction(lambda x, y, z: x * 100 + y * 10 + z, (6, 5, 4))
+        for ax in range(3):
+            np.testing.assert_allclose(e.cumsum(axis=ax).asarray(),
+                                       ne.cumsum(axis=ax))
+
+    def test_nd_axis_none_raises(self, ra):
+        c = ra.fromfunction(lambda x, y: x + y, (5, 5))
+        with pytest.raises(ValueError):
+            c.cumsum()
