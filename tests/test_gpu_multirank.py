@@ -110,3 +110,20 @@ def test_wide_shift_temp_path_2rank_gpu():
         A = np_.arange(3000) * 1.0
         return A[:-300] + A[300:]
     """)
+
+
+def test_axis_cumsum_and_mask_2rank_gpu():
+    """Cross-rank slab-exchange fixup (cumsum_axis_op) and mask-getitem
+    count allgather + uneven divisions, on the real HIP backend."""
+    run_spmd_gpu("""
+        c = np_.fromfunction(lambda x, y: x * 97 + y, (401, 37))
+        r0 = c.cumsum(axis=0)
+        r1 = c.cumsum(axis=1)
+        a = np_.arange(10000) * 1.0
+        sel = a[(a % 7.0) == 0.0]
+        if np_ is np:
+            return np.concatenate([r0.reshape(-1), r1.reshape(-1), sel])
+        import numpy as _np
+        return _np.concatenate([r0.asarray().reshape(-1),
+                                r1.asarray().reshape(-1), sel.asarray()])
+    """, tol=1e-12)
