@@ -114,7 +114,7 @@ def build_program(seed):
             action = rng.choice(
                 ["un", "bin", "view", "setitem", "mask", "reduce",
                  "axred", "cumsum", "astype", "where", "clip", "mean",
-                 "outer"])
+                 "outer", "axcumsum", "maskget"])
             i = int(rng.integers(0, len(pool)))
             val, kind = pool[i]
             if action == "un":
@@ -213,6 +213,16 @@ def build_program(seed):
                 if val.ndim == 1 and val.shape[0] <= 200:
                     r = val[:, None] * (val[None, :] + 1)
                     pool.append((r, kind))
+            elif action == "axcumsum":
+                if val.ndim == 2 and kind in ("f", "i"):
+                    ax = int(rng.integers(0, 2))
+                    pool.append((val.cumsum(axis=ax), kind))
+            elif action == "maskget":
+                # integer-valued membership only: a threshold on computed
+                # floats would make selection itself ulp-sensitive
+                if kind == "i":
+                    k = int(rng.integers(2, 9))
+                    pool.append((val[(val % k) == 0], kind))
 
         # result: flattened concat of the last few pool values + scalars
         outs = []
