@@ -18,6 +18,10 @@ from . import deferred as _deferred
 from .runtime import Runtime
 from .ndarray import *          # noqa: F401,F403  (the API surface)
 from .ndarray import ndarray    # noqa: F401
+from .ndarray import _MOD_DELEGATES as _mdel, _module_delegate as _mkdel
+for _n in _mdel:
+    globals()[_n] = _mkdel(_n)
+del _mdel, _mkdel
 from .common import dprint, get_timing, reset_timing  # noqa: F401
 
 __version__ = "0.1.0"

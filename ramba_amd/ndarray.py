@@ -490,6 +490,93 @@ class ndarray:
             cnt *= self.shape[a + self.ndim if a < 0 else a]
         return s * (1.0 / cnt)
 
+    # -- nan-aware / tolerance / axis-permutation vocabulary (the remaining
+    #    names of the reference's mod_to_array export list, ramba.py:9697;
+    #    all expressed through the fused elementwise/reduction machinery) --
+
+    def nansum(self, axis=None, dtype=None, keepdims=False, **kw):
+        if np.dtype(self.dtype).kind != "f":
+            return self.sum(axis=axis, dtype=dtype, keepdims=keepdims)
+        cleaned = where(self.isnan(), 0.0, self).astype(
+            self.dtype if dtype is None else dtype)
+        return cleaned.sum(axis=axis, keepdims=keepdims)
+
+    def nanmean(self, axis=None, keepdims=False, **kw):
+        if np.dtype(self.dtype).kind != "f":
+            return self.mean(axis=axis, keepdims=keepdims)
+        valid = self.isnan().logical_not()
+        cnt = valid.astype(np.int64).sum(axis=axis, keepdims=keepdims)
+        tot = self.nansum(axis=axis, keepdims=keepdims)
+        return tot / cnt
+
+    def isposinf(self):
+        return self == np.inf
+
+    def isneginf(self):
+        return self == -np.inf
+
+    def isclose(self, other, rtol=1e-5, atol=1e-8, **kw):
+        # numpy semantics: |a-b| <= atol + rtol*|b| where both finite,
+        # exact equality otherwise (inf==inf True, NaN always False)
+        if isinstance(other, np.ndarray):
+            other = fromarray(other)
+        diff_ok = abs(self - other) <= (atol + rtol * abs(other))
+        if isinstance(other, ndarray):
+            finite = self.isfinite().logical_and(other.isfinite())
+        else:
+            finite = self.isfinite() if np.isfinite(other)                 else self.isfinite().logical_and(False)
+        return where(finite, diff_ok, self == other)
+
+    def allclose(self, other, rtol=1e-5, atol=1e-8, **kw):
+        return bool(self.isclose(other, rtol=rtol, atol=atol).all())
+
+    def var(self, axis=None, ddof=0, keepdims=False, **kw):
+        m = self.mean(axis=axis, keepdims=True) if axis is not None             else self.mean()
+        dev2 = (self - m) * (self - m)
+        if axis is None:
+            n = self.size
+            return dev2.sum() / (n - ddof)
+        axes = (axis,) if isinstance(axis, (int, np.integer)) else             tuple(axis)
+        n = 1
+        for a in axes:
+            n *= self.shape[a + self.ndim if a < 0 else a]
+        return dev2.sum(axis=axis, keepdims=keepdims) * (1.0 / (n - ddof))
+
+    def std(self, axis=None, ddof=0, keepdims=False, **kw):
+        v = self.var(axis=axis, ddof=ddof, keepdims=keepdims)
+        return v ** 0.5 if isinstance(v, ndarray) else float(v) ** 0.5
+
+    def swapaxes(self, i, j):
+        n = self.ndim
+        i = i + n if i < 0 else i
+        j = j + n if j < 0 else j
+        perm = list(range(n))
+        perm[i], perm[j] = perm[j], perm[i]
+        return self.transpose(perm)
+
+    def moveaxis(self, source, destination):
+        n = self.ndim
+        src = [source] if isinstance(source, (int, np.integer))             else list(source)
+        dst = [destination] if isinstance(destination, (int, np.integer))             else list(destination)
+        src = [a + n if a < 0 else a for a in src]
+        dst = [a + n if a < 0 else a for a in dst]
+        order = [a for a in range(n) if a not in src]
+        for d, s_ in sorted(zip(dst, src)):
+            order.insert(d, s_)
+        return self.transpose(order)
+
+    def rollaxis(self, axis, start=0):
+        n = self.ndim
+        axis = axis + n if axis < 0 else axis
+        if start < 0:
+            start += n
+        if axis < start:
+            start -= 1
+        axes = list(range(n))
+        axes.remove(axis)
+        axes.insert(start, axis)
+        return self.transpose(axes)
+
 
 # -- attach the op tables (reference make_method loops, ramba.py:7893-7973) --
 
@@ -890,4 +977,41 @@ ndarray._ARRAY_FUNC.update({
     "shape": lambda a: a.shape,
     "ndim": lambda a: a.ndim,
     "size": lambda a: a.size,
+})
+
+
+# remaining mod_to_array names (reference ramba.py:9697): delegate to the
+# method when given a ramba array, to numpy otherwise.  These are exported
+# through the PACKAGE namespace (ramba_amd/__init__.py), NOT injected into
+# this module's globals — `all`/`any`/`abs`/`sum` must keep their builtin
+# meaning inside this file.
+def _module_delegate(name):
+    def f(a, *args, **kwargs):
+        if isinstance(a, ndarray):
+            return getattr(a, name)(*args, **kwargs)
+        if name == "abs":
+            import builtins
+            return builtins.abs(a)
+        return getattr(np, name)(a, *args, **kwargs)
+    f.__name__ = name
+    return f
+
+
+_MOD_DELEGATES = ("mean", "nanmean", "nansum", "isneginf", "isposinf",
+                  "all", "any", "isclose", "allclose", "swapaxes",
+                  "moveaxis", "rollaxis", "var", "std", "prod", "abs")
+
+
+ndarray._ARRAY_FUNC.update({
+    "nansum": lambda a, axis=None, **kw: a.nansum(axis=axis),
+    "nanmean": lambda a, axis=None, **kw: a.nanmean(axis=axis),
+    "isclose": lambda a, b, rtol=1e-5, atol=1e-8, **kw:
+        a.isclose(b, rtol=rtol, atol=atol),
+    "allclose": lambda a, b, rtol=1e-5, atol=1e-8, **kw:
+        a.allclose(b, rtol=rtol, atol=atol),
+    "swapaxes": lambda a, i, j: a.swapaxes(i, j),
+    "moveaxis": lambda a, s, d: a.moveaxis(s, d),
+    "rollaxis": lambda a, ax, start=0: a.rollaxis(ax, start),
+    "var": lambda a, axis=None, ddof=0, **kw: a.var(axis=axis, ddof=ddof),
+    "std": lambda a, axis=None, ddof=0, **kw: a.std(axis=axis, ddof=ddof),
 })

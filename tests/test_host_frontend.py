@@ -797,3 +797,89 @@ class TestAxisCumsum:
         c = ra.fromfunction(lambda x, y: x + y, (5, 5))
         with pytest.raises(ValueError):
             c.cumsum()
+
+
+class TestExtendedVocabulary:
+    """The remaining mod_to_array exports of the reference (ramba.py:9697):
+    nan-aware reductions, isclose/allclose, axis permutations, var/std."""
+
+    def test_nansum_nanmean(self, ra):
+        a = ra.arange(50) * 1.0
+        b = ra.where(a % 7.0 == 0.0, ra.full(50, np.nan), a)
+        i = np.arange(50) * 1.0
+        nb = np.where(i % 7 == 0, np.nan, i)
+        assert abs(float(b.nansum()) - np.nansum(nb)) < 1e-9
+        assert abs(float(b.nanmean()) - np.nanmean(nb)) < 1e-12
+        # module + protocol forms
+        assert abs(float(ra.nansum(b)) - np.nansum(nb)) < 1e-9
+        assert abs(float(np.nansum(b)) - np.nansum(nb)) < 1e-9
+        # int arrays: plain sum/mean
+        c = ra.arange(10)
+        assert int(c.nansum()) == 45
+
+    def test_nan_axis(self, ra):
+        a = ra.fromfunction(lambda x, y: x * 7 + y, (6, 7))
+        b = ra.where(a % 5.0 == 0.0, ra.full((6, 7), np.nan), a)
+        na = np.fromfunction(lambda x, y: x * 7 + y, (6, 7))
+        nb = np.where(na % 5 == 0, np.nan, na)
+        np.testing.assert_allclose(b.nansum(axis=0).asarray(),
+                                   np.nansum(nb, axis=0))
+        np.testing.assert_allclose(b.nanmean(axis=1).asarray(),
+                                   np.nanmean(nb, axis=1))
+
+    def test_isposinf_isneginf(self, ra):
+        a = ra.arange(10) * 1.0 - 5.0
+        b = 1.0 / ra.where(a == 0.0, 0.0, a)   # inf at the zero
+        i = np.arange(10) * 1.0 - 5.0
+        with np.errstate(all="ignore"):
+            nb = 1.0 / np.where(i == 0, 0.0, i)
+        np.testing.assert_array_equal(b.isposinf().asarray(),
+                                      np.isposinf(nb))
+        np.testing.assert_array_equal(b.isneginf().asarray(),
+                                      np.isneginf(nb))
+
+    def test_isclose_allclose(self, ra):
+        a = ra.arange(100) * 1.0
+        b = a + 1e-9
+        i = np.arange(100) * 1.0
+        np.testing.assert_array_equal(a.isclose(b).asarray(),
+                                      np.isclose(i, i + 1e-9))
+        assert a.allclose(b)
+        assert not a.allclose(a + 1.0)
+        assert ra.allclose(a, b)
+        # NaN never close; inf == inf close
+        c = ra.full(4, np.nan)
+        assert not bool(c.isclose(c).any())
+        d = ra.full(4, np.inf)
+        assert bool(d.isclose(d).all())
+
+    def test_axis_permutations(self, ra):
+        a = ra.fromfunction(lambda x, y, z: x * 20 + y * 4 + z, (3, 5, 4))
+        n = np.fromfunction(lambda x, y, z: x * 20 + y * 4 + z, (3, 5, 4))
+        np.testing.assert_array_equal(a.swapaxes(0, 2).asarray(),
+                                      n.swapaxes(0, 2))
+        np.testing.assert_array_equal(a.moveaxis(0, -1).asarray(),
+                                      np.moveaxis(n, 0, -1))
+        np.testing.assert_array_equal(a.moveaxis([0, 1], [2, 0]).asarray(),
+                                      np.moveaxis(n, [0, 1], [2, 0]))
+        for ax, st in ((2, 0), (0, 2), (1, 0), (2, 1)):
+            np.testing.assert_array_equal(
+                a.rollaxis(ax, st).asarray(), np.rollaxis(n, ax, st),
+                err_msg=f"rollaxis({ax},{st})")
+        np.testing.assert_array_equal(np.swapaxes(a, 0, 1).asarray(),
+                                      n.swapaxes(0, 1))
+
+    def test_var_std(self, ra):
+        a = ra.arange(1000) * 0.37
+        i = np.arange(1000) * 0.37
+        assert abs(float(a.var()) - i.var()) < 1e-9
+        assert abs(float(a.std()) - i.std()) < 1e-10
+        assert abs(float(a.var(ddof=1)) - i.var(ddof=1)) < 1e-9
+        b = ra.fromfunction(lambda x, y: x * 3.0 + y * y, (8, 9))
+        n = np.fromfunction(lambda x, y: x * 3.0 + y * y, (8, 9))
+        np.testing.assert_allclose(b.var(axis=0).asarray(), n.var(axis=0),
+                                   rtol=1e-12, atol=1e-9)
+        np.testing.assert_allclose(b.std(axis=1).asarray(), n.std(axis=1),
+                                   rtol=1e-12, atol=1e-9)
+        np.testing.assert_allclose(np.var(b, axis=0).asarray(),
+                                   n.var(axis=0), rtol=1e-12, atol=1e-9)
