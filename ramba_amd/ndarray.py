@@ -233,7 +233,8 @@ class ndarray:
                   "logical_and": "logical_and", "logical_or": "logical_or",
                   "logical_xor": "logical_xor", "bitwise_and": "__and__",
                   "bitwise_or": "__or__", "bitwise_xor": "__xor__",
-                  "left_shift": "__lshift__", "right_shift": "__rshift__"}
+                  "left_shift": "__lshift__", "right_shift": "__rshift__",
+                  "matmul": "__matmul__"}
     _UFUNC_UN = {"sin": "sin", "cos": "cos", "tan": "tan", "sinh": "sinh",
                  "cosh": "cosh", "tanh": "tanh", "arcsin": "arcsin",
                  "arccos": "arccos", "arctan": "arctan", "exp": "exp",
@@ -1014,4 +1015,129 @@ ndarray._ARRAY_FUNC.update({
     "rollaxis": lambda a, ax, start=0: a.rollaxis(ax, start),
     "var": lambda a, axis=None, ddof=0, **kw: a.var(axis=axis, ddof=ddof),
     "std": lambda a, axis=None, ddof=0, **kw: a.std(axis=axis, ddof=ddof),
+})
+
+
+# ---------------------------------------------------------------------------
+# matmul / expand_dims / triu / meshgrid / item — the remaining user-facing
+# surface of the reference around the hot path
+# ---------------------------------------------------------------------------
+
+def expand_dims(a, axis):
+    """reference expand_dims (ramba.py:9438)."""
+    if not isinstance(a, ndarray):
+        return np.expand_dims(a, axis)
+    nd = a.ndim + 1
+    axis = axis + nd if axis < 0 else axis
+    idx = (slice(None),) * axis + (None,) + (slice(None),) * (a.ndim - axis)
+    return a[idx]
+
+
+def matmul(a, b):
+    """reference matmul (ramba.py:6953): every case lowers to the
+    broadcast-views × fused-multiply × axis-sum composition the reference
+    itself uses for its N-D path (`(a*b).sum(axis=-2)`).  The K-axis
+    product is materialised before the axis reduction (no dense-GEMM MFMA
+    path — `north_star` has no dense contraction in scope), so this is
+    meant for the reference's moderate-size matmul uses, not as a BLAS."""
+    if isinstance(a, np.ndarray):
+        a = fromarray(a)
+    if isinstance(b, np.ndarray):
+        b = fromarray(b)
+    if not isinstance(a, ndarray) or not isinstance(b, ndarray):
+        raise ValueError("matmul cannot be used with scalar arguments")
+    ashape, bshape = a.shape, b.shape
+    if a.ndim == 1 and b.ndim == 1:
+        if ashape[0] != bshape[0]:
+            raise ValueError(f"matmul: mismatched shapes {ashape}/{bshape}")
+        return (a * b).sum()
+    if b.ndim == 1:
+        if ashape[-1] != bshape[0]:
+            raise ValueError(f"matmul: mismatched shapes {ashape}/{bshape}")
+        return (a * b.broadcast_to(ashape)).sum(axis=-1)
+    if a.ndim == 1:
+        if ashape[0] != bshape[-2]:
+            raise ValueError(f"matmul: mismatched shapes {ashape}/{bshape}")
+        aa = expand_dims(a, -1).broadcast_to(bshape)
+        return (aa * b).sum(axis=-2)
+    if ashape[-1] != bshape[-2]:
+        raise ValueError(f"matmul: mismatched shapes {ashape}/{bshape}")
+    aa = expand_dims(a, -1)
+    bb = expand_dims(b, -3)
+    shp = np.broadcast_shapes(ashape[:-2], bshape[:-2]) \
+        + ashape[-2:-1] + bshape[-2:]
+    return (aa.broadcast_to(shp) * bb.broadcast_to(shp)).sum(axis=-2)
+
+
+def _nd_matmul(self, rhs):
+    return matmul(self, rhs)
+
+
+def _nd_rmatmul(self, lhs):
+    return matmul(lhs, self)
+
+
+def _nd_item(self, *args):
+    if not args:
+        assert self.size == 1, "item(): array has more than one element"
+        return self.asarray().reshape(-1)[0].item()
+    if len(args) == 1 and isinstance(args[0], tuple):
+        args = args[0]
+    if len(args) == 1 and self.ndim != 1:
+        return self.asarray().reshape(-1)[int(args[0])].item()
+    v = self[tuple(int(x) for x in args)]
+    return v.item() if isinstance(v, np.generic) else v
+
+
+ndarray.__matmul__ = _nd_matmul
+ndarray.__rmatmul__ = _nd_rmatmul
+ndarray.item = _nd_item
+
+
+def triu(m, k=0):
+    """Upper triangle (reference triu, ramba.py:9053/:2091; 2-D only like
+    the reference worker)."""
+    if not isinstance(m, ndarray):
+        return np.triu(m, k)
+    assert m.ndim == 2, "triu: 2-D only (reference triu_executor)"
+    ij = fromfunction(lambda i, j: j - i, m.shape, dtype=np.int64)
+    return where(ij >= k, m, zeros(m.shape, dtype=m.dtype))
+
+
+def tril(m, k=0):
+    if not isinstance(m, ndarray):
+        return np.tril(m, k)
+    assert m.ndim == 2, "tril: 2-D only"
+    ij = fromfunction(lambda i, j: j - i, m.shape, dtype=np.int64)
+    return where(ij <= k, m, zeros(m.shape, dtype=m.dtype))
+
+
+def meshgrid(*xi, copy=True, sparse=False, indexing="ij"):
+    """reference meshgrid (ramba.py:9028): 'ij' indexing only, returns ONE
+    stacked (len(xi), n1, ..., nk) distributed array (the reference stacks
+    np.meshgrid output into a single backing array, :3921-3927)."""
+    if indexing != "ij":
+        raise ValueError(f"Unsupported meshgrid indexing option {indexing}")
+    if sparse or not copy:
+        raise ValueError("Unsupported meshgrid sparse/copy option")
+    xs = [np.asarray(x) for x in xi]
+    if any(x.ndim != 1 for x in xs):
+        raise ValueError("Unsupported argument to meshgrid")
+    if not all(x.dtype == xs[0].dtype for x in xs):
+        raise ValueError("Mis-matching dtypes to meshgrid")
+    nd = len(xs)
+    sizes = tuple(len(x) for x in xs)
+    out = empty((nd,) + sizes, dtype=xs[0].dtype)
+    for i, x in enumerate(xs):
+        idx = tuple(slice(None) if d == i else None for d in range(nd))
+        out[i] = fromarray(x)[idx].broadcast_to(sizes)
+    return out
+
+
+ndarray._ARRAY_FUNC.update({
+    "matmul": lambda a, b, **kw: matmul(a, b),
+    "dot": lambda a, b, **kw: matmul(a, b),
+    "expand_dims": lambda a, axis: expand_dims(a, axis),
+    "triu": lambda m, k=0: triu(m, k),
+    "tril": lambda m, k=0: tril(m, k),
 })

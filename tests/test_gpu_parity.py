@@ -407,3 +407,38 @@ def test_axis_cumsum_gpu(ra_gpu):
                                             dtype=_np.float64).reshape(-1)
                                 for o in outs])
     run_both(impl, ra_gpu)
+
+
+def test_extended_vocabulary_gpu(ra_gpu):
+    """NaN-aware reductions, isclose, var/std through the HIP kernels."""
+    def impl(np_):
+        a = np_.arange(100_000) * 1.0
+        b = np_.where(a % 97.0 == 0.0,
+                      np_.full(100_000, np.nan) if np_ is not np
+                      else np.full(100_000, np.nan), a)
+        ns = b.nansum() if hasattr(b, "nansum") else np.nansum(b)
+        nm = b.nanmean() if hasattr(b, "nanmean") else np.nanmean(b)
+        v = a.var()
+        sd = a.std()
+        ic = np_.isclose(a, a + 1e-9)
+        icn = float(ic.sum() if hasattr(ic, "asarray") else ic.sum())
+        return np.asarray([float(ns), float(nm), float(v), float(sd), icn])
+    r, n = run_both(impl, ra_gpu, tol=1e-9)
+
+
+def test_matmul_gpu(ra_gpu):
+    """@ through broadcast views + fused multiply + axis-sum kernels."""
+    def impl(np_):
+        A = np_.fromfunction(lambda i, j: i * 7.0 + j, (65, 33))
+        B = np_.fromfunction(lambda i, j: i - 2.0 * j, (33, 49))
+        v = np_.arange(33) * 1.0
+        M = A @ B
+        w = A @ v
+        t = np_.triu(np_.fromfunction(lambda i, j: i * 10.0 + j, (31, 37)),
+                     2)
+        outs = [M, w, t]
+        if np_ is np:
+            return np.concatenate([np.asarray(o).reshape(-1) for o in outs])
+        import numpy as _np
+        return _np.concatenate([o.asarray().reshape(-1) for o in outs])
+    run_both(impl, ra_gpu, tol=1e-10)

@@ -883,3 +883,57 @@ class TestExtendedVocabulary:
                                    rtol=1e-12, atol=1e-9)
         np.testing.assert_allclose(np.var(b, axis=0).asarray(),
                                    n.var(axis=0), rtol=1e-12, atol=1e-9)
+
+
+class TestMatmulTriuMeshgrid:
+    """matmul (reference ramba.py:6953: broadcast × fused-mul × axis-sum
+    composition), triu/tril (:9053), stacked 'ij' meshgrid (:9028), item."""
+
+    def test_matmul_all_arities(self, ra):
+        A = ra.fromfunction(lambda i, j: i * 7.0 + j, (13, 9))
+        B = ra.fromfunction(lambda i, j: i - 2.0 * j, (9, 11))
+        nA = np.fromfunction(lambda i, j: i * 7.0 + j, (13, 9))
+        nB = np.fromfunction(lambda i, j: i - 2.0 * j, (9, 11))
+        v, nv = ra.arange(9) * 1.0, np.arange(9) * 1.0
+        np.testing.assert_allclose((A @ B).asarray(), nA @ nB, rtol=1e-12)
+        np.testing.assert_allclose((A @ v).asarray(), nA @ nv)
+        np.testing.assert_allclose((v @ B).asarray(), nv @ nB)
+        assert abs(float(v @ v) - nv @ nv) < 1e-9
+        np.testing.assert_allclose(np.matmul(A, B).asarray(), nA @ nB,
+                                   rtol=1e-12)
+        np.testing.assert_allclose((nA @ B).asarray(), nA @ nB, rtol=1e-12)
+        with pytest.raises(ValueError):
+            A @ ra.arange(5)
+
+    def test_matmul_batched_3d(self, ra):
+        C = ra.fromfunction(lambda b, i, j: b + i * 2.0 + j, (4, 5, 6))
+        D = ra.fromfunction(lambda b, i, j: b - i + 3.0 * j, (4, 6, 7))
+        nC = np.fromfunction(lambda b, i, j: b + i * 2.0 + j, (4, 5, 6))
+        nD = np.fromfunction(lambda b, i, j: b - i + 3.0 * j, (4, 6, 7))
+        np.testing.assert_allclose((C @ D).asarray(), nC @ nD, rtol=1e-12)
+
+    def test_triu_tril(self, ra):
+        M = ra.fromfunction(lambda i, j: i * 10.0 + j + 1, (6, 8))
+        nM = np.fromfunction(lambda i, j: i * 10.0 + j + 1, (6, 8))
+        for k in (-2, 0, 1, 3):
+            np.testing.assert_array_equal(ra.triu(M, k).asarray(),
+                                          np.triu(nM, k))
+            np.testing.assert_array_equal(ra.tril(M, k).asarray(),
+                                          np.tril(nM, k))
+        np.testing.assert_array_equal(np.triu(M).asarray(), np.triu(nM))
+
+    def test_meshgrid_stacked_ij(self, ra):
+        g = ra.meshgrid(np.arange(4), np.arange(5), np.arange(3))
+        ng = np.stack(np.meshgrid(np.arange(4), np.arange(5), np.arange(3),
+                                  indexing="ij"))
+        assert g.shape == ng.shape
+        np.testing.assert_array_equal(g.asarray(), ng)
+        with pytest.raises(ValueError):
+            ra.meshgrid(np.arange(3), indexing="xy")
+
+    def test_item(self, ra):
+        a = ra.arange(10)
+        assert a[3:4].item() == 3
+        A = ra.fromfunction(lambda i, j: i * 7.0 + j, (4, 5))
+        assert A.item(2, 3) == 17.0
+        assert A.item((1, 1)) == 8.0
