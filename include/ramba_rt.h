@@ -107,6 +107,13 @@ int rt_mask_compact(uintptr_t stream, const void *a, const void *m,
                     const int64_t *a_strides, const int64_t *m_strides,
                     void *bcounts, int64_t nchunks, int dtype, int phase);
 
+/* Flat C-order gather (scatter=0) / scatter (scatter=1) between a strided
+ * local box and a dense buffer — the data movement of reshape (reference
+ * flat-index remap worker, ramba/ramba.py:2409-2492). */
+int rt_flat_copy(uintptr_t stream, void *boxed, void *dense, int nd,
+                 const int64_t *shape, const int64_t *strides,
+                 int64_t flat0, int64_t n, int elemsize, int scatter);
+
 int rt_stream_sync(uintptr_t stream);
 int rt_device_sync(void);
 

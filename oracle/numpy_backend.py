@@ -310,3 +310,34 @@ NumpyBackend.axis_scan_local = _nb_axis_scan_local
 NumpyBackend.axcs_init_offsets = _nb_axcs_init_offsets
 NumpyBackend.axcs_accumulate = _nb_axcs_accumulate
 NumpyBackend.axcs_apply = _nb_axcs_apply
+
+
+def _nb_flat_view(cont, off0, strides, shape):
+    base = cont.reshape(-1)
+    isz = base.itemsize
+    return np.lib.stride_tricks.as_strided(
+        base[off0:] if off0 else base, shape=tuple(shape),
+        strides=tuple(s * isz for s in strides))
+
+
+def _nb_flat_gather(self, cont, off0, strides, shape, flat0, n):
+    import torch
+    v = _nb_flat_view(cont, off0, strides, shape).reshape(-1)
+    return torch.from_numpy(np.ascontiguousarray(v[flat0:flat0 + n]))
+
+
+def _nb_flat_scatter(self, cont, off0, strides, shape, flat0, buf):
+    v = _nb_flat_view(cont, off0, strides, shape)
+    flat = v.reshape(-1) if v.flags["C_CONTIGUOUS"] else None
+    b = buf.numpy()
+    if flat is not None:
+        flat[flat0:flat0 + b.size] = b
+        return
+    # strided destination: walk coords
+    idx = np.arange(flat0, flat0 + b.size)
+    coords = np.unravel_index(idx, tuple(shape))
+    v[coords] = b
+
+
+NumpyBackend.flat_gather = _nb_flat_gather
+NumpyBackend.flat_scatter = _nb_flat_scatter

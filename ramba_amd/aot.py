@@ -123,6 +123,12 @@ class AotCompileBackend:
     def axcs_apply(self, *a):
         pass
 
+    def flat_gather(self, cont, off0, strides, shape, flat0, n):
+        return None
+
+    def flat_scatter(self, *a):
+        pass
+
     def write_local_dense(self, out_bd, rt, local):
         pass
 

@@ -216,3 +216,29 @@ def default_divisions(num_workers, shape):
     if len(shape) == 0:
         raise ValueError("0-d arrays are not distributed")
     return compute_regular_schedule(num_workers, tuple(int(s) for s in shape))
+
+
+def contiguous_divisions(world, shape):
+    """Axis-0-only balanced split: every shard covers full trailing axes,
+    so shards are C-contiguous flat intervals (reshape/compaction need
+    this)."""
+    import numpy as np
+    nd = len(shape)
+    divs = np.zeros((world, 2, nd), dtype=np.int64)
+    if any(int(s) == 0 for s in shape):
+        divs[:, 1, :] = -1
+        return divs
+    rows = int(shape[0])
+    base, rem = divmod(rows, world)
+    lo = 0
+    for r in range(world):
+        cnt = base + (1 if r < rem else 0)
+        if cnt == 0:
+            divs[r, 1, :] = -1
+            continue
+        divs[r, 0, 0] = lo
+        divs[r, 1, 0] = lo + cnt - 1
+        for d in range(1, nd):
+            divs[r, 1, d] = int(shape[d]) - 1
+        lo += cnt
+    return divs

@@ -1168,3 +1168,101 @@ extern "C" int rt_axis_scan(uintptr_t stream, const void *in, void *out,
             return 1;
     }
 }
+
+// ---------------------------------------------------------------------------
+// flat-order gather/scatter between a strided local box and a dense buffer
+// (reshape data movement; the reference's per-element flat-index remap
+// worker, ramba/ramba.py:2409-2492, becomes interval exchange + these two
+// kernels).  flat0/n select a C-order flat subrange of the box.
+// ---------------------------------------------------------------------------
+
+namespace {
+
+struct FlatArgs {
+    int64_t n;          // elements in the subrange
+    int64_t flat0;      // first flat index within the box
+    int64_t shape[4];
+    int64_t str[4];     // element strides of the boxed side
+    int nd;
+};
+
+template <typename T, int SCATTER>
+__global__ __launch_bounds__(256) void flat_copy_kernel(
+    T *__restrict__ boxed, T *__restrict__ dense, FlatArgs g) {
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < g.n; i += stride) {
+        int64_t rem = g.flat0 + i, off = 0;
+        for (int d = g.nd - 1; d >= 0; --d) {
+            int64_t idx = rem % g.shape[d];
+            rem /= g.shape[d];
+            off += idx * g.str[d];
+        }
+        if (SCATTER) boxed[off] = dense[i];
+        else dense[i] = boxed[off];
+    }
+}
+
+template <typename T>
+int flat_copy_launch(uintptr_t stream, void *boxed, void *dense,
+                     const FlatArgs &g, int scatter) {
+    int64_t blocks = (g.n + 255) / 256;
+    if (blocks > 4096) blocks = 4096;
+    if (blocks < 1) blocks = 1;
+    hipStream_t st = reinterpret_cast<hipStream_t>(stream);
+    if (scatter)
+        hipLaunchKernelGGL((flat_copy_kernel<T, 1>), dim3((unsigned)blocks),
+                           dim3(256), 0, st, static_cast<T *>(boxed),
+                           static_cast<T *>(dense), g);
+    else
+        hipLaunchKernelGGL((flat_copy_kernel<T, 0>), dim3((unsigned)blocks),
+                           dim3(256), 0, st, static_cast<T *>(boxed),
+                           static_cast<T *>(dense), g);
+    hipError_t e = hipGetLastError();
+    if (e != hipSuccess) {
+        set_error(std::string("flat_copy launch: ") + hipGetErrorString(e));
+        return 1;
+    }
+    return 0;
+}
+
+}  // namespace
+
+// boxed pointer is pre-offset to the box origin; scatter=0 gathers the
+// flat subrange [flat0, flat0+n) of the box into dense, scatter=1 writes
+// dense back into that subrange.
+extern "C" int rt_flat_copy(uintptr_t stream, void *boxed, void *dense,
+                            int nd, const int64_t *shape,
+                            const int64_t *strides, int64_t flat0,
+                            int64_t n, int elemsize, int scatter) {
+    if (nd < 1 || nd > 4) {
+        set_error("rt_flat_copy: nd out of range");
+        return 1;
+    }
+    FlatArgs g;
+    g.nd = nd;
+    g.n = n;
+    g.flat0 = flat0;
+    for (int d = 0; d < nd; ++d) {
+        g.shape[d] = shape[d];
+        g.str[d] = strides[d];
+    }
+    for (int d = nd; d < 4; ++d) {
+        g.shape[d] = 1;
+        g.str[d] = 0;
+    }
+    if (n <= 0) return 0;
+    switch (elemsize) {
+        case 1: return flat_copy_launch<uint8_t>(stream, boxed, dense, g,
+                                                 scatter);
+        case 2: return flat_copy_launch<uint16_t>(stream, boxed, dense, g,
+                                                  scatter);
+        case 4: return flat_copy_launch<uint32_t>(stream, boxed, dense, g,
+                                                  scatter);
+        case 8: return flat_copy_launch<uint64_t>(stream, boxed, dense, g,
+                                                  scatter);
+        default:
+            set_error("rt_flat_copy: bad elemsize");
+            return 1;
+    }
+}

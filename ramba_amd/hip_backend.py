@@ -83,6 +83,12 @@ def _load_lib():
                                  ctypes.POINTER(ctypes.c_int64),
                                  ctypes.POINTER(ctypes.c_int64),
                                  ctypes.c_int, ctypes.c_int]
+    lib.rt_flat_copy.argtypes = [ctypes.c_size_t, ctypes.c_void_p,
+                                 ctypes.c_void_p, ctypes.c_int,
+                                 ctypes.POINTER(ctypes.c_int64),
+                                 ctypes.POINTER(ctypes.c_int64),
+                                 ctypes.c_int64, ctypes.c_int64,
+                                 ctypes.c_int, ctypes.c_int]
     lib.rt_stream_sync.argtypes = [ctypes.c_size_t]
     return lib
 
@@ -808,3 +814,32 @@ HipBackend.axis_scan_local = _hb_axis_scan_local
 HipBackend.axcs_init_offsets = _hb_axcs_init_offsets
 HipBackend.axcs_accumulate = _hb_axcs_accumulate
 HipBackend.axcs_apply = _hb_axcs_apply
+
+
+# -- reshape flat gather/scatter ---------------------------------------------
+
+
+def _hb_flat_gather(self, cont, off0, strides, shape, flat0, n):
+    """Dense device buffer = C-order flat subrange [flat0, flat0+n) of the
+    strided box at cont[off0...]."""
+    dense = self.torch.empty(n, dtype=cont.dtype, device="cuda")
+    p = cont.data_ptr() + off0 * cont.element_size()
+    self._check(self.lib.rt_flat_copy(
+        self._stream(), ctypes.c_void_p(p),
+        ctypes.c_void_p(dense.data_ptr()), len(shape), _i64arr(shape),
+        _i64arr(strides), int(flat0), int(n), cont.element_size(), 0),
+        "rt_flat_copy(gather)")
+    return dense
+
+
+def _hb_flat_scatter(self, cont, off0, strides, shape, flat0, buf):
+    p = cont.data_ptr() + off0 * cont.element_size()
+    self._check(self.lib.rt_flat_copy(
+        self._stream(), ctypes.c_void_p(p),
+        ctypes.c_void_p(buf.data_ptr()), len(shape), _i64arr(shape),
+        _i64arr(strides), int(flat0), int(buf.numel()),
+        cont.element_size(), 1), "rt_flat_copy(scatter)")
+
+
+HipBackend.flat_gather = _hb_flat_gather
+HipBackend.flat_scatter = _hb_flat_scatter

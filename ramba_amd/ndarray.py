@@ -1141,3 +1141,77 @@ ndarray._ARRAY_FUNC.update({
     "triu": lambda m, k=0: triu(m, k),
     "tril": lambda m, k=0: tril(m, k),
 })
+
+
+# ---------------------------------------------------------------------------
+# reshape / ravel / flatten (reference ramba.py:6716-6721 + the reshape
+# worker 2409-2492; always a copy here, like the reference's general path)
+# ---------------------------------------------------------------------------
+
+def _nd_reshape(self, *newshape):
+    if len(newshape) == 1 and isinstance(newshape[0], (tuple, list)):
+        newshape = tuple(newshape[0])
+    newshape = [int(x) for x in newshape]
+    neg = [i for i, x in enumerate(newshape) if x < 0]
+    if neg:
+        assert len(neg) == 1, "can only specify one unknown dimension"
+        known = 1
+        for i, x in enumerate(newshape):
+            if i != neg[0]:
+                known *= x
+        assert known > 0 and self.size % known == 0, \
+            f"cannot reshape array of size {self.size} into {newshape}"
+        newshape[neg[0]] = self.size // known
+    newshape = tuple(newshape)
+    if newshape == self.shape:
+        return self.copy()
+    deferred.flush()
+    rt = deferred.get_runtime()
+    src = self
+    # interval exchange needs C-contiguous source shards; repartition
+    # through the fused engine (a plain assignment onto an array with
+    # prescribed axis-0 divisions) when the schedule split trailing axes
+    from .runtime import _flat_interval
+    from .shardview import exec_boxes as _eb
+    lbs = _eb(src.view, src.bdarray.divisions)
+    if any(b is not None and _flat_interval(b, src.view.shape) is None
+           for b in lbs):
+        from .common import contiguous_divisions
+        tmp_bd = deferred.bdarray(
+            self.shape, self.dtype,
+            contiguous_divisions(rt.world, self.shape),
+            default_border, flex=False)
+        tmp = ndarray(tmp_bd, View.identity(self.shape))
+        deferred.add_op(tmp, "=", self, empty_like=empty_like)
+        deferred.flush()
+        src = tmp
+    out_bd = rt.reshape_op(src, newshape)
+    return ndarray(out_bd, View.identity(newshape))
+
+
+def _nd_ravel(self):
+    return self.reshape(self.size)
+
+
+ndarray.reshape = _nd_reshape
+ndarray.reshape_copy = _nd_reshape
+ndarray.ravel = _nd_ravel
+ndarray.flatten = _nd_ravel
+
+
+def reshape(a, newshape):
+    if isinstance(a, ndarray):
+        return a.reshape(newshape)
+    return np.reshape(a, newshape)
+
+
+def ravel(a):
+    if isinstance(a, ndarray):
+        return a.ravel()
+    return np.ravel(a)
+
+
+ndarray._ARRAY_FUNC.update({
+    "reshape": lambda a, shape, **kw: a.reshape(shape),
+    "ravel": lambda a, **kw: a.ravel(),
+})

@@ -803,3 +803,120 @@ def cumsum_axis_op(self, arr, axis, out_dtype):
 
 
 Runtime.cumsum_axis_op = cumsum_axis_op
+
+# ---------------------------------------------------------------------------
+# reshape (reference reshape worker, ramba.py:2409-2492 + frontend 6716/
+# 9438-area).  The reference remaps every element's flat C-order index and
+# ships per-element lists; here both sides' shards are required to cover
+# C-contiguous flat INTERVALS (always true at world 1; multi-rank needs
+# axis-0-only splits spanning full trailing axes), so the movement
+# collapses to interval intersections + the rt_flat_copy gather/scatter
+# kernels.
+# ---------------------------------------------------------------------------
+
+def _flat_interval(box, shape):
+    """(lo, hi_exclusive) flat C-order range of `box` if it is contiguous
+    (spans full extent on every axis but the first), else None."""
+    nd = len(shape)
+    for d in range(1, nd):
+        if not (int(box[0, d]) == 0 and int(box[1, d]) == shape[d] - 1):
+            return None
+    inner = 1
+    for d in range(1, nd):
+        inner *= shape[d]
+    return int(box[0, 0]) * inner, (int(box[1, 0]) + 1) * inner
+
+
+def reshape_op(self, arr, newshape):
+    bd, v = arr.bdarray, arr.view
+    from .shardview import exec_boxes as _eb
+    newshape = tuple(int(x) for x in newshape)
+    total = 1
+    for x in v.shape:
+        total *= x
+    ntotal = 1
+    for x in newshape:
+        ntotal *= x
+    assert total == ntotal, f"cannot reshape {v.shape} into {newshape}"
+
+    from .common import contiguous_divisions
+    out_bd = deferred.bdarray(newshape, bd.dtype,
+                              contiguous_divisions(self.world, newshape),
+                              default_border, flex=False)
+    self.backend.alloc_container(out_bd, self)
+    out_bd.constructed = True
+    if total == 0:
+        return out_bd
+
+    lbs = _eb(v, bd.divisions)
+    src_iv, dst_iv = [], []
+    for r in range(self.world):
+        b = lbs[r]
+        iv = None if b is None else _flat_interval(b, v.shape)
+        if b is not None and iv is None:
+            # the frontend repartitions first; reaching here is a bug
+            raise AssertionError(
+                "reshape_op: non-contiguous source shard "
+                f"{b.tolist()} of {v.shape} (frontend must repartition)")
+        src_iv.append(iv)
+        ob = self.core_box(out_bd, r)
+        oiv = None if ob is None else _flat_interval(ob, newshape)
+        if ob is not None and oiv is None:
+            raise NotImplementedError(
+                "reshape needs C-contiguous destination shards")
+        dst_iv.append(oiv)
+
+    msgs = []   # (dst, src, lo, hi) flat intervals, deterministic
+    for q in range(self.world):
+        if dst_iv[q] is None:
+            continue
+        for r in range(self.world):
+            if src_iv[r] is None:
+                continue
+            lo = max(dst_iv[q][0], src_iv[r][0])
+            hi = min(dst_iv[q][1], src_iv[r][1])
+            if lo < hi:
+                msgs.append((q, r, lo, hi))
+    msgs.sort()
+
+    lb = lbs[self.rank]
+    if lb is not None:
+        d_, _, cstrides, pads = self.shard_geometry(bd)
+        off0, strides = v.operand_addressing(lb[0], cstrides, d_[0], pads)
+        src_shape = box_shape(lb)
+    ob = self.core_box(out_bd, self.rank)
+    if ob is not None:
+        _, _, ocs, opads = self.shard_geometry(out_bd)
+        out_off = sum(opads[i] * ocs[i] for i in range(len(newshape)))
+        dst_shape = box_shape(ob)
+
+    cont_in = self.backend._cont(bd) if lb is not None else None
+    cont_out = self.backend._cont(out_bd) if ob is not None else None
+    sends, recvs, locals_ = [], [], []
+    for (q, r, lo, hi) in msgs:
+        if r == self.rank and q == self.rank:
+            locals_.append((lo, hi))
+            continue
+        if r == self.rank:
+            buf = self.backend.flat_gather(
+                cont_in, off0, strides, src_shape, lo - src_iv[r][0],
+                hi - lo)
+            sends.append((q, buf))
+        if q == self.rank:
+            recvs.append((r, self.backend.new_message_buffer(
+                (hi - lo,), bd.dtype), lo))
+    for (lo, hi) in locals_:
+        buf = self.backend.flat_gather(cont_in, off0, strides, src_shape,
+                                       lo - src_iv[self.rank][0], hi - lo)
+        self.backend.flat_scatter(cont_out, out_off, ocs, dst_shape,
+                                  lo - dst_iv[self.rank][0], buf)
+    if sends or recvs:
+        self.backend.exchange(sends, [(s, b) for (s, b, _) in recvs])
+    for (r, buf, lo) in recvs:
+        self.backend.flat_scatter(cont_out, out_off, ocs, dst_shape,
+                                  lo - dst_iv[self.rank][0], buf)
+    self.backend.free_temps()
+    return out_bd
+
+
+Runtime.reshape_op = reshape_op

@@ -293,3 +293,34 @@ def test_axis_cumsum_spmd():
             import numpy as _np
             return _np.concatenate([o.asarray().reshape(-1) for o in outs])
         """, world=world, tol=1e-12)
+
+
+def test_reshape_spmd():
+    """Distributed reshape: flat-interval intersections + gather/scatter
+    (runtime.reshape_op); surface-split sources repartition through the
+    fused engine first."""
+    for world in (2, 3, 4):
+        run_spmd("""
+            a = np_.arange(840) * 1.0
+            r1 = a.reshape(21, 40)
+            r2 = a.reshape(8, 105)
+            b = np_.fromfunction(lambda i, j: i * 3.0 + j, (281, 3))
+            r3 = b.ravel()
+            r4 = b.reshape(3, 281)
+            # square 2-D: the surface heuristic splits BOTH axes at
+            # world 4 -> exercises the repartition-first path
+            c = np_.fromfunction(lambda i, j: i * 37.0 + j, (36, 37))
+            r5 = c.reshape(37, 36)
+            r6 = c.T.ravel()
+            if np_ is np:
+                return np.concatenate([r1.reshape(-1), r2.reshape(-1),
+                                       r3, r4.reshape(-1), r5.reshape(-1),
+                                       r6])
+            import numpy as _np
+            return _np.concatenate([r1.asarray().reshape(-1),
+                                    r2.asarray().reshape(-1),
+                                    r3.asarray(),
+                                    r4.asarray().reshape(-1),
+                                    r5.asarray().reshape(-1),
+                                    r6.asarray()])
+        """, world=world, tol=0.0)

@@ -442,3 +442,20 @@ def test_matmul_gpu(ra_gpu):
         import numpy as _np
         return _np.concatenate([o.asarray().reshape(-1) for o in outs])
     run_both(impl, ra_gpu, tol=1e-10)
+
+
+def test_reshape_gpu(ra_gpu):
+    """reshape through rt_flat_copy gather/scatter."""
+    def impl(np_):
+        a = np_.arange(1_000_000) * 1.0
+        r1 = a.reshape(1000, 1000)
+        b = np_.fromfunction(lambda i, j: i * 513.0 + j, (512, 513))
+        r2 = b.ravel()
+        r3 = b.T.reshape(513, 512)
+        s1 = r1.sum(axis=0)
+        if np_ is np:
+            return np.concatenate([s1, r2[::997], r3.reshape(-1)[::997]])
+        import numpy as _np
+        return _np.concatenate([s1.asarray(), r2.asarray()[::997],
+                                r3.asarray().reshape(-1)[::997]])
+    run_both(impl, ra_gpu, tol=1e-9)

@@ -937,3 +937,45 @@ class TestMatmulTriuMeshgrid:
         A = ra.fromfunction(lambda i, j: i * 7.0 + j, (4, 5))
         assert A.item(2, 3) == 17.0
         assert A.item((1, 1)) == 8.0
+
+
+class TestReshape:
+    """reshape/ravel/flatten (reference ramba.py:6716 + flat-remap worker
+    2409; always-copy semantics, interval exchange)."""
+
+    def test_basic_shapes(self, ra):
+        a = ra.arange(60) * 1.0
+        na = np.arange(60) * 1.0
+        np.testing.assert_array_equal(a.reshape(6, 10).asarray(),
+                                      na.reshape(6, 10))
+        np.testing.assert_array_equal(a.reshape((5, 4, 3)).asarray(),
+                                      na.reshape(5, 4, 3))
+        np.testing.assert_array_equal(a.reshape(-1, 12).asarray(),
+                                      na.reshape(-1, 12))
+
+    def test_2d_and_views(self, ra):
+        b = ra.fromfunction(lambda i, j: i * 8.0 + j, (7, 8))
+        nb = np.fromfunction(lambda i, j: i * 8.0 + j, (7, 8))
+        np.testing.assert_array_equal(b.ravel().asarray(), nb.ravel())
+        np.testing.assert_array_equal(b.flatten().asarray(), nb.flatten())
+        np.testing.assert_array_equal(b.reshape(4, 14).asarray(),
+                                      nb.reshape(4, 14))
+        np.testing.assert_array_equal(b.T.reshape(56).asarray(),
+                                      nb.T.reshape(56))
+        a = ra.arange(60) * 1.0
+        np.testing.assert_array_equal(a[::2].reshape(5, 6).asarray(),
+                                      np.arange(60)[::2].reshape(5, 6) * 1.0)
+
+    def test_protocol_and_copy_semantics(self, ra):
+        b = ra.fromfunction(lambda i, j: i * 8.0 + j, (7, 8))
+        nb = np.fromfunction(lambda i, j: i * 8.0 + j, (7, 8))
+        np.testing.assert_array_equal(np.reshape(b, (2, 28)).asarray(),
+                                      nb.reshape(2, 28))
+        r = b.reshape(56)
+        b[0:1, 0:1] = -99.0
+        assert float(r[0]) == 0.0  # copy, not a view (reference semantics)
+
+    def test_errors(self, ra):
+        a = ra.arange(10)
+        with pytest.raises(AssertionError):
+            a.reshape(3, 4)
