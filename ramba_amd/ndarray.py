@@ -1215,3 +1215,105 @@ ndarray._ARRAY_FUNC.update({
     "reshape": lambda a, shape, **kw: a.reshape(shape),
     "ravel": lambda a, **kw: a.ravel(),
 })
+
+
+# ---------------------------------------------------------------------------
+# joining / splitting / squeezing (reference ramba.py:9455-9620:
+# concatenate, stack, split, squeeze — vstack/block etc. are unimplemented
+# comments in the reference and stay out of scope here too)
+# ---------------------------------------------------------------------------
+
+def concatenate(arrayseq, axis=0, out=None, **kwargs):
+    """reference concatenate (ramba.py:9541): same-dtype parts written
+    into region views of a fresh array through the fused engine."""
+    assert out is None, "concatenate: out= unsupported"
+    parts = [fromarray(np.asarray(a)) if not isinstance(a, ndarray) else a
+             for a in arrayseq]
+    assert parts, "need at least one array"
+    nd = parts[0].ndim
+    axis = axis + nd if axis < 0 else axis
+    out_shape = list(parts[0].shape)
+    for p in parts[1:]:
+        assert p.ndim == nd, "concatenate: rank mismatch"
+        assert p.dtype == parts[0].dtype, \
+            "concatenate: dtype mismatch (reference asserts equal dtypes)"
+        for i in range(nd):
+            if i == axis:
+                out_shape[i] += p.shape[i]
+            else:
+                assert out_shape[i] == p.shape[i], \
+                    f"concatenate: shape mismatch on axis {i}"
+    res = empty(tuple(out_shape), dtype=parts[0].dtype)
+    pos = 0
+    for p in parts:
+        idx = tuple(slice(pos, pos + p.shape[i]) if i == axis
+                    else slice(None) for i in range(nd))
+        res[idx] = p
+        pos += p.shape[axis]
+    return res
+
+
+def stack(arrays, axis=0, out=None):
+    """reference stack (ramba.py:9581)."""
+    assert out is None
+    parts = [fromarray(np.asarray(a)) if not isinstance(a, ndarray) else a
+             for a in arrays]
+    assert all(p.shape == parts[0].shape for p in parts)
+    nd_out = parts[0].ndim + 1
+    axis = axis + nd_out if axis < 0 else axis
+    return concatenate([expand_dims(p, axis) for p in parts], axis=axis)
+
+
+def split(arr, indices_or_sections, axis=0):
+    """reference split (ramba.py:9611): equal integer sections only,
+    returns slice VIEWS (zero copy)."""
+    ashape = arr.shape
+    nd = len(ashape)
+    if axis > nd:
+        raise ValueError("Wrong axis")
+    if not isinstance(indices_or_sections, numbers.Integral):
+        raise ValueError("split with indices not implemented.")
+    axis_len = ashape[axis]
+    if axis_len % indices_or_sections != 0:
+        raise ValueError(
+            f"Cannot evenly divide array dimension of length {axis_len} "
+            f"into {indices_or_sections} equal sections.")
+    k = axis_len // indices_or_sections
+    return [arr[tuple(slice(None) if y != axis else slice(x * k, (x + 1) * k)
+                      for y in range(nd))]
+            for x in range(indices_or_sections)]
+
+
+def squeeze(a, axis=None):
+    """reference squeeze (ramba.py:9455) — but as a zero-copy VIEW (integer
+    index drops the axis in the affine view algebra)."""
+    if not isinstance(a, ndarray):
+        return np.squeeze(a, axis)
+    ashape = a.shape
+    if axis is None:
+        axis = tuple(i for i in range(len(ashape)) if ashape[i] == 1)
+    if not isinstance(axis, (list, tuple)):
+        axis = (axis,)
+    axis = tuple(x + len(ashape) if x < 0 else x for x in axis)
+    if not all(ashape[x] == 1 for x in axis):
+        raise ValueError("cannot squeeze out an axis with size not equal "
+                         "to 1")
+    idx = tuple(0 if i in axis else slice(None) for i in range(len(ashape)))
+    return a[idx]
+
+
+def dot(a, b, out=None):
+    """reference dot (ramba.py:6933)."""
+    assert out is None
+    return matmul(a, b)
+
+
+ndarray.dot = lambda self, b: matmul(self, b)
+ndarray.squeeze = lambda self, axis=None: squeeze(self, axis)
+
+ndarray._ARRAY_FUNC.update({
+    "concatenate": lambda seq, axis=0, **kw: concatenate(seq, axis=axis),
+    "stack": lambda seq, axis=0, **kw: stack(seq, axis=axis),
+    "split": lambda a, n, axis=0: split(a, n, axis),
+    "squeeze": lambda a, axis=None: squeeze(a, axis),
+})

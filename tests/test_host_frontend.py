@@ -979,3 +979,67 @@ class TestReshape:
         a = ra.arange(10)
         with pytest.raises(AssertionError):
             a.reshape(3, 4)
+
+
+class TestJoinSplit:
+    """concatenate/stack/split/squeeze/dot (reference ramba.py:9455-9620)."""
+
+    def test_concatenate(self, ra):
+        a = ra.arange(10) * 1.0
+        b = ra.arange(6) * 2.0
+        na, nb = np.arange(10) * 1.0, np.arange(6) * 2.0
+        np.testing.assert_array_equal(ra.concatenate([a, b]).asarray(),
+                                      np.concatenate([na, nb]))
+        A = ra.fromfunction(lambda i, j: i * 5.0 + j, (4, 5))
+        B = ra.fromfunction(lambda i, j: i - j * 1.0, (3, 5))
+        nA = np.fromfunction(lambda i, j: i * 5.0 + j, (4, 5))
+        nB = np.fromfunction(lambda i, j: i - j * 1.0, (3, 5))
+        np.testing.assert_array_equal(
+            ra.concatenate([A, B], axis=0).asarray(),
+            np.concatenate([nA, nB], axis=0))
+        C = ra.fromfunction(lambda i, j: i + j * 7.0, (4, 3))
+        nC = np.fromfunction(lambda i, j: i + j * 7.0, (4, 3))
+        np.testing.assert_array_equal(
+            ra.concatenate([A, C], axis=1).asarray(),
+            np.concatenate([nA, nC], axis=1))
+        np.testing.assert_array_equal(
+            np.concatenate([A, B], axis=0).asarray(),
+            np.concatenate([nA, nB], axis=0))
+
+    def test_stack(self, ra):
+        a = ra.arange(8) * 1.0
+        b = ra.arange(8) * 3.0
+        na, nb = np.arange(8) * 1.0, np.arange(8) * 3.0
+        for ax in (0, 1, -1):
+            np.testing.assert_array_equal(
+                ra.stack([a, b], axis=ax).asarray(),
+                np.stack([na, nb], axis=ax))
+
+    def test_split_views(self, ra):
+        A = ra.fromfunction(lambda i, j: i * 6.0 + j, (8, 6))
+        nA = np.fromfunction(lambda i, j: i * 6.0 + j, (8, 6))
+        for got, ref in zip(ra.split(A, 4, axis=0), np.split(nA, 4, axis=0)):
+            np.testing.assert_array_equal(got.asarray(), ref)
+        for got, ref in zip(ra.split(A, 3, axis=1), np.split(nA, 3, axis=1)):
+            np.testing.assert_array_equal(got.asarray(), ref)
+        with pytest.raises(ValueError):
+            ra.split(A, 3, axis=0)
+
+    def test_squeeze_expand_dims(self, ra):
+        A = ra.fromfunction(lambda i, j: i * 3.0 + j, (5, 3))
+        nA = np.fromfunction(lambda i, j: i * 3.0 + j, (5, 3))
+        e = ra.expand_dims(A, 1)
+        assert e.shape == (5, 1, 3)
+        np.testing.assert_array_equal(ra.squeeze(e).asarray(), nA)
+        assert ra.squeeze(e, axis=1).shape == (5, 3)
+        with pytest.raises(ValueError):
+            ra.squeeze(A, axis=0)
+        np.testing.assert_array_equal(np.squeeze(e, axis=1).asarray(), nA)
+
+    def test_dot(self, ra):
+        A = ra.fromfunction(lambda i, j: i * 4.0 + j, (3, 4))
+        v = ra.arange(4) * 1.0
+        nA = np.fromfunction(lambda i, j: i * 4.0 + j, (3, 4))
+        nv = np.arange(4) * 1.0
+        np.testing.assert_allclose(ra.dot(A, v).asarray(), nA @ nv)
+        np.testing.assert_allclose(A.dot(v).asarray(), nA @ nv)
