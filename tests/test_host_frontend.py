@@ -1043,3 +1043,45 @@ class TestJoinSplit:
         nv = np.arange(4) * 1.0
         np.testing.assert_allclose(ra.dot(A, v).asarray(), nA @ nv)
         np.testing.assert_allclose(A.dot(v).asarray(), nA @ nv)
+
+
+class TestSmapSreduce:
+    """smap/smap_index/sreduce (reference ramba.py:9926-9985), supported
+    by tracing the user function through the deferred surface."""
+
+    def test_smap(self, ra):
+        a = ra.arange(50) * 1.0
+        b = ra.arange(50) * 3.0
+        na, nb = np.arange(50) * 1.0, np.arange(50) * 3.0
+        r = ra.smap(lambda x, y: x * 2 + np.float64(1.5) * y, a, b)
+        np.testing.assert_allclose(r.asarray(), na * 2 + 1.5 * nb)
+        # lambda source string, like the reference accepts
+        r2 = ra.smap("lambda x: x * x + 1", a)
+        np.testing.assert_allclose(r2.asarray(), na * na + 1)
+        r3 = ra.smap(lambda x: x + 1, a, dtype=np.float32)
+        assert r3.dtype == np.float32
+
+    def test_smap_index(self, ra):
+        a = ra.fromfunction(lambda i, j: i + j * 1.0, (6, 7))
+        na = np.fromfunction(lambda i, j: i + j * 1.0, (6, 7))
+        r = ra.smap_index(lambda idx, x: idx[0] * 100 + idx[1] * 10 + x, a)
+        ni, nj = np.indices((6, 7))
+        np.testing.assert_allclose(r.asarray(), ni * 100 + nj * 10 + na)
+
+    def test_sreduce(self, ra):
+        a = ra.arange(100) * 1.0
+        na = np.arange(100) * 1.0
+        s = ra.sreduce(lambda x: x * x, lambda x, y: x + y, 0, a)
+        assert abs(float(s) - (na * na).sum()) < 1e-9
+        m = ra.sreduce(lambda x: x + 1, lambda x, y: x if x > y else y,
+                       -np.inf, a)
+        assert float(m) == 100.0
+        with pytest.raises(NotImplementedError):
+            ra.sreduce(lambda x: x, lambda x, y: x - y, 0, a)
+
+    def test_sreduce_index(self, ra):
+        a = ra.arange(20) * 1.0
+        s = ra.sreduce_index(lambda idx, x: idx[0] * x,
+                             lambda x, y: x + y, 0, a)
+        na = np.arange(20) * 1.0
+        assert abs(float(s) - (np.arange(20) * na).sum()) < 1e-9
