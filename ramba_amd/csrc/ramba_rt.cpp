@@ -899,21 +899,36 @@ template <typename T>
 __global__ __launch_bounds__(256) void mask_write_k(
     const T *__restrict__ a, const uint8_t *__restrict__ m,
     T *__restrict__ out, MCArgs g, const int64_t *__restrict__ bexcl) {
-    // thread t owns elements [t*ITEMS, t*ITEMS+ITEMS) of the chunk, so the
-    // intra-chunk exclusive scan of per-thread counts preserves C order
+    // Everything is staged through LDS so HBM sees only coalesced
+    // streams: values+mask in, the block-compacted run out (the naive
+    // per-thread version's global writes were data-dependent partial
+    // lines: measured 0.58 TB/s).  Thread t owns elements
+    // [t*ITEMS, t*ITEMS+ITEMS) of the chunk, so the intra-chunk
+    // exclusive scan of per-thread counts preserves C order.
+#define LIDX(x) ((x) + ((x) >> 4))
+    __shared__ T vals[SCAN_CHUNK + SCAN_THREADS];
+    __shared__ T outbuf[SCAN_CHUNK];
+    __shared__ unsigned char msk[SCAN_CHUNK + SCAN_THREADS];
     __shared__ int64_t wsum[4];
     int64_t blk = blockIdx.x;
     int64_t nblocks = gridDim.x;
     for (; blk * SCAN_CHUNK < g.n; blk += nblocks) {
         int64_t b0 = blk * SCAN_CHUNK;
-        int64_t l0 = b0 + (int64_t)threadIdx.x * SCAN_ITEMS;
-        unsigned char sel[SCAN_ITEMS];
-        int64_t s = 0;
         for (int j = 0; j < SCAN_ITEMS; ++j) {
-            int64_t i = l0 + j;
-            sel[j] = (i < g.n) && m[mc_addr(g, i, g.mstr)] != 0;
-            s += sel[j];
+            int64_t k = j * SCAN_THREADS + threadIdx.x;
+            int64_t i = b0 + k;
+            if (i < g.n) {
+                msk[LIDX(k)] = m[mc_addr(g, i, g.mstr)];
+                vals[LIDX(k)] = a[mc_addr(g, i, g.astr)];
+            } else {
+                msk[LIDX(k)] = 0;
+            }
         }
+        __syncthreads();
+        int64_t l0 = (int64_t)threadIdx.x * SCAN_ITEMS;
+        int64_t s = 0;
+        for (int j = 0; j < SCAN_ITEMS; ++j)
+            s += msk[LIDX(l0 + j)] != 0;
         int64_t x = s;
         int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
         for (int off = 1; off < 64; off <<= 1) {
@@ -924,11 +939,18 @@ __global__ __launch_bounds__(256) void mask_write_k(
         __syncthreads();
         int64_t wbase = 0;
         for (int w = 0; w < wid; ++w) wbase += wsum[w];
-        int64_t pos = bexcl[blk] + wbase + x - s;
+        int64_t pos = wbase + x - s;        // block-local compacted index
         for (int j = 0; j < SCAN_ITEMS; ++j)
-            if (sel[j]) out[pos++] = a[mc_addr(g, l0 + j, g.astr)];
+            if (msk[LIDX(l0 + j)] != 0) outbuf[pos++] = vals[LIDX(l0 + j)];
+        int64_t btotal = 0;
+        for (int w = 0; w < 4; ++w) btotal += wsum[w];
+        __syncthreads();
+        int64_t obase = bexcl[blk];
+        for (int64_t k = threadIdx.x; k < btotal; k += SCAN_THREADS)
+            out[obase + k] = outbuf[k];
         __syncthreads();
     }
+#undef LIDX
 }
 
 template <typename T>
@@ -1093,6 +1115,107 @@ __global__ __launch_bounds__(256) void axis_scan_waves_k(
     }
 }
 
+// chunked variant for axis != last with FEW lines: nlines threads alone
+// starve the chip (16384 columns = 64 workgroups: measured 0.17 TB/s), so
+// the scan axis is split into `nchunks` ranges — k1 scans each (line,
+// chunk) range locally and records its sum, k2 turns the per-chunk sums
+// into exclusive offsets per line (and writes the final line totals), k3
+// adds the offsets back.  3 passes over out ≈ 32 B/elem vs 16, but the
+// chip is full.
+template <typename T>
+__global__ __launch_bounds__(256) void axis_scan_ck1(
+    const T *__restrict__ in, T *__restrict__ out, T *__restrict__ tot2,
+    ASArgs g, int64_t nchunks, int64_t clen) {
+    int64_t w = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    int64_t work = g.nlines * nchunks;
+    for (; w < work; w += stride) {
+        int64_t line = w % g.nlines;
+        int64_t c = w / g.nlines;
+        int64_t k0 = c * clen;
+        int64_t k1 = k0 + clen < g.len ? k0 + clen : g.len;
+        int64_t ioff, ooff;
+        as_addr(g, line, ioff, ooff);
+        T acc = (T)0;
+        for (int64_t k = k0; k < k1; ++k) {
+            acc += in[ioff + k * g.istr_ax];
+            out[ooff + k * g.ostr_ax] = acc;
+        }
+        tot2[c * g.nlines + line] = acc;
+    }
+}
+
+template <typename T>
+__global__ __launch_bounds__(256) void axis_scan_ck2(
+    T *__restrict__ tot2, T *__restrict__ totals, int64_t nlines,
+    int64_t nchunks) {
+    int64_t line = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; line < nlines; line += stride) {
+        T run = (T)0;
+        for (int64_t c = 0; c < nchunks; ++c) {
+            T t = tot2[c * nlines + line];
+            tot2[c * nlines + line] = run;
+            run += t;
+        }
+        totals[line] = run;
+    }
+}
+
+template <typename T>
+__global__ __launch_bounds__(256) void axis_scan_ck3(
+    T *__restrict__ out, const T *__restrict__ tot2, ASArgs g,
+    int64_t nchunks, int64_t clen) {
+    // chunks 1.. only; chunk 0's offset is zero
+    int64_t w = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    int64_t work = g.nlines * (nchunks - 1);
+    for (; w < work; w += stride) {
+        int64_t line = w % g.nlines;
+        int64_t c = 1 + w / g.nlines;
+        T off = tot2[c * g.nlines + line];
+        int64_t k0 = c * clen;
+        int64_t k1 = k0 + clen < g.len ? k0 + clen : g.len;
+        int64_t ioff, ooff;
+        as_addr(g, line, ioff, ooff);
+        for (int64_t k = k0; k < k1; ++k)
+            out[ooff + k * g.ostr_ax] += off;
+    }
+}
+
+template <typename T>
+int axis_scan_chunked_launch(uintptr_t stream, const void *in, void *out,
+                             void *totals, void *tot2, const ASArgs &g,
+                             int64_t nchunks) {
+    hipStream_t st = reinterpret_cast<hipStream_t>(stream);
+    int64_t clen = (g.len + nchunks - 1) / nchunks;
+    int64_t work = g.nlines * nchunks;
+    int64_t blocks = (work + 255) / 256;
+    if (blocks > 4096) blocks = 4096;
+    if (blocks < 1) blocks = 1;
+    hipLaunchKernelGGL((axis_scan_ck1<T>), dim3((unsigned)blocks), dim3(256),
+                       0, st, static_cast<const T *>(in),
+                       static_cast<T *>(out), static_cast<T *>(tot2), g,
+                       nchunks, clen);
+    int64_t b2 = (g.nlines + 255) / 256;
+    if (b2 > 4096) b2 = 4096;
+    if (b2 < 1) b2 = 1;
+    hipLaunchKernelGGL((axis_scan_ck2<T>), dim3((unsigned)b2), dim3(256), 0,
+                       st, static_cast<T *>(tot2), static_cast<T *>(totals),
+                       g.nlines, nchunks);
+    if (nchunks > 1)
+        hipLaunchKernelGGL((axis_scan_ck3<T>), dim3((unsigned)blocks),
+                           dim3(256), 0, st, static_cast<T *>(out),
+                           static_cast<const T *>(tot2), g, nchunks, clen);
+    hipError_t e = hipGetLastError();
+    if (e != hipSuccess) {
+        set_error(std::string("axis_scan chunked launch: ")
+                  + hipGetErrorString(e));
+        return 1;
+    }
+    return 0;
+}
+
 template <typename T>
 int axis_scan_launch(uintptr_t stream, const void *in, void *out,
                      void *totals, const ASArgs &g, int waves) {
@@ -1123,12 +1246,14 @@ int axis_scan_launch(uintptr_t stream, const void *in, void *out,
 
 // in/out pre-offset to the box origin; shape/strides over the FULL local
 // box (nd dims, nd 2..4); totals: dense C-order slab of the line space.
-// dtype: 0=f64 1=f32 2=i64 3=i32.
+// dtype: 0=f64 1=f32 2=i64 3=i32.  nchunks > 1 selects the chunked
+// variant (axis != last with few lines); tot2 then holds
+// nchunks × nlines scratch.
 extern "C" int rt_axis_scan(uintptr_t stream, const void *in, void *out,
                             void *totals, int nd, const int64_t *shape,
                             const int64_t *in_strides,
                             const int64_t *out_strides, int axis,
-                            int dtype) {
+                            int dtype, void *tot2, int64_t nchunks) {
     if (nd < 2 || nd > 4 || axis < 0 || axis >= nd) {
         set_error("rt_axis_scan: bad nd/axis");
         return 1;
@@ -1153,6 +1278,21 @@ extern "C" int rt_axis_scan(uintptr_t stream, const void *in, void *out,
         g.ostr[d] = 0;
     }
     if (g.nlines == 0 || g.len == 0) return 0;
+    if (nchunks > 1 && tot2) {
+        switch (dtype) {
+            case 0: return axis_scan_chunked_launch<double>(
+                stream, in, out, totals, tot2, g, nchunks);
+            case 1: return axis_scan_chunked_launch<float>(
+                stream, in, out, totals, tot2, g, nchunks);
+            case 2: return axis_scan_chunked_launch<int64_t>(
+                stream, in, out, totals, tot2, g, nchunks);
+            case 3: return axis_scan_chunked_launch<int32_t>(
+                stream, in, out, totals, tot2, g, nchunks);
+            default:
+                set_error("rt_axis_scan: bad dtype");
+                return 1;
+        }
+    }
     int waves = (axis == nd - 1) ? 1 : 0;
     switch (dtype) {
         case 0: return axis_scan_launch<double>(stream, in, out, totals, g,
@@ -1186,38 +1326,104 @@ struct FlatArgs {
     int nd;
 };
 
+// row-wise mapping: one div/mod chain per ROW per block instead of per
+// element (int64 division is ~100 cycles on CDNA4 and dominated the
+// naive per-element version: measured 0.40 TB/s -> row-wise streams).
+// A "row" is one span of the last axis; threads sweep it coalesced.
 template <typename T, int SCATTER>
-__global__ __launch_bounds__(256) void flat_copy_kernel(
-    T *__restrict__ boxed, T *__restrict__ dense, FlatArgs g) {
+__global__ __launch_bounds__(256) void flat_copy_1d_kernel(
+    T *__restrict__ boxed, T *__restrict__ dense, int64_t flat0,
+    int64_t n, int64_t str) {
     int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    for (; i < g.n; i += stride) {
-        int64_t rem = g.flat0 + i, off = 0;
-        for (int d = g.nd - 1; d >= 0; --d) {
+    for (; i < n; i += stride) {
+        if (SCATTER) boxed[(flat0 + i) * str] = dense[i];
+        else dense[i] = boxed[(flat0 + i) * str];
+    }
+}
+
+template <typename T, int SCATTER>
+__global__ __launch_bounds__(256) void flat_copy_kernel(
+    T *__restrict__ boxed, T *__restrict__ dense, FlatArgs g,
+    int64_t cpr /* column chunks per row (fills the chip when rows are
+                   few and the inner axis is huge) */) {
+    const int64_t inner = g.shape[g.nd - 1];
+    const int64_t sin_ = g.str[g.nd - 1];
+    const int64_t row_lo = g.flat0 / inner;
+    const int64_t row_hi = (g.flat0 + g.n - 1) / inner;   // inclusive
+    const int64_t span = (inner + cpr - 1) / cpr;
+    for (int64_t b = blockIdx.x;; b += gridDim.x) {
+        int64_t row = row_lo + b / cpr;
+        if (row > row_hi) break;
+        int64_t chunk = b % cpr;
+        // outer-coordinate offset for this row (one divmod chain)
+        int64_t rem = row, off = 0;
+        for (int d = g.nd - 2; d >= 0; --d) {
             int64_t idx = rem % g.shape[d];
             rem /= g.shape[d];
             off += idx * g.str[d];
         }
-        if (SCATTER) boxed[off] = dense[i];
-        else dense[i] = boxed[off];
+        int64_t fbase = row * inner;
+        int64_t c0 = fbase < g.flat0 ? g.flat0 - fbase : 0;
+        int64_t c1 = fbase + inner > g.flat0 + g.n ? g.flat0 + g.n - fbase
+                                                   : inner;
+        if (chunk * span > c0) c0 = chunk * span;
+        if ((chunk + 1) * span < c1) c1 = (chunk + 1) * span;
+        for (int64_t c = c0 + threadIdx.x; c < c1; c += blockDim.x) {
+            int64_t di = fbase + c - g.flat0;
+            if (SCATTER) boxed[off + c * sin_] = dense[di];
+            else dense[di] = boxed[off + c * sin_];
+        }
     }
 }
 
 template <typename T>
 int flat_copy_launch(uintptr_t stream, void *boxed, void *dense,
                      const FlatArgs &g, int scatter) {
-    int64_t blocks = (g.n + 255) / 256;
-    if (blocks > 4096) blocks = 4096;
-    if (blocks < 1) blocks = 1;
     hipStream_t st = reinterpret_cast<hipStream_t>(stream);
+    if (g.nd == 1) {
+        int64_t blocks = (g.n + 255) / 256;
+        if (blocks > 4096) blocks = 4096;
+        if (blocks < 1) blocks = 1;
+        if (scatter)
+            hipLaunchKernelGGL((flat_copy_1d_kernel<T, 1>),
+                               dim3((unsigned)blocks), dim3(256), 0, st,
+                               static_cast<T *>(boxed),
+                               static_cast<T *>(dense), g.flat0, g.n,
+                               g.str[0]);
+        else
+            hipLaunchKernelGGL((flat_copy_1d_kernel<T, 0>),
+                               dim3((unsigned)blocks), dim3(256), 0, st,
+                               static_cast<T *>(boxed),
+                               static_cast<T *>(dense), g.flat0, g.n,
+                               g.str[0]);
+        hipError_t e1 = hipGetLastError();
+        if (e1 != hipSuccess) {
+            set_error(std::string("flat_copy launch: ")
+                      + hipGetErrorString(e1));
+            return 1;
+        }
+        return 0;
+    }
+    int64_t inner = g.shape[g.nd - 1];
+    int64_t rows = (g.n + inner - 1) / inner + 1;
+    int64_t cpr = 1;
+    if (rows < 2048) {
+        cpr = 2048 / rows;
+        int64_t maxc = (inner + 65535) / 65536;
+        if (cpr > maxc) cpr = maxc;
+        if (cpr < 1) cpr = 1;
+    }
+    int64_t blocks = rows * cpr < 4096 ? rows * cpr : 4096;
+    if (blocks < 1) blocks = 1;
     if (scatter)
         hipLaunchKernelGGL((flat_copy_kernel<T, 1>), dim3((unsigned)blocks),
                            dim3(256), 0, st, static_cast<T *>(boxed),
-                           static_cast<T *>(dense), g);
+                           static_cast<T *>(dense), g, cpr);
     else
         hipLaunchKernelGGL((flat_copy_kernel<T, 0>), dim3((unsigned)blocks),
                            dim3(256), 0, st, static_cast<T *>(boxed),
-                           static_cast<T *>(dense), g);
+                           static_cast<T *>(dense), g, cpr);
     hipError_t e = hipGetLastError();
     if (e != hipSuccess) {
         set_error(std::string("flat_copy launch: ") + hipGetErrorString(e));

@@ -82,7 +82,8 @@ def _load_lib():
                                  ctypes.POINTER(ctypes.c_int64),
                                  ctypes.POINTER(ctypes.c_int64),
                                  ctypes.POINTER(ctypes.c_int64),
-                                 ctypes.c_int, ctypes.c_int]
+                                 ctypes.c_int, ctypes.c_int,
+                                 ctypes.c_void_p, ctypes.c_int64]
     lib.rt_flat_copy.argtypes = [ctypes.c_size_t, ctypes.c_void_p,
                                  ctypes.c_void_p, ctypes.c_int,
                                  ctypes.POINTER(ctypes.c_int64),
@@ -758,10 +759,27 @@ def _hb_axis_scan_local(self, bd_in, off0, strides, lshape, axis, out_bd,
     self.temps["__axcs_tot__"] = tot
     in_ptr = ca.data_ptr() + off0 * ca.element_size()
     out_ptr = co.data_ptr() + out_off * co.element_size()
+    # axis != last with few lines: single thread-per-line starves the
+    # chip -> chunked 3-pass variant (k1 local / k2 offsets / k3 apply)
+    nlines = 1
+    for d in range(nd):
+        if d != axis:
+            nlines *= lshape[d]
+    length = lshape[axis]
+    nchunks, tot2_ptr = 1, None
+    if axis != nd - 1 and nlines < 262144 and length >= 64:
+        # enough chunks to fill the chip, but keep >=32 elems per chunk
+        nchunks = min(-(-524288 // max(1, nlines)), -(-length // 32))
+        if nchunks > 1:
+            tot2 = self.torch.empty(nchunks * nlines, dtype=co.dtype,
+                                    device="cuda")
+            self.temps["__axcs_tot2__"] = tot2
+            tot2_ptr = ctypes.c_void_p(tot2.data_ptr())
     rc = self.lib.rt_axis_scan(
         self._stream(), ctypes.c_void_p(in_ptr), ctypes.c_void_p(out_ptr),
         ctypes.c_void_p(tot.data_ptr()), nd, _i64arr(lshape),
-        _i64arr(strides), _i64arr(out_strides), axis, dt)
+        _i64arr(strides), _i64arr(out_strides), axis, dt,
+        tot2_ptr, nchunks if tot2_ptr else 1)
     self._check(rc, "rt_axis_scan")
 
 
