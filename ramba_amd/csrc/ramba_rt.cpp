@@ -855,6 +855,9 @@ extern "C" int rt_cumsum(uintptr_t stream, const void *in, int64_t in_off,
 
 namespace {
 
+constexpr int MC_ITEMS = 8;                  // 2048-elem chunks: ~34 KB
+constexpr int MC_CHUNK = 256 * MC_ITEMS;     // LDS -> 4 blocks/CU
+
 struct MCArgs {
     int64_t n;
     int64_t shape[4];
@@ -880,10 +883,10 @@ __global__ __launch_bounds__(256) void mask_count_k(
     int64_t blk = blockIdx.x;
     int64_t nblocks = gridDim.x;
     __shared__ int64_t lds[4];
-    for (; blk * SCAN_CHUNK < g.n; blk += nblocks) {
-        int64_t base = blk * SCAN_CHUNK;
+    for (; blk * MC_CHUNK < g.n; blk += nblocks) {
+        int64_t base = blk * MC_CHUNK;
         int64_t acc = 0;
-        for (int j = 0; j < SCAN_ITEMS; ++j) {
+        for (int j = 0; j < MC_ITEMS; ++j) {
             int64_t i = base + j * SCAN_THREADS + threadIdx.x;
             if (i < g.n && m[mc_addr(g, i, g.mstr)] != 0) ++acc;
         }
@@ -907,15 +910,15 @@ __global__ __launch_bounds__(256) void mask_write_k(
     // [t*ITEMS, t*ITEMS+ITEMS) of the chunk, so the intra-chunk
     // exclusive scan of per-thread counts preserves C order.
 #define LIDX(x) ((x) + ((x) >> 4))
-    __shared__ T vals[SCAN_CHUNK + SCAN_THREADS];
-    __shared__ T outbuf[SCAN_CHUNK];
-    __shared__ unsigned char msk[SCAN_CHUNK + SCAN_THREADS];
+    __shared__ T vals[MC_CHUNK + SCAN_THREADS];
+    __shared__ T outbuf[MC_CHUNK];
+    __shared__ unsigned char msk[MC_CHUNK + SCAN_THREADS];
     __shared__ int64_t wsum[4];
     int64_t blk = blockIdx.x;
     int64_t nblocks = gridDim.x;
-    for (; blk * SCAN_CHUNK < g.n; blk += nblocks) {
-        int64_t b0 = blk * SCAN_CHUNK;
-        for (int j = 0; j < SCAN_ITEMS; ++j) {
+    for (; blk * MC_CHUNK < g.n; blk += nblocks) {
+        int64_t b0 = blk * MC_CHUNK;
+        for (int j = 0; j < MC_ITEMS; ++j) {
             int64_t k = j * SCAN_THREADS + threadIdx.x;
             int64_t i = b0 + k;
             if (i < g.n) {
@@ -926,9 +929,9 @@ __global__ __launch_bounds__(256) void mask_write_k(
             }
         }
         __syncthreads();
-        int64_t l0 = (int64_t)threadIdx.x * SCAN_ITEMS;
+        int64_t l0 = (int64_t)threadIdx.x * MC_ITEMS;
         int64_t s = 0;
-        for (int j = 0; j < SCAN_ITEMS; ++j)
+        for (int j = 0; j < MC_ITEMS; ++j)
             s += msk[LIDX(l0 + j)] != 0;
         int64_t x = s;
         int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
@@ -941,7 +944,7 @@ __global__ __launch_bounds__(256) void mask_write_k(
         int64_t wbase = 0;
         for (int w = 0; w < wid; ++w) wbase += wsum[w];
         int64_t pos = wbase + x - s;        // block-local compacted index
-        for (int j = 0; j < SCAN_ITEMS; ++j)
+        for (int j = 0; j < MC_ITEMS; ++j)
             if (msk[LIDX(l0 + j)] != 0) outbuf[pos++] = vals[LIDX(l0 + j)];
         int64_t btotal = 0;
         for (int w = 0; w < 4; ++w) btotal += wsum[w];

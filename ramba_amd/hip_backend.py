@@ -706,7 +706,8 @@ def _hb_mask_compact(self, bd_a, bd_m, rt):
     a_ptr = ca.data_ptr() + off_a * ca.element_size()
     m_ptr = cm.data_ptr() + off_m * cm.element_size()
     dtc = _MC_DT[str(np.dtype(bd_a.dtype))]
-    nchunks = max(1, (n + _SCAN_CHUNK - 1) // _SCAN_CHUNK)
+    _MC_CHUNK = 2048     # must match MC_CHUNK in ramba_rt.cpp
+    nchunks = max(1, (n + _MC_CHUNK - 1) // _MC_CHUNK)
     bcounts = self.torch.empty(nchunks, dtype=self.torch.int64,
                                device="cuda")
     total = self.torch.empty(1, dtype=self.torch.int64, device="cuda")
