@@ -1398,3 +1398,14 @@ def sreduce_index(func, reducer, identity, *args, parallel=True):
     mapped = func(idx, *args)
     return getattr(mapped, {"sum": "sum", "prod": "prod", "min": "min",
                             "max": "max"}[kind])()
+
+
+# small numpy-compatible accessors
+def _nd_tolist(self):
+    return self.asarray().tolist()
+
+
+ndarray.tolist = _nd_tolist
+ndarray.itemsize = property(lambda self: np.dtype(self.dtype).itemsize)
+ndarray.nbytes = property(
+    lambda self: self.size * np.dtype(self.dtype).itemsize)

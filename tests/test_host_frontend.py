@@ -1085,3 +1085,40 @@ class TestSmapSreduce:
                              lambda x, y: x + y, 0, a)
         na = np.arange(20) * 1.0
         assert abs(float(s) - (np.arange(20) * na).sum()) < 1e-9
+
+
+class TestNewOpDtypeMatrix:
+    """dtype coverage for the round-1 extension ops (mask-getitem, axis
+    cumsum, reshape) beyond the default fp64/int64 cases."""
+
+    @pytest.mark.parametrize("dt", [np.float32, np.int32, np.int64,
+                                    np.uint8, np.int8])
+    def test_mask_getitem_dtypes(self, ra, dt):
+        a = (ra.arange(200) % 120).astype(dt)
+        na = (np.arange(200) % 120).astype(dt)
+        sel = a[(a % 3).astype(np.int64) == 0]
+        nsel = na[(na % 3).astype(np.int64) == 0]
+        np.testing.assert_array_equal(sel.asarray(), nsel)
+        assert sel.dtype == nsel.dtype
+
+    @pytest.mark.parametrize("dt", [np.float32, np.int32, np.int64])
+    def test_axis_cumsum_dtypes(self, ra, dt):
+        a = ra.fromfunction(lambda i, j: i + j, (11, 9), dtype=dt)
+        na = np.fromfunction(lambda i, j: i + j, (11, 9)).astype(dt)
+        for ax in (0, 1):
+            g = a.cumsum(axis=ax)
+            n = na.cumsum(axis=ax)
+            assert g.dtype == n.dtype, (g.dtype, n.dtype)
+            np.testing.assert_allclose(g.asarray(), n, rtol=1e-6)
+
+    @pytest.mark.parametrize("dt", [np.float32, np.int16, np.uint8])
+    def test_reshape_dtypes(self, ra, dt):
+        a = (ra.arange(120) % 100).astype(dt)
+        na = (np.arange(120) % 100).astype(dt)
+        np.testing.assert_array_equal(a.reshape(10, 12).asarray(),
+                                      na.reshape(10, 12))
+
+    def test_tolist_nbytes(self, ra):
+        a = ra.arange(6)
+        assert a.tolist() == [0, 1, 2, 3, 4, 5]
+        assert a.nbytes == 48 and a.itemsize == 8
