@@ -130,8 +130,12 @@ def main():
             A[1:-1, 1:-1] = (ssin[:-2, 1:-1] + ssin[2:, 1:-1]
                              + ssin[1:-1, :-2] + ssin[1:-1, 2:]
                              - 4.0 * ssin[1:-1, 1:-1])
-            s = A.sum()
-            return s
+            # force the reduction like the reference does (its sum is
+            # eager: internal_reduction2b gathers immediately).  Leaving
+            # the scalar pending would also let the LAST step's group
+            # escape the timed region (host-side flush happens after the
+            # barrier) and triggers the WAR temp-dance on A every step.
+            return float(A.sum())
         metric = ("GElem/s, 8192^2 fp64 iota→sin→stencil→sum pipeline "
                   "(BASELINE configs[4])")
         wl_desc = ("8192^2 fp64: fused iota+sin fill, 5-pt stencil, global "
