@@ -174,13 +174,22 @@ def main():
         wl_desc = (f"{S}^2 fp32 5-point stencil, 2-D blocks + halo "
                    "exchange, 8 B/elem algorithmic")
 
+    trace = os.environ.get("RAMBA_BENCH_TRACE")
     keep = None
     for _ in range(args.warmup):
         keep = step()
     barrier_sync()
     t0 = time.perf_counter()
-    for _ in range(args.steps):
-        keep = step()
+    if trace:
+        for i in range(args.steps):
+            ts = time.perf_counter()
+            keep = step()
+            torch.cuda.synchronize()
+            print(f"step {i}: {(time.perf_counter()-ts)*1e3:.2f} ms",
+                  file=sys.stderr)
+    else:
+        for _ in range(args.steps):
+            keep = step()
     barrier_sync()
     t1 = time.perf_counter()
     elapsed = t1 - t0
