@@ -176,8 +176,19 @@ def main():
 
     trace = os.environ.get("RAMBA_BENCH_TRACE")
     keep = None
-    for _ in range(args.warmup):
+    # the box exhibits ONE ~36 ms runtime stall per process, 3-5 s after
+    # start (observed at arbitrary step indices; independent of GC and of
+    # our kernels — gpurun_out/s1err.log).  Warm up for at least 6 s of
+    # wall time so it lands before the timed window; the JSON reports the
+    # warmup steps actually done.
+    warm_done = 0
+    tw = time.perf_counter()
+    while warm_done < args.warmup or time.perf_counter() - tw < 6.0:
         keep = step()
+        warm_done += 1
+        if warm_done > args.warmup + 100_000:
+            break
+    args.warmup = warm_done
     barrier_sync()
     t0 = time.perf_counter()
     if trace:
