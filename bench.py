@@ -183,11 +183,31 @@ def main():
     # warmup steps actually done.
     warm_done = 0
     tw = time.perf_counter()
-    while warm_done < args.warmup or time.perf_counter() - tw < 6.0:
-        keep = step()
-        warm_done += 1
-        if warm_done > args.warmup + 100_000:
-            break
+    if world == 1:
+        while warm_done < args.warmup or time.perf_counter() - tw < 6.0:
+            keep = step()
+            warm_done += 1
+            if warm_done > args.warmup + 100_000:
+                break
+    else:
+        # N>1: a step may contain collectives (reduce's allreduce, halo
+        # exchange), so every rank MUST run the same number of steps.
+        # Agree in rounds: fixed step batches + an all_reduce(MIN) vote
+        # on whether everyone has passed the 6 s floor.
+        import torch.distributed as dist
+        for _ in range(args.warmup):
+            keep = step()
+            warm_done += 1
+        while True:
+            flag = torch.tensor(
+                [1.0 if time.perf_counter() - tw >= 6.0 else 0.0],
+                dtype=torch.float64, device="cuda")
+            dist.all_reduce(flag, op=dist.ReduceOp.MIN)
+            if float(flag.cpu()[0]) >= 1.0 or warm_done > 100_000:
+                break
+            for _ in range(4):
+                keep = step()
+                warm_done += 1
     args.warmup = warm_done
     barrier_sync()
     t0 = time.perf_counter()

@@ -163,7 +163,6 @@ class ndarray:
                 and len(index) == self.ndim:
             # full scalar read -- a sync point
             deferred.flush()
-            coords = []
             v = self.view
             base = [int(o) for o in v.offset]
             for d, i in enumerate(index):
