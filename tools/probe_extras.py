@@ -12,7 +12,7 @@ import ramba_amd as ra  # noqa: E402
 from ramba_amd import deferred  # noqa: E402
 
 
-def t(label, fn, nbytes, iters=5):
+def t(label, fn, nbytes, iters=3):
     import torch
     fn()
     torch.cuda.synchronize()
@@ -25,11 +25,12 @@ def t(label, fn, nbytes, iters=5):
 
 
 def main():
+    small = "--small" in sys.argv   # profiled runs: keep rocprof cheap
     ra.init()
     rt = deferred.get_runtime()
     be = rt.backend
 
-    n = 16384
+    n = 8192 if small else 16384
     A = ra.fromfunction(lambda i, j: (i * 3 + j) * 1e-9, (n, n))
     ra.sync()
     out = ra.zeros((n, n))
@@ -57,7 +58,7 @@ def main():
         be._cont(obd), ooff, ocs, (n, n), 0, dense), nn * 8 * 2)
 
     # raw mask compaction on prebuilt co-partitioned pair (50%)
-    m = 500_000_000
+    m = 100_000_000 if small else 500_000_000
     V = ra.arange(m) * 1.0
     M8 = (ra.arange(m) % 2).astype(np.uint8)
     ra.sync()
