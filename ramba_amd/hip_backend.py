@@ -349,6 +349,9 @@ class HipBackend:
         return gk
 
     def launch(self, plan, recipe=None):
+        from .common import ntiming, add_time
+        import time as _time
+        _t0 = _time.perf_counter() if ntiming else 0.0
         if recipe is not None and recipe.backend_kernel is not None:
             gk = recipe.backend_kernel
         else:
@@ -397,6 +400,9 @@ class HipBackend:
             return self._cont(op.bd).data_ptr()
 
         args = codegen.pack_args(gk, plan, ptr_of)
+        if ntiming:
+            add_time("hb_prep", _time.perf_counter() - _t0)
+            _t0 = _time.perf_counter()
         stream = self._stream()
         if self.time_kernels:
             ev0 = self.torch.cuda.Event(enable_timing=True)
@@ -420,6 +426,8 @@ class HipBackend:
             for spec, o in zip(plan.reductions, outs):
                 v = o.cpu().numpy()[0]
                 results.append(np.asarray(v, dtype=spec.dtype)[()])
+        if ntiming:
+            add_time("hb_submit", _time.perf_counter() - _t0)
         return results
 
     def sync(self):

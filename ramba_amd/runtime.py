@@ -173,6 +173,8 @@ class Runtime:
             add_time("run_deferred_ops", time.perf_counter() - t0)
 
     def _execute_group(self, group):
+        from .common import ntiming
+        t0 = time.perf_counter() if ntiming else 0.0
         live, dead = deferred.compute_live_vars(group)
         cache = getattr(self, "_recipe_cache", None)
         if cache is None:
@@ -181,6 +183,8 @@ class Runtime:
             sig = self._group_signature(group, live)
         except TypeError:
             sig = None
+        if ntiming:
+            add_time("eg_signature", time.perf_counter() - t0)
         recipe = cache.get(sig) if sig is not None else None
         if recipe is not None:
             self._run_recipe(recipe, group, live)
@@ -219,12 +223,18 @@ class Runtime:
             msgs = [(dst, src, live[owner].bd, bx, tgt)
                     for (dst, src, owner, bx, tgt) in recipe.comm_msgs]
             self._do_comms(msgs, recipe.temp_geom)
+        from .common import ntiming
+        tl = time.perf_counter() if ntiming else 0.0
         if plan.itershape is not None:
             partials = self.backend.launch(plan, recipe)
+            if ntiming:
+                add_time("eg_launch_host", time.perf_counter() - tl)
         else:
             partials = [np.asarray(ir.reduction_init(spec.kind, spec.dtype),
                                    dtype=spec.dtype)[()]
                         for spec in plan.reductions]
+            if ntiming:
+                add_time("eg_launch_host", time.perf_counter() - tl)
         for (spec, pend), val in zip(group.reductions, partials):
             pend.partial = np.asarray(val, dtype=spec.dtype)[()]
         self.backend.free_temps()
