@@ -210,12 +210,26 @@ def compute_regular_schedule(num_workers, size, dims_do_not_distribute=()):
     return divisions
 
 
+_div_cache = {}
+
+
 def default_divisions(num_workers, shape):
     """Divisions for a fresh array (reference: shardview_array.py:908
-    default_distribution -> compute_regular_schedule)."""
+    default_distribution -> compute_regular_schedule).  Cached and shared
+    per (world, shape) — divisions are never mutated after construction
+    (flex adoption REBINDS bd.divisions), and the cached array is marked
+    read-only to enforce that."""
     if len(shape) == 0:
         raise ValueError("0-d arrays are not distributed")
-    return compute_regular_schedule(num_workers, tuple(int(s) for s in shape))
+    key = (num_workers, tuple(int(s) for s in shape))
+    d = _div_cache.get(key)
+    if d is None:
+        d = compute_regular_schedule(num_workers, key[1])
+        d.flags.writeable = False
+        if len(_div_cache) > 4096:
+            _div_cache.clear()
+        _div_cache[key] = d
+    return d
 
 
 def contiguous_divisions(world, shape):

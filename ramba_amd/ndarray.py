@@ -28,14 +28,29 @@ def _dtype_of(x):
     return x  # python scalar: weak promotion via np.result_type
 
 
+_rdt_cache = {}
+
+
 def _result_dtype(op, a, b):
     if op in ir.BOOL_RESULT_BINOPS:
         return ir.BOOL
     da, db = _dtype_of(a), _dtype_of(b)
-    rt = np.result_type(da, db)
-    if op == "div":
-        if rt.kind in "bui":
+    # NEP-50: python scalar TYPES are weak (value-independent), so the
+    # result is cacheable on (op, kinds)
+    ka = da if isinstance(da, np.dtype) else type(da)
+    kb = db if isinstance(db, np.dtype) else type(db)
+    try:
+        key = (op == "div", ka, kb)
+        rt = _rdt_cache.get(key)
+    except TypeError:
+        rt = None
+        key = None
+    if rt is None:
+        rt = np.result_type(da, db)
+        if op == "div" and rt.kind in "bui":
             rt = np.dtype(np.float64)
+        if key is not None:
+            _rdt_cache[key] = rt
     return rt
 
 
@@ -56,6 +71,10 @@ def _bcast(x, shape):
 # ---------------------------------------------------------------------------
 # ndarray
 # ---------------------------------------------------------------------------
+
+# hot-path caches: iterating workloads re-execute the same source lines,
+# so slice-view composition and dtype resolution repeat exactly
+_getitem_cache = {}
 
 class ndarray:
     __slots__ = ("bdarray", "view", "readonly", "__weakref__")
@@ -174,7 +193,16 @@ class ndarray:
                     base[b] += i * st
             rt = deferred.get_runtime()
             return rt.read_element(self.bdarray, base)
-        view = self.view.apply_index(index)
+        try:
+            key = (self.view, index)
+            view = _getitem_cache.get(key)
+            if view is None:
+                view = self.view.apply_index(index)
+                if len(_getitem_cache) > 8192:
+                    _getitem_cache.clear()
+                _getitem_cache[key] = view
+        except TypeError:       # unhashable index element
+            view = self.view.apply_index(index)
         return self._with_view(view)
 
     def __setitem__(self, index, value):
