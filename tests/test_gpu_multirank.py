@@ -127,3 +127,22 @@ def test_axis_cumsum_and_mask_2rank_gpu():
         return _np.concatenate([r0.asarray().reshape(-1),
                                 r1.asarray().reshape(-1), sel.asarray()])
     """, tol=1e-12)
+
+
+def test_mixed_pipeline_2rank_gpu():
+    """The full configs[4] shape at world 2: fused fill, stencil with halo
+    exchange, fused reduction + allreduce, eager force each step."""
+    run_spmd_gpu("""
+        S = 512
+        A = np_.zeros((S, S), dtype=np.float64)
+        out = []
+        for _ in range(3):
+            src = np_.fromfunction(
+                lambda x, y: (x * S + y) * 1e-6, (S, S), dtype=np.float64)
+            ssin = np_.sin(src)
+            A[1:-1, 1:-1] = (ssin[:-2, 1:-1] + ssin[2:, 1:-1]
+                             + ssin[1:-1, :-2] + ssin[1:-1, 2:]
+                             - 4.0 * ssin[1:-1, 1:-1])
+            out.append(float(A.sum()))
+        return np.asarray(out)
+    """, tol=1e-9)
