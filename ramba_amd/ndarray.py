@@ -170,10 +170,23 @@ class ndarray:
                 raise IndexError(
                     f"boolean index shape {index.shape} does not match "
                     f"array shape {self.shape}")
-            ta = self.copy()
-            tm = index.astype(np.uint8)
-            deferred.flush()
+            # copy value+mask onto ONE fresh C-contiguous (axis-0 split)
+            # partition through the fused engine: cross-rank C-order
+            # concatenation of per-rank compactions is then always valid,
+            # whatever the source partitions look like
+            from .common import contiguous_divisions
             rt = deferred.get_runtime()
+            cdivs = contiguous_divisions(rt.world, self.shape)
+            ta_bd = deferred.bdarray(self.shape, self.dtype, cdivs,
+                                     default_border, flex=False)
+            ta = ndarray(ta_bd, View.identity(self.shape))
+            deferred.add_op(ta, "=", self, empty_like=empty_like)
+            tm_bd = deferred.bdarray(self.shape, np.dtype(np.uint8), cdivs,
+                                     default_border, flex=False)
+            tm = ndarray(tm_bd, View.identity(self.shape))
+            deferred.add_op(tm, "=", ir.Cast(index, np.dtype(np.uint8)),
+                            empty_like=empty_like)
+            deferred.flush()
             out_bd = rt.mask_compact_op(ta.bdarray, tm.bdarray)
             return ndarray(out_bd, View.identity(out_bd.shape))
         if not isinstance(index, tuple):

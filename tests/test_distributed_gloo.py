@@ -254,22 +254,29 @@ def test_poison_sanitizer_halo_correct():
 
 
 def test_mask_getitem_spmd():
-    """Compressing a[mask] across ranks: uneven result divisions ordered
-    by core-box origin; no data-path exchange (runtime.mask_compact_op)."""
-    for world in (2, 3):
+    """Compressing a[mask] across ranks: value+mask repartitioned onto a
+    C-contiguous split, per-rank compaction, uneven result divisions
+    (runtime.mask_compact_op).  world=4 puts a square 2-D input on a 2x2
+    grid — the case the repartition makes legal."""
+    for world in (2, 3, 4):
         run_spmd("""
             a = np_.arange(1000) * 1.0
             b = np_.sin(a * 0.01)
             sel = b[b > 0.3]
             t = np_.fromfunction(lambda x, y: x * 3 + y, (401, 3))
             sel2 = t[(t % 7.0) == 0.0]
+            q = np_.fromfunction(lambda x, y: x * 37.0 + y, (36, 37))
+            sel3 = q[(q % 5.0) == 0.0]
             if np_ is np:
-                return np.concatenate([sel, sel2,
-                                       [float(sel.size), float(sel2.size)]])
+                return np.concatenate([sel, sel2, sel3,
+                                       [float(sel.size), float(sel2.size),
+                                        float(sel3.size)]])
             import numpy as _np
             return _np.concatenate([sel.asarray(), sel2.asarray(),
+                                    sel3.asarray(),
                                     [float(sel.shape[0]),
-                                     float(sel2.shape[0])]])
+                                     float(sel2.shape[0]),
+                                     float(sel3.shape[0])]])
         """, world=world, tol=1e-12)
 
 
