@@ -181,6 +181,7 @@ def main():
     # our kernels — gpurun_out/s1err.log).  Warm up for at least 6 s of
     # wall time so it lands before the timed window; the JSON reports the
     # warmup steps actually done.
+    warm_requested = args.warmup
     warm_done = 0
     tw = time.perf_counter()
     if world == 1:
@@ -331,6 +332,7 @@ def main():
                 "workload": wl_desc,
                 "elems": N,
                 "parallelism": f"shard{world}",
+                "warmup_requested": warm_requested,
             },
             "roofline": roofline,
             "cpu_baseline": cpu_baseline,
