@@ -331,3 +331,21 @@ def test_reshape_spmd():
                                     r5.asarray().reshape(-1),
                                     r6.asarray()])
         """, world=world, tol=0.0)
+
+
+def test_scalar_reads_and_fromarray_spmd():
+    """Cross-rank scalar reads (owner broadcast) and numpy ingestion
+    (scatter) at worlds 3/4."""
+    for world in (3, 4):
+        run_spmd("""
+            a = np_.arange(997) * 3.0
+            reads = [float(a[0]), float(a[500]), float(a[996]),
+                     float(a[-1])]
+            import numpy as _np
+            src = _np.fromfunction(lambda i, j: i * 11.0 - j, (23, 17))
+            b = np_.fromarray(src) if np_ is not np else src
+            c = (b * 2.0).sum(axis=1)
+            if np_ is np:
+                return np.concatenate([reads, (src * 2).sum(axis=1)])
+            return _np.concatenate([reads, c.asarray()])
+        """, world=world, tol=1e-12)
