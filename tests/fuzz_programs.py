@@ -85,9 +85,11 @@ def _rand_slice(rng, n):
 def build_program(seed):
     """Returns (impl, tol): impl(np_) runs the program; tol for comparison."""
     master = np.random.default_rng(seed)
-    two_d = bool(master.integers(0, 2))
+    ndim = int(master.integers(1, 4))   # 1-, 2- or 3-D base arrays
+    two_d = ndim == 2
     n0 = int(master.integers(37, 400))
     n1 = int(master.integers(17, 120))
+    n2 = int(master.integers(5, 23))
     nsteps = int(master.integers(3, 9))
     # pre-draw all decisions so both interpretations agree
     plan_seed = int(master.integers(0, 2 ** 31))
@@ -99,7 +101,11 @@ def build_program(seed):
         scalars = []
 
         def mk_base():
-            if two_d:
+            if ndim == 3:
+                a = np_.fromfunction(
+                    lambda x, y, z: x * 23 + y * 5 + z, (n1, n2, 11),
+                    dtype=np.int64)
+            elif two_d:
                 a = np_.fromfunction(
                     lambda x, y: x * 7 + y * 3, (n0, n1), dtype=np.int64)
             else:
@@ -143,12 +149,20 @@ def build_program(seed):
                 if val.ndim == 1:
                     sl = _rand_slice(rng, val.shape[0])
                     pool.append((val[sl], kind))
-                else:
+                elif val.ndim == 2:
                     sl = (_rand_slice(rng, val.shape[0]),
                           _rand_slice(rng, val.shape[1]))
                     v = val[sl]
                     if rng.integers(0, 3) == 0:
                         v = v.T
+                    pool.append((v, kind))
+                else:
+                    sl = tuple(_rand_slice(rng, val.shape[d])
+                               for d in range(3))
+                    v = val[sl]
+                    if rng.integers(0, 3) == 0:
+                        perm = list(rng.permutation(3))
+                        v = v.transpose(perm)
                     pool.append((v, kind))
             elif action == "setitem":
                 # write a computed value into a slice of a FRESH array
@@ -181,8 +195,8 @@ def build_program(seed):
                 r = getattr(val, op)()
                 scalars.append(float(r))
             elif action == "axred":
-                if val.ndim == 2 and min(val.shape) >= 2:
-                    ax = int(rng.integers(0, 2))
+                if val.ndim >= 2 and min(val.shape) >= 2:
+                    ax = int(rng.integers(0, val.ndim))
                     r = val.sum(axis=ax)
                     pool.append((r, "i" if kind == "i" else "f"))
             elif action == "cumsum":
@@ -214,8 +228,8 @@ def build_program(seed):
                     r = val[:, None] * (val[None, :] + 1)
                     pool.append((r, kind))
             elif action == "axcumsum":
-                if val.ndim == 2 and kind in ("f", "i"):
-                    ax = int(rng.integers(0, 2))
+                if val.ndim >= 2 and kind in ("f", "i"):
+                    ax = int(rng.integers(0, val.ndim))
                     pool.append((val.cumsum(axis=ax), kind))
             elif action == "maskget":
                 # integer-valued membership only: a threshold on computed
