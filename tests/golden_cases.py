@@ -66,6 +66,50 @@ def _pi(np_):
     return np.array([float(4.0 * h * (1.0 / (1.0 + x * x)).sum())])
 
 
+def _axis_cumsum(np_):
+    c = np_.fromfunction(lambda x, y: x * 97 + y, (211, 67))
+    r0 = c.cumsum(axis=0)
+    r1 = c.cumsum(axis=1)
+    i = np_.fromfunction(lambda x, y: x + y, (50, 33),
+                         dtype=np.int32).cumsum(axis=0)
+    out = [r0, r1, i]
+    if hasattr(r0, "asarray"):
+        out = [o.asarray() for o in out]
+    return np.concatenate([np.asarray(o, dtype=np.float64).reshape(-1)
+                           for o in out])
+
+
+def _mask_getitem(np_):
+    a = np_.arange(100_000) * 1.0
+    sel = a[(a % 7.0) == 0.0]
+    t = np_.fromfunction(lambda x, y: x * 31 + y, (300, 31))
+    sel2 = t[(t % 13.0) == 0.0]
+    if hasattr(sel, "asarray"):
+        sel, sel2 = sel.asarray(), sel2.asarray()
+    return np.concatenate([sel, sel2, [float(sel.size), float(sel2.size)]])
+
+
+def _reshape(np_):
+    a = np_.arange(60_060) * 1.0
+    r1 = a.reshape(231, 260)
+    b = np_.fromfunction(lambda i, j: i * 77.0 + j, (91, 77))
+    r2 = b.T.reshape(7007)
+    if hasattr(r1, "asarray"):
+        r1, r2 = r1.asarray(), r2.asarray()
+    return np.concatenate([np.asarray(r1).reshape(-1), np.asarray(r2)])
+
+
+def _matmul(np_):
+    A = np_.fromfunction(lambda i, j: i * 7.0 + j, (65, 33))
+    B = np_.fromfunction(lambda i, j: i - 2.0 * j, (33, 49))
+    v = np_.arange(33) * 1.0
+    M = A @ B
+    w = A @ v
+    if hasattr(M, "asarray"):
+        M, w = M.asarray(), w.asarray()
+    return np.concatenate([np.asarray(M).reshape(-1), np.asarray(w)])
+
+
 # name -> (fn, tolerance; 0 = bit-exact)
 CASES_TOL = {
     "flagship": (_flagship, 1e-12),
@@ -76,6 +120,10 @@ CASES_TOL = {
     "arange_exact": (_arange_exact, 0),
     "reductions": (_reductions, 1e-12),
     "pi": (_pi, 1e-12),
+    "axis_cumsum": (_axis_cumsum, 1e-9),
+    "mask_getitem": (_mask_getitem, 0),
+    "reshape": (_reshape, 0),
+    "matmul": (_matmul, 1e-9),
 }
 
 CASES = {k: v[0] for k, v in CASES_TOL.items()}
