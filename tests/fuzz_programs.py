@@ -12,8 +12,13 @@ import numpy as np
 
 FLOAT_UN = ["sin", "cos", "sqrt", "tanh", "exp_neg", "arctan"]
 ANY_UN = ["neg", "abs", "square", "floor", "sign"]
-BIN = ["add", "sub", "mul", "maximum", "minimum", "mod1", "div1"]
-INT_BIN = ["floordiv1", "bitand", "bitxor"]
+# mod/floordiv are DISCONTINUOUS: on float chains a 1-ulp upstream
+# difference (the div->mul-by-reciprocal rewrite both we and the
+# reference apply, ramba.py:6121) lands O(divisor) apart at an
+# exact-multiple boundary (first seen at seed 23716 after 33k clean
+# seeds), so the comparison is only sound over exact integer arithmetic
+BIN = ["add", "sub", "mul", "maximum", "minimum", "div1"]
+INT_BIN = ["mod1", "floordiv1", "bitand", "bitxor"]
 
 
 def _apply_un(np_, x, op):
