@@ -11,7 +11,11 @@ import numpy as np
 
 
 FLOAT_UN = ["sin", "cos", "sqrt", "tanh", "exp_neg", "arctan"]
-ANY_UN = ["neg", "abs", "square", "floor", "sign"]
+# floor/sign are discontinuous too (seed 5828: floor lands 27 vs 28 when
+# the 1-ulp div-rewrite difference straddles an integer) -> int-only,
+# like mod/floordiv below
+ANY_UN = ["neg", "abs", "square"]
+INT_UN = ["floor", "sign"]
 # mod/floordiv are DISCONTINUOUS: on float chains a 1-ulp upstream
 # difference (the div->mul-by-reciprocal rewrite both we and the
 # reference apply, ramba.py:6121) lands O(divisor) apart at an
@@ -130,7 +134,7 @@ def build_program(seed):
             val, kind = pool[i]
             if action == "un":
                 if kind == "i":
-                    op = str(rng.choice(ANY_UN))
+                    op = str(rng.choice(ANY_UN + INT_UN))
                 else:
                     op = str(rng.choice(FLOAT_UN + ANY_UN))
                 r = _apply_un(np_, val, op)
