@@ -212,9 +212,11 @@ def build_program(seed):
                 if val.ndim == 1 and val.shape[0] > 0 and kind == "i":
                     pool.append((val.cumsum(), "i"))
             elif action == "astype":
-                dt = np.float32 if kind == "i" else np.int64
-                pool.append((val.astype(dt), "g" if dt == np.float32
-                             else "i"))
+                # float->int truncation is discontinuous like floor
+                # (seed 28097: 27 vs 28 across a 1-ulp boundary) ->
+                # int widens to f32, floats narrow to f32 (continuous)
+                dt = np.float32
+                pool.append((val.astype(dt), "g"))
             elif action == "where":
                 c = float(rng.uniform(-5.0, 50.0))
                 r = np_.where(val > c, val, -val)
