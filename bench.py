@@ -238,6 +238,7 @@ def main():
     # ---- roofline leg: HIP-event time of the fused kernel itself ----------
     backend.time_kernels = True
     backend.kernel_times_ms = []
+    backend.kernel_keys = []
     for _ in range(3):
         keep = step()
     backend.time_kernels = False
@@ -258,12 +259,33 @@ def main():
     roofline = None
     if kms:
         achieved = alg_bytes_per_elem * local_elems / (kms / 1e3)
+        # PMC-measured per-launch HBM traffic is valid evidence only for
+        # the EXACT kernel it was captured against: match the live
+        # kernel's content-addressed key against the manifest written at
+        # profiling time (tools/make_pmc_manifest.py); no match -> null.
         traffic = os.environ.get("RAMBA_BENCH_TRAFFIC")
-        if traffic is None and args.workload == "flagship" \
-                and N == 1_000_000_000 and world == 1:
-            # measured via rocprofv3 PMC on this workload (profiles/README.md
-            # r01): FETCH 4.00GB (x2 gfx950 correction = 8GB) + WRITE 24.00GB
-            traffic = 32.0e9
+        traffic_src = "RAMBA_BENCH_TRAFFIC env" if traffic else None
+        if traffic is None:
+            per_step_keys = backend.kernel_keys[:max(1, len(
+                backend.kernel_keys) // 3)]
+            dom = None
+            if kt:
+                di = max(range(len(per_step_keys)),
+                         key=lambda i: kt[i]) if per_step_keys else None
+                dom = per_step_keys[di] if di is not None else None
+            mpath = os.path.join(ROOT, "profiles", "pmc_manifest.json")
+            if dom and os.path.exists(mpath):
+                try:
+                    man = json.load(open(mpath))
+                    ent = man.get(dom)
+                    if ent and ent.get("workload") == args.workload \
+                            and int(ent.get("elems", -1)) == N \
+                            and int(ent.get("world", -1)) == world:
+                        traffic = float(ent["bytes_per_launch"])
+                        traffic_src = ent.get("source",
+                                              "rocprofv3 PMC manifest")
+                except Exception:  # noqa: BLE001
+                    pass
         roofline = {
             "bound": "hbm",
             "achieved": achieved / 1e9,          # GB/s
@@ -271,8 +293,7 @@ def main():
             "unit": "GB/s",
             "frac": achieved / HBM_PEAK_BYTES,
             "traffic": float(traffic) if traffic else None,
-            "traffic_source": "rocprofv3 PMC, profiles/r01_pmc_*.csv"
-                              if traffic else None,
+            "traffic_source": traffic_src if traffic else None,
             "kernel_ms": kms,
         }
 

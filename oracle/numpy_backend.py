@@ -133,6 +133,11 @@ class NumpyBackend:
         dist = self._d()
         v = np.asarray(value)
         dt = v.dtype
+        if kind in ("min", "max") and dt.kind == "f":
+            # same NaN-propagation contract as the HIP backend
+            parts = self.allgather_scalars(value, dt)
+            red = np.minimum.reduce if kind == "min" else np.maximum.reduce
+            return np.asarray(red(np.asarray(parts, dtype=dt)), dtype=dt)[()]
         if dt == np.bool_:
             v = v.astype(np.uint8)
         t = torch.from_numpy(v.reshape(1).copy())

@@ -496,8 +496,35 @@ __device__ __forceinline__ void rt_sincos(double x, double *sr, double *cr) {
     double r = __builtin_fma(-fn, pio2_1, x);
     double w = fn * pio2_1t;
     double y0 = r - w;
-    double y1 = (r - y0) - w;
-    (void)pio2_2; (void)pio2_2t;
+    // second/third Cody-Waite stages when cancellation near a multiple
+    // of pi/2 ate the leading bits (fdlibm __ieee754_rem_pio2 medium
+    // path): keeps RELATIVE accuracy of sin/cos near their zeros, not
+    // just the 1e-12 absolute parity bar
+    long long hx = (long long)__builtin_bit_cast(unsigned long long, x);
+    long long hy = (long long)__builtin_bit_cast(unsigned long long, y0);
+    int bits_lost = (int)((hx >> 52) & 0x7ff) - (int)((hy >> 52) & 0x7ff);
+    double y1;
+    if (bits_lost > 16) {
+        double t = r;
+        w = fn * pio2_2;
+        r = t - w;
+        w = __builtin_fma(fn, pio2_2t, -((t - r) - w));
+        y0 = r - w;
+        hy = (long long)__builtin_bit_cast(unsigned long long, y0);
+        bits_lost = (int)((hx >> 52) & 0x7ff) - (int)((hy >> 52) & 0x7ff);
+        if (bits_lost > 49) {
+            const double pio2_3  = 2.02226624871116645580e-21;
+            const double pio2_3t = 8.47842766036889956997e-32;
+            t = r;
+            w = fn * pio2_3;
+            r = t - w;
+            w = __builtin_fma(fn, pio2_3t, -((t - r) - w));
+            y0 = r - w;
+        }
+        y1 = (r - y0) - w;
+    } else {
+        y1 = (r - y0) - w;
+    }
     // kernel_sin(y0, y1)
     const double S1 = -1.66666666666666324348e-01;
     const double S2 = 8.33333333332248946124e-03;

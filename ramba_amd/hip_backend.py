@@ -135,6 +135,7 @@ class HipBackend:
         # optional HIP-event kernel timing (bench roofline leg)
         self.time_kernels = False
         self.kernel_times_ms = []
+        self.kernel_keys = []
 
     # ------------------------------------------------------------------
     def _check(self, rc, what):
@@ -310,6 +311,13 @@ class HipBackend:
         dist = self._d()
         v = np.asarray(value)
         dt = v.dtype
+        if kind in ("min", "max") and dt.kind == "f":
+            # NCCL/RCCL MIN/MAX does not reliably propagate NaN across
+            # ranks, while the in-kernel combine and NumPy do:
+            # gather the per-rank partials and combine on the host
+            parts = self.allgather_scalars(value, dt)
+            red = np.minimum.reduce if kind == "min" else np.maximum.reduce
+            return np.asarray(red(np.asarray(parts, dtype=dt)), dtype=dt)[()]
         if dt == np.bool_:
             v = v.astype(np.uint8)
         t = self.torch.from_numpy(v.reshape(1).copy())
@@ -415,6 +423,7 @@ class HipBackend:
             ev1.record()
             ev1.synchronize()
             self.kernel_times_ms.append(ev0.elapsed_time(ev1))
+            self.kernel_keys.append(gk.key)
         results = []
         if gk.nred:
             fargs = codegen.pack_finish_args(

@@ -371,6 +371,13 @@ class ndarray:
         dt = _result_dtype(op, a, b) if out_dtype is None else out_dtype
         a, b = _bcast(a, shape), _bcast(b, shape)
         if op == "pow" and isinstance(b, (int, np.integer)) \
+                and not isinstance(b, (bool, np.bool_)) and b < 0 \
+                and np.dtype(dt).kind in "iu":
+            # match NumPy/the oracle: the HIP rt_ipow loop would silently
+            # return 1 here
+            raise ValueError(
+                "Integers to negative integer powers are not allowed.")
+        if op == "pow" and isinstance(b, (int, np.integer)) \
                 and not isinstance(b, (bool, np.bool_)) and 0 <= b <= 8:
             # small-int exponent baked as a compile-time constant so the
             # kernel emits a multiply chain (x**2 == x*x, NumPy's fast path)
