@@ -349,3 +349,16 @@ def test_scalar_reads_and_fromarray_spmd():
                 return np.concatenate([reads, (src * 2).sum(axis=1)])
             return _np.concatenate([reads, c.asarray()])
         """, world=world, tol=1e-12)
+
+
+@pytest.mark.parametrize("world", [2])
+def test_minmax_nan_spmd(world):
+    """Float min/max propagate a NaN living on only one rank (ADVICE r1:
+    the backend allgathers float min/max partials instead of trusting
+    collective MIN/MAX with NaN)."""
+    run_spmd("""
+        A = np_.arange(10000) * 1.0
+        B = np_.where(A == 9999.0, (A - 20000.0) ** 0.5, A)  # one NaN
+        return np.asarray([float(B.min()), float(B.max()),
+                           float(A.min()), float(A.max())])
+    """, world=world, tol=0.0)

@@ -1122,3 +1122,18 @@ class TestNewOpDtypeMatrix:
         a = ra.arange(6)
         assert a.tolist() == [0, 1, 2, 3, 4, 5]
         assert a.nbytes == 48 and a.itemsize == 8
+
+
+def test_negative_int_pow_raises(ra):
+    """int_array ** negative int must raise like NumPy (the HIP
+    rt_ipow loop would silently return 1)."""
+    a = ra.arange(10)
+    with pytest.raises(ValueError):
+        a ** -2
+    with pytest.raises(ValueError):
+        a ** np.int64(-1)
+    # float base stays fine
+    b = ra.arange(10) * 1.0
+    r = (b[1:] ** -1).asarray()
+    np.testing.assert_allclose(r, (np.arange(10) * 1.0)[1:] ** -1,
+                               rtol=1e-12)
