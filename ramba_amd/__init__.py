@@ -45,7 +45,9 @@ def init(backend=None, world_size=None, rank=None):
         from .hip_backend import HipBackend
         backend = HipBackend()
 
-    if world > 1:
+    if world > 1 or int(os.environ.get("RAMBA_FORCE_PG", "0")):
+        # RAMBA_FORCE_PG=1 initialises the process group even at world 1
+        # (world-1 RCCL transport tests on a single-GPU box)
         backend.init_process_group(rank, world)
 
     rt = Runtime(backend, rank=rank, world=world)
