@@ -362,3 +362,22 @@ def test_minmax_nan_spmd(world):
         return np.asarray([float(B.min()), float(B.max()),
                            float(A.min()), float(A.max())])
     """, world=world, tol=0.0)
+
+
+@pytest.mark.parametrize("world", [2, 4])
+def test_multi_axis_reduction_spmd(world):
+    """Axis-TUPLE reductions across ranks: the combining box exchange
+    projects several reduced axes at once."""
+    run_spmd("""
+        a = np_.fromfunction(lambda x, y, z: x * 1000 + y * 31 + z,
+                             (32, 21, 45))
+        s02 = a.sum(axis=(0, 2))
+        s12 = a.sum(axis=(1, 2))
+        m01 = a.max(axis=(0, 1))
+        k = a.sum(axis=(0, 1), keepdims=True)
+        if np_ is np:
+            return np.concatenate([s02, s12, m01, k.reshape(-1)])
+        import numpy as _np
+        return _np.concatenate([s02.asarray(), s12.asarray(),
+                                m01.asarray(), k.asarray().reshape(-1)])
+    """, world=world, tol=1e-12)

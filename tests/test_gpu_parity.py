@@ -268,6 +268,55 @@ class TestAxisReduction:
                              int(m.all(axis=1).sum())])
         run_both(impl, ra_gpu)
 
+    # -- multi-axis tuples (reference reduce_axes
+    #    handles axis tuples, shardview_array.py:1054, ramba.py:5818-5849)
+
+    def test_sum_axis_tuple_02(self, ra_gpu):
+        def impl(np_):
+            a = np_.fromfunction(lambda x, y, z: x * 1000 + y * 31 + z,
+                                 (16, 31, 64), dtype=np.int64)
+            return a.sum(axis=(0, 2))
+        run_both(impl, ra_gpu)
+
+    def test_sum_axis_tuple_12_lane_split(self, ra_gpu):
+        def impl(np_):
+            a = np_.fromfunction(lambda x, y, z: x * 0.5 + y * 0.25 + z,
+                                 (24, 37, 129), dtype=np.float64)
+            return a.sum(axis=(1, 2))
+        run_both(impl, ra_gpu, tol=1e-12)
+
+    def test_sum_axis_tuple_01_keepdims(self, ra_gpu):
+        def impl(np_):
+            a = np_.fromfunction(lambda x, y, z: x * 7 + y * 3 + z,
+                                 (19, 23, 55), dtype=np.int64)
+            return a.sum(axis=(0, 1), keepdims=True)
+        run_both(impl, ra_gpu)
+
+    def test_minmax_prod_axis_tuple(self, ra_gpu):
+        def impl(np_):
+            a = np_.fromfunction(
+                lambda x, y, z: (x * 37 + y * 11 + z) % 13, (17, 29, 41),
+                dtype=np.int64)
+            return np.array([int(a.min(axis=(0, 2)).sum()),
+                             int(a.max(axis=(0, 1)).sum()),
+                             int((a % 3 + 1).prod(axis=(1, 2))[3])])
+        run_both(impl, ra_gpu)
+
+    def test_axis_tuple_4d(self, ra_gpu):
+        def impl(np_):
+            a = np_.fromfunction(
+                lambda w, x, y, z: w * 101 + x * 17 + y * 5 + z,
+                (6, 7, 8, 9), dtype=np.int64)
+            return a.sum(axis=(1, 3))
+        run_both(impl, ra_gpu)
+
+    def test_negative_axis_tuple(self, ra_gpu):
+        def impl(np_):
+            a = np_.fromfunction(lambda x, y, z: x + y + z, (10, 11, 12),
+                                 dtype=np.int64)
+            return a.sum(axis=(-1, 0))
+        run_both(impl, ra_gpu)
+
 
 class TestCumsum:
     def test_cumsum_large_int(self, ra_gpu):

@@ -1137,3 +1137,17 @@ def test_negative_int_pow_raises(ra):
     r = (b[1:] ** -1).asarray()
     np.testing.assert_allclose(r, (np.arange(10) * 1.0)[1:] ** -1,
                                rtol=1e-12)
+
+
+def test_multi_axis_reductions(ra):
+    """Axis tuples through reduce_axes_op."""
+    def impl(np_):
+        a = np_.fromfunction(lambda x, y, z: x * 100 + y * 10 + z,
+                             (7, 9, 11))
+        return np.concatenate([
+            np.asarray(a.sum(axis=(0, 2))).reshape(-1),
+            np.asarray(a.min(axis=(1, 2))).reshape(-1),
+            np.asarray(a.max(axis=(0, 1), keepdims=True)).reshape(-1),
+            np.asarray(a.sum(axis=(-1, 0))).reshape(-1)])
+    from conftest import run_both
+    run_both(impl, ra, tol=1e-12)
