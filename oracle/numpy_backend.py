@@ -85,17 +85,26 @@ class NumpyBackend:
         import torch
         return torch.from_numpy(np.empty(shape, dtype=dtype))
 
-    def exchange(self, sends, recvs):
+    def exchange_begin(self, sends, recvs):
+        """Post the pairwise exchange (gloo over host tensors)."""
         if not sends and not recvs:
-            return
+            return None
         dist = self._d()
         ops = []
         for (dst, buf) in sends:
             ops.append(dist.P2POp(dist.isend, buf, dst))
         for (src, buf) in recvs:
             ops.append(dist.P2POp(dist.irecv, buf, src))
-        for req in dist.batch_isend_irecv(ops):
+        return dist.batch_isend_irecv(ops) if ops else []
+
+    def exchange_finish(self, token):
+        if token is None:
+            return
+        for req in token:
             req.wait()
+
+    def exchange(self, sends, recvs):
+        self.exchange_finish(self.exchange_begin(sends, recvs))
 
     def unpack_box_to_container(self, bd, rt, box, buf):
         self._cont(bd)[rt.container_slice(bd, box)] = buf.numpy()

@@ -192,7 +192,9 @@ def seed(fuzz_seeds=None):
         from fuzz_programs import build_program
         with np.errstate(all="ignore"):
             for sd in range(fuzz_seeds):
-                impl, _ = build_program(sd)
+                impl, _ = build_program(sd, mode="numpy")
+                impl(ra)
+                impl, _ = build_program(sd, mode="oracle")
                 impl(ra)
     n = be.compiled
     ra.shutdown()

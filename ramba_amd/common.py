@@ -27,6 +27,11 @@ ntiming = int(os.environ.get("RAMBA_TIMING", "0"))
 # of an unfilled border or uninitialised cell corrupts results visibly —
 # the analog of the reference build plan's hipMemsetD poisoning
 debug_poison = int(os.environ.get("RAMBA_DEBUG_POISON", "0"))
+# halo/compute overlap: split each rank's iteration box into interior
+# (launched while the exchange is in flight) + rim slabs (after it lands)
+# — BASELINE configs[4] "overlapped RCCL"; 0 restores the reference-style
+# serial exchange-then-execute (ramba.py:3547-3693) for A/B measurement
+overlap_exchange = int(os.environ.get("RAMBA_OVERLAP", "1"))
 
 
 def dprint(level, *args):

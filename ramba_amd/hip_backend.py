@@ -240,13 +240,20 @@ class HipBackend:
     def new_message_buffer(self, shape, dtype):
         return self.torch.empty(shape, dtype=self._tdt(dtype), device="cuda")
 
-    def exchange(self, sends, recvs):
+    def exchange_begin(self, sends, recvs):
+        """POST the pairwise exchange; returns a token for
+        exchange_finish.  On the nccl(=RCCL) path this is non-blocking on
+        both host and device: torch orders the RCCL stream after the
+        current stream (our pack kernels), so interior compute launched
+        after this call overlaps the wire time.  The gloo path stages
+        through host buffers (tests only)."""
         if not sends and not recvs:
-            return
+            return None
         dist = self._d()
-        self.torch.cuda.synchronize()  # pack kernels complete before NCCL
         staged = getattr(self, "_pg", "nccl") != "nccl"
+        hrecvs = None
         if staged:
+            self.torch.cuda.synchronize()  # packs -> host staging copies
             sends = [(d, b.cpu()) for (d, b) in sends]
             hrecvs = [(s, b, b.cpu()) for (s, b) in recvs]
         ops = []
@@ -258,11 +265,21 @@ class HipBackend:
         else:
             for (src, buf) in recvs:
                 ops.append(dist.P2POp(dist.irecv, buf, src))
-        for req in dist.batch_isend_irecv(ops):
+        reqs = dist.batch_isend_irecv(ops) if ops else []
+        return (reqs, hrecvs)
+
+    def exchange_finish(self, token):
+        if token is None:
+            return
+        reqs, hrecvs = token
+        for req in reqs:
             req.wait()
-        if staged:
+        if hrecvs:
             for (_, dbuf, hb) in hrecvs:
                 dbuf.copy_(hb.to(dbuf.device))
+
+    def exchange(self, sends, recvs):
+        self.exchange_finish(self.exchange_begin(sends, recvs))
 
     def unpack_box_to_container(self, bd, rt, box, buf):
         cont = self._cont(bd)

@@ -448,6 +448,15 @@ class ndarray:
             axes = tuple(sorted(a + self.ndim if a < 0 else a for a in axes))
             if len(axes) == self.ndim:
                 axis = None
+        if kind in ("min", "max"):
+            # NumPy parity (fuzz v6 seed 5235): no identity for min/max —
+            # raises iff a REDUCED extent is zero (axis=None: any extent)
+            red = range(self.ndim) if axis is None else axes
+            if any(self.shape[i] == 0 for i in red):
+                raise ValueError(
+                    "zero-size array to reduction operation "
+                    f"{'minimum' if kind == 'min' else 'maximum'} "
+                    "which has no identity")
         if axis is None:
             src = self
             if kind in ("all", "any") and self.dtype != ir.BOOL:

@@ -58,10 +58,21 @@ class KernelPlan:
         self.temps = {}             # temp_key -> (shape, dtype) buffers
 
 
+class LaunchUnit:
+    """One kernel launch over one sub-box of the rank's iteration space
+    (interior or rim slab); carries its own cached kernel handle."""
+    __slots__ = ("plan", "backend_kernel")
+
+    def __init__(self, plan):
+        self.plan = plan
+        self.backend_kernel = None
+
+
 class Recipe:
     """Cached execution geometry for one fused-group structure."""
     __slots__ = ("plan", "temp_specs", "comm_msgs", "temp_geom",
-                 "adopted_divs", "backend_kernel")
+                 "adopted_divs", "backend_kernel", "units",
+                 "pre_wait_units")
 
     def __init__(self):
         self.plan = None
@@ -70,6 +81,12 @@ class Recipe:
         self.temp_geom = {}
         self.adopted_divs = None
         self.backend_kernel = None
+        # halo/compute overlap (BASELINE configs[4] "overlapped RCCL",
+        # replacing the reference's serial exchange-then-execute,
+        # ramba.py:3547-3693): `units` is the launch list; the first
+        # `pre_wait_units` of them run between comms-post and comms-wait
+        self.units = None
+        self.pre_wait_units = 0
 
 
 class Runtime:
@@ -219,20 +236,41 @@ class Runtime:
             if part is not None and bd.constructed:
                 self.backend.copy_container_to_temp(bd, self, part, vname,
                                                     need)
+        state = None
         if recipe.comm_msgs:
             msgs = [(dst, src, live[owner].bd, bx, tgt)
                     for (dst, src, owner, bx, tgt) in recipe.comm_msgs]
-            self._do_comms(msgs, recipe.temp_geom)
+            state = self._comms_begin(msgs, recipe.temp_geom)
         from .common import ntiming
         tl = time.perf_counter() if ntiming else 0.0
-        if plan.itershape is not None:
-            partials = self.backend.launch(plan, recipe)
+        if recipe.units is not None:
+            # overlapped split: interior unit(s) launch while the
+            # exchange is in flight, rim units after it lands
+            for u in recipe.units:
+                u.plan.scalars = plan.scalars
+                for op in u.plan.operands:
+                    op.bd = live[op.name].bd
+            for i, u in enumerate(recipe.units):
+                if i == recipe.pre_wait_units and state is not None:
+                    self._comms_finish(state)
+                    state = None
+                self.backend.launch(u.plan, u)
+            if state is not None:
+                self._comms_finish(state)
+                state = None
+            partials = []
             if ntiming:
                 add_time("eg_launch_host", time.perf_counter() - tl)
         else:
-            partials = [np.asarray(ir.reduction_init(spec.kind, spec.dtype),
-                                   dtype=spec.dtype)[()]
-                        for spec in plan.reductions]
+            if state is not None:
+                self._comms_finish(state)
+                state = None
+            if plan.itershape is not None:
+                partials = self.backend.launch(plan, recipe)
+            else:
+                partials = [np.asarray(
+                    ir.reduction_init(spec.kind, spec.dtype),
+                    dtype=spec.dtype)[()] for spec in plan.reductions]
             if ntiming:
                 add_time("eg_launch_host", time.perf_counter() - tl)
         for (spec, pend), val in zip(group.reductions, partials):
@@ -357,44 +395,46 @@ class Runtime:
         recipe.temp_geom = temp_geom
         recipe.comm_msgs = comm_msgs
 
-        # ---- execute transfers
+        # ---- launches, with the exchange posted first and (when the
+        # geometry allows) the interior launched while it is in flight
+        ib = eboxes[self.rank]
+        split_boxes, pre_wait = self._plan_overlap_split(
+            ib, comm_msgs, plan, temp_geom, live)
+        state = None
         if comm_msgs:
             msgs = [(dst, src, live[owner].bd, bx, tgt)
                     for (dst, src, owner, bx, tgt) in comm_msgs]
-            self._do_comms(msgs, temp_geom)
-
-        # ---- operand descriptors for the kernel
-        ib = eboxes[self.rank]
-        if ib is not None:
-            plan.itershape = box_shape(ib)
-            plan.global_start = tuple(int(x) for x in ib[0])
-            for name, oi in live.items():
-                if name in temp_geom:
-                    need, strides = temp_geom[name]
-                    off0, s = oi.view.operand_addressing(
-                        ib[0], strides, need[0], (0,) * len(strides))
-                    plan.operands.append(OperandPlan(
-                        name, "temp", oi.bd, name, off0, s, oi.dtype))
-                else:
-                    d, cshape, cstrides, border = self.shard_geometry(oi.bd)
-                    if d is None:
-                        # no local shard; var must be unused here
-                        need = oi.view.image_box(ib)
-                        assert need is None, "operand needed but unowned"
-                        plan.operands.append(OperandPlan(
-                            name, "container", oi.bd, None, 0,
-                            (0,) * len(plan.itershape), oi.dtype))
-                        continue
-                    off0, s = oi.view.operand_addressing(
-                        ib[0], cstrides, d[0], border)
-                    plan.operands.append(OperandPlan(
-                        name, "container", oi.bd, None, off0, s, oi.dtype))
-            partials = self.backend.launch(plan, recipe)
-        else:
+            state = self._comms_begin(msgs, temp_geom)
+        if split_boxes is not None:
+            recipe.units = [
+                LaunchUnit(self._address_plan(plan, live, b, temp_geom))
+                for b in split_boxes]
+            recipe.pre_wait_units = pre_wait
             plan.itershape = None
-            partials = [np.asarray(ir.reduction_init(spec.kind, spec.dtype),
-                                   dtype=spec.dtype)[()]
-                        for spec in plan.reductions]
+            for i, u in enumerate(recipe.units):
+                if i == pre_wait and state is not None:
+                    self._comms_finish(state)
+                    state = None
+                self.backend.launch(u.plan, u)
+            if state is not None:
+                self._comms_finish(state)
+                state = None
+            partials = []
+        else:
+            if state is not None:
+                self._comms_finish(state)
+                state = None
+            if ib is not None:
+                ap = self._address_plan(plan, live, ib, temp_geom)
+                plan.itershape = ap.itershape
+                plan.global_start = ap.global_start
+                plan.operands = ap.operands
+                partials = self.backend.launch(plan, recipe)
+            else:
+                plan.itershape = None
+                partials = [np.asarray(
+                    ir.reduction_init(spec.kind, spec.dtype),
+                    dtype=spec.dtype)[()] for spec in plan.reductions]
         recipe.plan = plan
 
         for (spec, pend), val in zip(group.reductions, partials):
@@ -410,9 +450,77 @@ class Runtime:
         return recipe
 
     # ------------------------------------------------------------------
-    def _do_comms(self, msgs, temp_geom):
-        """Move sub-boxes between ranks.  The message list is identical on
-        every rank; pairwise ordering comes from a deterministic sort."""
+    def _address_plan(self, base_plan, live, ibox, temp_geom):
+        """Operand descriptors for a launch over iteration box `ibox`
+        (shares statements/scalars/dead_vars with the group's plan)."""
+        p = KernelPlan()
+        p.itershape = box_shape(ibox)
+        p.global_start = tuple(int(x) for x in ibox[0])
+        p.scalars = base_plan.scalars
+        p.statements = base_plan.statements
+        p.reductions = base_plan.reductions
+        p.dead_vars = base_plan.dead_vars
+        for name, oi in live.items():
+            if name in temp_geom:
+                need, strides = temp_geom[name]
+                off0, s = oi.view.operand_addressing(
+                    ibox[0], strides, need[0], (0,) * len(strides))
+                p.operands.append(OperandPlan(
+                    name, "temp", oi.bd, name, off0, s, oi.dtype))
+            else:
+                d, cshape, cstrides, border = self.shard_geometry(oi.bd)
+                if d is None:
+                    # no local shard; var must be unused here
+                    need = oi.view.image_box(ibox)
+                    assert need is None, "operand needed but unowned"
+                    p.operands.append(OperandPlan(
+                        name, "container", oi.bd, None, 0,
+                        (0,) * len(p.itershape), oi.dtype))
+                    continue
+                off0, s = oi.view.operand_addressing(
+                    ibox[0], cstrides, d[0], border)
+                p.operands.append(OperandPlan(
+                    name, "container", oi.bd, None, off0, s, oi.dtype))
+        return p
+
+    def _plan_overlap_split(self, ib, comm_msgs, plan, temp_geom, live):
+        """(boxes, pre_wait) for overlapped halo/compute, or (None, 0).
+
+        Interior = the subset of the rank's iteration box whose reads all
+        land inside locally-owned cores (no halo dependency): it launches
+        while the exchange is in flight; rim slabs launch after.  A rank
+        that only SENDS launches its whole box early.  Groups with
+        reductions or temp-materialised operands stay sequential (the
+        reference's serial order, ramba.py:3547-3693, is the fallback)."""
+        from .common import overlap_exchange
+        if (not overlap_exchange or ib is None or not comm_msgs
+                or plan.reductions or temp_geom):
+            return None, 0
+        if not any(dst == self.rank for (dst, _, _, _, _) in comm_msgs):
+            return [ib], 1            # send-only rank: nothing to wait for
+        interior = ib
+        for name, oi in live.items():
+            if oi.written:
+                continue              # writes are core-contained (asserted)
+            need = oi.view.image_box(ib)
+            if need is None:
+                continue
+            core = self.core_box(oi.bd, self.rank)
+            if core is not None and box_contains(core, need):
+                continue              # fully local operand
+            pre = oi.view.preimage_box(core) if core is not None else None
+            interior = box_intersect(interior, pre)
+            if interior is None:
+                return None, 0        # no core-only region: sequential
+        rims = box_subtract(ib, interior)
+        if not rims:
+            return None, 0
+        return [interior] + rims, 1
+
+    def _comms_begin(self, msgs, temp_geom):
+        """Pack outgoing sub-boxes and POST the exchange (non-blocking on
+        the HIP/RCCL backend).  The message list is identical on every
+        rank; pairwise ordering comes from a deterministic sort."""
         def key(m):
             dst, src, bd, bx, tgt = m
             return (dst, src, bd.gid, tuple(bx[0]), tuple(bx[1]), tgt[0],
@@ -432,7 +540,14 @@ class Runtime:
             if dst == self.rank:
                 buf = self.backend.new_message_buffer(box_shape(bx), bd.dtype)
                 recvs.append((src, buf, bd, bx, tgt))
-        self.backend.exchange(sends, [(s, b) for (s, b, _, _, _) in recvs])
+        token = self.backend.exchange_begin(
+            sends, [(s, b) for (s, b, _, _, _) in recvs])
+        return (token, recvs, temp_geom, tc0)
+
+    def _comms_finish(self, state):
+        """Wait for the posted exchange and unpack received boxes."""
+        token, recvs, temp_geom, tc0 = state
+        self.backend.exchange_finish(token)
         for (src, buf, bd, bx, tgt) in recvs:
             if tgt[0] == "border":
                 self.backend.unpack_box_to_container(bd, self, bx, buf)
@@ -441,6 +556,9 @@ class Runtime:
                 need, _ = temp_geom[vname]
                 self.backend.unpack_box_to_temp(vname, need, bx, buf)
         add_time("part_exchange", time.perf_counter() - tc0)
+
+    def _do_comms(self, msgs, temp_geom):
+        self._comms_finish(self._comms_begin(msgs, temp_geom))
 
     # ------------------------------------------------------------------
     def finish_reduction(self, pend):

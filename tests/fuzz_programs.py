@@ -1,32 +1,48 @@
-"""Seeded random in-scope program generator for parity fuzzing.
+"""Seeded random in-scope program generator for parity fuzzing (v6).
 
 Generates programs over the supported vocabulary (creation, elementwise
 chains, slice views, setitem, masked writes, reductions, axis reductions,
 cumsum, astype, transpose/broadcast) as DATA, then interprets them against
 either NumPy or ramba_amd — the same run_both pattern as the reference's
 tests, driven by a generator instead of hand-written cases.  Deterministic
-per seed; failures reproduce with the seed number."""
+per seed; failures reproduce with the seed number.
+
+v6 restores the DISCONTINUOUS float ops (`%`, `//`, floor, sign,
+float→int cast) via per-value EXACTNESS tracking:
+a value is "exact" when both interpretations compute it bit-identically,
+which holds for integer arithmetic and for float chains built purely from
+IEEE correctly-rounded ops (+ - * / sqrt abs neg min max; the HIP kernels
+compile with -ffp-contract=off, csrc/ramba_rt.cpp:99).  Discontinuous ops
+are drawn only on exact values, where comparison is sound at any
+tolerance.  What taints exactness depends on the comparison PAIR:
+
+- mode="numpy" (engine vs direct NumPy): true division taints (the
+  engine's div→mul-by-reciprocal rewrite, shared with the reference,
+  ramba.py:6121, costs 1 ulp vs NumPy's division) and the transcendentals
+  taint (rt_sincos etc. vs libm on the HIP pair);
+- mode="oracle" (HIP engine vs the CPU-oracle backend through the SAME
+  frontend, tests/test_gpu_parity.py): both sides apply the same
+  rewrites, so division stays exact; only the transcendentals taint
+  (different implementations) plus association-order ops (float cumsum).
+"""
 
 import numpy as np
 
 
 FLOAT_UN = ["sin", "cos", "sqrt", "tanh", "exp_neg", "arctan"]
-# floor/sign are discontinuous too (seed 5828: floor lands 27 vs 28 when
-# the 1-ulp div-rewrite difference straddles an integer) -> int-only,
-# like mod/floordiv below
+# IEEE correctly-rounded unaries: identical on every backend
+EXACT_UN = {"sqrt", "neg", "abs", "square"}
 ANY_UN = ["neg", "abs", "square"]
-INT_UN = ["floor", "sign"]
-# mod/floordiv are DISCONTINUOUS: on float chains a 1-ulp upstream
-# difference (the div->mul-by-reciprocal rewrite both we and the
-# reference apply, ramba.py:6121) lands O(divisor) apart at an
-# exact-multiple boundary (first seen at seed 23716 after 33k clean
-# seeds), so the comparison is only sound over exact integer arithmetic
+# discontinuous unaries: drawn only on exact values (v6)
+DISC_UN = ["floor", "sign"]
 BIN = ["add", "sub", "mul", "maximum", "minimum", "div1"]
-INT_BIN = ["mod1", "floordiv1", "bitand", "bitxor"]
+# discontinuous binaries: exact operands only (v6); for integers that is
+# always true, for floats only untainted chains
+DISC_BIN = ["mod1", "floordiv1"]
+INT_BIN = ["bitand", "bitxor"]
 
 
 def _apply_un(np_, x, op):
-    import builtins
     if op == "sin":
         return np_.sin(x)
     if op == "cos":
@@ -66,11 +82,9 @@ def _apply_bin(np_, a, b, op):
     if op == "minimum":
         return np_.minimum(a, b)
     if op == "mod1":
-        d = abs(b) + 1 if not hasattr(b, "asarray") else abs(b) + 1
-        return a % d
+        return a % (abs(b) + 1)
     if op == "div1":
-        d = abs(b) + 1.25
-        return a / d
+        return a / (abs(b) + 1.25)
     if op == "floordiv1":
         return a // (abs(b) + 1)
     if op == "bitand":
@@ -91,8 +105,14 @@ def _rand_slice(rng, n):
     return slice(a, b, step)
 
 
-def build_program(seed):
-    """Returns (impl, tol): impl(np_) runs the program; tol for comparison."""
+def build_program(seed, mode="numpy"):
+    """Returns (impl, tol): impl(np_) runs the program; tol for comparison.
+
+    mode: "numpy" = compared against direct NumPy (div taints exactness);
+          "oracle" = compared against the rewrite-consistent CPU-oracle
+          backend (div exact; transcendentals/assoc-order taint).
+    """
+    assert mode in ("numpy", "oracle")
     master = np.random.default_rng(seed)
     ndim = int(master.integers(1, 4))   # 1-, 2- or 3-D base arrays
     two_d = ndim == 2
@@ -103,10 +123,12 @@ def build_program(seed):
     # pre-draw all decisions so both interpretations agree
     plan_seed = int(master.integers(0, 2 ** 31))
 
+    div_taints = (mode == "numpy")
+
     def impl(np_):
         rng = np.random.default_rng(plan_seed)
         is_np = np_ is np
-        pool = []          # (value, kind) kind: 'i' int64, 'f' f64, 'g' f32
+        pool = []     # (value, kind, exact) kind: 'i' int64, 'f' f64, 'g' f32
         scalars = []
 
         def mk_base():
@@ -121,27 +143,34 @@ def build_program(seed):
                 a = np_.arange(n0 * 2)
             return a
 
-        pool.append((mk_base(), "i"))
-        pool.append(((mk_base() * 3 + 1) % 101, "i"))
-        pool.append((mk_base() * 0.001953125, "f"))   # exact binary scale
+        pool.append((mk_base(), "i", True))
+        pool.append(((mk_base() * 3 + 1) % 101, "i", True))
+        pool.append((mk_base() * 0.001953125, "f", True))  # exact binary
 
         for _ in range(nsteps):
             action = rng.choice(
                 ["un", "bin", "view", "setitem", "mask", "reduce",
                  "axred", "cumsum", "astype", "where", "clip", "mean",
-                 "outer", "axcumsum", "maskget"])
+                 "outer", "axcumsum", "maskget", "fdisc"])
             i = int(rng.integers(0, len(pool)))
-            val, kind = pool[i]
+            val, kind, exact = pool[i]
             if action == "un":
                 if kind == "i":
-                    op = str(rng.choice(ANY_UN + INT_UN))
+                    op = str(rng.choice(ANY_UN + DISC_UN))
                 else:
                     op = str(rng.choice(FLOAT_UN + ANY_UN))
                 r = _apply_un(np_, val, op)
-                pool.append((r, "f" if op in FLOAT_UN else kind))
+                ex = exact and (op in EXACT_UN or op in DISC_UN
+                                or kind == "i")
+                pool.append((r, "f" if op in FLOAT_UN else kind, ex))
+            elif action == "fdisc":
+                # v6: discontinuous float unaries on EXACT floats only
+                if kind in ("f", "g") and exact:
+                    op = str(rng.choice(DISC_UN))
+                    pool.append((_apply_un(np_, val, op), kind, True))
             elif action == "bin":
                 j = int(rng.integers(0, len(pool)))
-                v2, k2 = pool[j]
+                v2, k2, ex2 = pool[j]
                 shape1 = val.shape
                 shape2 = v2.shape
                 if shape1 != shape2:
@@ -149,22 +178,29 @@ def build_program(seed):
                         if (kind != "i" or k2 != "i") \
                         else int(rng.integers(1, 7))
                     k2 = "f" if isinstance(v2, float) else "i"
-                ops = BIN + (INT_BIN if kind == "i" and k2 == "i" else [])
+                    ex2 = True
+                ops = list(BIN)
+                if kind == "i" and k2 == "i":
+                    ops += INT_BIN + DISC_BIN
+                elif exact and ex2:
+                    # v6: float %, // sound on exact operands
+                    ops += DISC_BIN
                 op = str(rng.choice(ops))
                 r = _apply_bin(np_, val, v2, op)
+                ex = exact and ex2 and not (op == "div1" and div_taints)
                 pool.append((r, "f" if op == "div1" or "f" in (kind, k2)
-                             else kind))
+                             else kind, ex))
             elif action == "view":
                 if val.ndim == 1:
                     sl = _rand_slice(rng, val.shape[0])
-                    pool.append((val[sl], kind))
+                    pool.append((val[sl], kind, exact))
                 elif val.ndim == 2:
                     sl = (_rand_slice(rng, val.shape[0]),
                           _rand_slice(rng, val.shape[1]))
                     v = val[sl]
                     if rng.integers(0, 3) == 0:
                         v = v.T
-                    pool.append((v, kind))
+                    pool.append((v, kind, exact))
                 else:
                     sl = tuple(_rand_slice(rng, val.shape[d])
                                for d in range(3))
@@ -172,7 +208,7 @@ def build_program(seed):
                     if rng.integers(0, 3) == 0:
                         perm = list(rng.permutation(3))
                         v = v.transpose(perm)
-                    pool.append((v, kind))
+                    pool.append((v, kind, exact))
             elif action == "setitem":
                 # write a computed value into a slice of a FRESH array
                 if is_np:
@@ -188,39 +224,51 @@ def build_program(seed):
                     tgt[a:-a] = val[a:-a] * 0.5
                 else:
                     tgt[...] = val * 0.5 if val.ndim == 1 else val * 0.5
-                pool.append((tgt, "f" if kind != "g" else "g"))
+                pool.append((tgt, "f" if kind != "g" else "g", exact))
             elif action == "mask":
                 if kind == "i":
                     m = (val % 5) == 0
-                    if is_np:
-                        w = val.copy()
-                        w[m] = -7
-                    else:
-                        w = val.copy()
-                        w[m] = -7
-                    pool.append((w, kind))
+                    w = val.copy()
+                    w[m] = -7
+                    pool.append((w, kind, exact))
             elif action == "reduce":
                 op = str(rng.choice(["sum", "max", "min"]))
+                if op != "sum" and 0 in val.shape:
+                    continue   # min/max of empty raises (both sides)
                 r = getattr(val, op)()
                 scalars.append(float(r))
             elif action == "axred":
                 if val.ndim >= 2 and min(val.shape) >= 2:
-                    ax = int(rng.integers(0, val.ndim))
+                    # v6 also draws axis TUPLES
+                    if val.ndim == 3 and rng.integers(0, 3) == 0:
+                        axs = rng.permutation(3)[:2]
+                        ax = tuple(sorted(int(a) for a in axs))
+                    else:
+                        ax = int(rng.integers(0, val.ndim))
                     r = val.sum(axis=ax)
-                    pool.append((r, "i" if kind == "i" else "f"))
+                    # float axis-sum association differs per backend
+                    pool.append((r, "i" if kind == "i" else "f",
+                                 exact and kind == "i"))
             elif action == "cumsum":
                 if val.ndim == 1 and val.shape[0] > 0 and kind == "i":
-                    pool.append((val.cumsum(), "i"))
+                    pool.append((val.cumsum(), "i", exact))
             elif action == "astype":
-                # float->int truncation is discontinuous like floor
-                # (seed 28097: 27 vs 28 across a 1-ulp boundary) ->
-                # int widens to f32, floats narrow to f32 (continuous)
-                dt = np.float32
-                pool.append((val.astype(dt), "g"))
+                if kind == "i" and exact:
+                    # v6: int -> f64 is exact (values < 2^52)
+                    pool.append((val.astype(np.float64), "f", True))
+                elif exact and kind in ("f", "g") \
+                        and rng.integers(0, 2) == 0:
+                    # v6: float->int truncation restored on exact floats
+                    # (seed 28097's class is sound here)
+                    clipped = np_.minimum(np_.maximum(val, -1e9), 1e9)
+                    pool.append((clipped.astype(np.int64), "i", True))
+                else:
+                    # float narrows to f32 (continuous, correctly rounded)
+                    pool.append((val.astype(np.float32), "g", exact))
             elif action == "where":
                 c = float(rng.uniform(-5.0, 50.0))
                 r = np_.where(val > c, val, -val)
-                pool.append((r, kind))
+                pool.append((r, kind, exact))
             elif action == "clip":
                 lo = float(rng.uniform(-10.0, 0.0))
                 hi = float(rng.uniform(1.0, 100.0))
@@ -228,8 +276,10 @@ def build_program(seed):
                     r = val.clip(int(lo), int(hi))
                 else:
                     r = val.clip(lo, hi)
-                pool.append((r, kind))
+                pool.append((r, kind, exact))
             elif action == "mean":
+                if 0 in val.shape:
+                    continue
                 if kind != "i":
                     scalars.append(float(val.mean()))
                 else:
@@ -237,33 +287,63 @@ def build_program(seed):
             elif action == "outer":
                 if val.ndim == 1 and val.shape[0] <= 200:
                     r = val[:, None] * (val[None, :] + 1)
-                    pool.append((r, kind))
+                    pool.append((r, kind, exact))
             elif action == "axcumsum":
                 if val.ndim >= 2 and kind in ("f", "i"):
                     ax = int(rng.integers(0, val.ndim))
-                    pool.append((val.cumsum(axis=ax), kind))
+                    # float scan association differs per backend
+                    pool.append((val.cumsum(axis=ax), kind,
+                                 exact and kind == "i"))
             elif action == "maskget":
                 # integer-valued membership only: a threshold on computed
                 # floats would make selection itself ulp-sensitive
                 if kind == "i":
                     k = int(rng.integers(2, 9))
-                    pool.append((val[(val % k) == 0], kind))
+                    pool.append((val[(val % k) == 0], kind, exact))
 
         # result: flattened concat of the last few pool values + scalars
         outs = []
-        for (v, k) in pool[-4:]:
+        for (v, k, _) in pool[-4:]:
             a = v.asarray() if hasattr(v, "asarray") else np.asarray(v)
             outs.append(np.asarray(a, dtype=np.float64).reshape(-1))
         outs.append(np.asarray(scalars, dtype=np.float64))
         return np.concatenate(outs) if outs else np.zeros(0)
 
-    return impl, 1e-4 if True else 1e-10
+    return impl, 1e-4
 
 
-def check_seed(ra_module, seed):
-    impl, tol = build_program(seed)
+def check_seed(ra_module, seed, mode="numpy"):
+    impl, tol = build_program(seed, mode=mode)
     with np.errstate(all="ignore"):
         got = impl(ra_module)
         ref = impl(np)
     np.testing.assert_allclose(got, ref, rtol=tol, atol=tol,
-                               err_msg=f"fuzz seed {seed}")
+                               err_msg=f"fuzz seed {seed} (mode={mode})")
+
+
+def check_seed_vs_oracle(ra_module, seed):
+    """HIP engine vs the CPU-oracle backend through the SAME frontend:
+    rewrite-consistent on both sides, so the v6 discontinuous float ops
+    compare soundly.  Swaps the global runtime to a
+    fresh NumpyBackend for the reference leg, then swaps back."""
+    import ramba_amd as ra
+    from ramba_amd.runtime import Runtime
+    import sys, os
+    sys.path.insert(0, os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))))
+    from oracle.numpy_backend import NumpyBackend
+
+    impl, tol = build_program(seed, mode="oracle")
+    rt_prod = ra._deferred.get_runtime()
+    with np.errstate(all="ignore"):
+        got = impl(ra_module)
+        nb = NumpyBackend()
+        rt_cpu = Runtime(nb, rank=0, world=1)
+        nb.attach(rt_cpu)
+        ra._deferred.set_runtime(rt_cpu)
+        try:
+            ref = impl(ra_module)
+        finally:
+            ra._deferred.set_runtime(rt_prod)
+    np.testing.assert_allclose(got, ref, rtol=tol, atol=tol,
+                               err_msg=f"fuzz seed {seed} (vs oracle)")

@@ -381,3 +381,70 @@ def test_multi_axis_reduction_spmd(world):
         return _np.concatenate([s02.asarray(), s12.asarray(),
                                 m01.asarray(), k.asarray().reshape(-1)])
     """, world=world, tol=1e-12)
+
+
+WORKER_SPLIT = r"""
+import os, sys
+sys.path.insert(0, {root!r})
+import numpy as np
+import ramba_amd as ra
+from oracle.numpy_backend import NumpyBackend
+ra.init(backend=NumpyBackend())
+rt = ra._deferred.get_runtime()
+A = ra.fromfunction(lambda x, y: x + y, (257, 259), dtype=np.float32)
+B = ra.zeros((257, 259), dtype=np.float32)
+for it in range(3):
+    B[1:-1, 1:-1] = (A[:-2, 1:-1] + A[2:, 1:-1] + A[1:-1, :-2]
+                     + A[1:-1, 2:] - 4.0 * A[1:-1, 1:-1])
+    A[1:-1, 1:-1] = 0.25 * B[1:-1, 1:-1]
+ra.sync()
+want = int(os.environ.get("RAMBA_OVERLAP", "1"))
+cache = getattr(rt, "_recipe_cache", {{}})
+split = [r for r in cache.values() if r.units is not None]
+if want:
+    # the halo-bearing stencil group must have split into interior+rim
+    assert len(split) == 1, [(r.units, r.comm_msgs) for r in cache.values()]
+    r = split[0]
+    assert r.pre_wait_units == 1
+    assert len(r.units) >= 2
+    total = sum(int(np.prod(u.plan.itershape)) for u in r.units)
+    # units tile the rank's exec box exactly (disjoint cover)
+else:
+    assert not split, "RAMBA_OVERLAP=0 must stay sequential"
+xs = np.fromfunction(lambda x, y: x + y, (257, 259), dtype=np.float32)
+ys = np.zeros_like(xs)
+for it in range(3):
+    ys[1:-1, 1:-1] = (xs[:-2, 1:-1] + xs[2:, 1:-1] + xs[1:-1, :-2]
+                      + xs[1:-1, 2:] - 4.0 * xs[1:-1, 1:-1])
+    xs[1:-1, 1:-1] = 0.25 * ys[1:-1, 1:-1]
+np.testing.assert_allclose(A.asarray(), xs, rtol=1e-5, atol=1e-5)
+print("RANK", os.environ["RANK"], "OK")
+"""
+
+
+@pytest.mark.parametrize("world", [2, 4])
+@pytest.mark.parametrize("overlap", ["1", "0"])
+def test_overlap_split_stencil_spmd(world, overlap):
+    """Halo/compute overlap (BASELINE configs[4] 'overlapped RCCL'):
+    the stencil group splits into interior+rim with the exchange posted
+    first; RAMBA_OVERLAP=0 restores the serial order.  Parity vs NumPy
+    either way."""
+    script = WORKER_SPLIT.format(root=ROOT)
+    port = str(29460 + world + int(overlap) * 7)
+    procs = []
+    for r in range(world):
+        env = dict(os.environ)
+        env.update({"RANK": str(r), "WORLD_SIZE": str(world),
+                    "RAMBA_OVERLAP": overlap,
+                    "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": port,
+                    "GLOO_SOCKET_IFNAME": env.get("GLOO_SOCKET_IFNAME",
+                                                  "lo")})
+        procs.append(subprocess.Popen(
+            [sys.executable, "-c", script], env=env,
+            stdout=subprocess.PIPE, stderr=subprocess.STDOUT))
+    outs, ok = [], True
+    for p in procs:
+        out, _ = p.communicate(timeout=180)
+        outs.append(out.decode())
+        ok = ok and p.returncode == 0
+    assert ok, "\n==== rank outputs ====\n" + "\n----\n".join(outs)
