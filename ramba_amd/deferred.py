@@ -101,6 +101,7 @@ class FusedGroup:
         self._exec_boxes = None
         self.statements = []
         self.arr_vars = {}          # name -> OperandInfo
+        self._var_by_key = {}       # (gid, view) -> var name
         self.vars_by_gid = {}       # gid -> list of var names
         self.scalars = {}           # name -> (value, np.dtype)
         self.reductions = []        # list of (ReductionSpec, PendingReduction)
@@ -118,7 +119,8 @@ class FusedGroup:
 
     def exec_boxes(self):
         if self._exec_boxes is None:
-            self._exec_boxes = exec_boxes(self.part_view, self.part_divs)
+            self._exec_boxes = exec_boxes_cached(self.part_view,
+                                                 self.part_divs)
         return self._exec_boxes
 
     def adopt_partition(self, arr):
@@ -134,15 +136,16 @@ class FusedGroup:
     def arr_var(self, arr, written):
         """Same (gid, view) -> same var (reference add_gid, ramba.py:8078)."""
         gid = arr.bdarray.gid
-        for name in self.vars_by_gid.get(gid, ()):
-            oi = self.arr_vars[name]
-            if oi.view == arr.view:
-                if written:
-                    oi.written = True
-                return name
+        key = (gid, arr.view)
+        name = self._var_by_key.get(key)
+        if name is not None:
+            if written:
+                self.arr_vars[name].written = True
+            return name
         name = self.fresh("v")
         oi = OperandInfo(name, arr.bdarray, arr.view, arr.dtype, written)
         self.arr_vars[name] = oi
+        self._var_by_key[key] = name
         self.vars_by_gid.setdefault(gid, []).append(name)
         self.keepalives.append(arr.bdarray)
         return name
@@ -164,6 +167,23 @@ class FusedGroup:
 # ---------------------------------------------------------------------------
 # module state
 # ---------------------------------------------------------------------------
+
+_eb_cache = {}
+
+
+def exec_boxes_cached(view, divisions):
+    """exec_boxes with a content-keyed cache: _ensure_group recomputes the
+    per-rank preimages for EVERY op of an iterating workload otherwise
+    (~5-10 us x ops x steps of pure frontend overhead)."""
+    key = (view, divisions.shape, divisions.tobytes())
+    r = _eb_cache.get(key)
+    if r is None:
+        r = exec_boxes(view, divisions)
+        if len(_eb_cache) > 4096:
+            _eb_cache.clear()
+        _eb_cache[key] = r
+    return r
+
 
 _state = {"group": None, "runtime": None}
 
@@ -197,7 +217,12 @@ def flush():
 # expression-tree substitution
 # ---------------------------------------------------------------------------
 
+NDARRAY_CLS = None   # set by ndarray.py at import (breaks the cycle)
+
+
 def _is_ndarray(x):
+    if NDARRAY_CLS is not None:
+        return type(x) is NDARRAY_CLS
     return hasattr(x, "bdarray") and hasattr(x, "view")
 
 
@@ -268,7 +293,7 @@ def _ensure_group(arr):
             flush()
             g = None
         elif not arr_flex and not g.flex:
-            boxes = exec_boxes(arr.view, arr.bdarray.divisions)
+            boxes = exec_boxes_cached(arr.view, arr.bdarray.divisions)
             if not exec_boxes_eq(g.exec_boxes(), boxes):
                 dprint(2, "deferred ops partition mismatch; flushing")
                 flush()

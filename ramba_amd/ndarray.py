@@ -206,16 +206,32 @@ class ndarray:
                     base[b] += i * st
             rt = deferred.get_runtime()
             return rt.read_element(self.bdarray, base)
-        try:
-            key = (self.view, index)
+        # slices are unhashable before Python 3.12, so normalise the index
+        # to a hashable key (the r1 cache keyed on raw slices and silently
+        # never hit on 3.10 — every getitem recomputed apply_index)
+        key = None
+        parts = []
+        for ix in index:
+            t = type(ix)
+            if t is slice:
+                parts.append((ix.start, ix.stop, ix.step))
+            elif t is int or ix is None or ix is Ellipsis:
+                parts.append(ix)
+            elif isinstance(ix, (int, np.integer)):
+                parts.append(int(ix))
+            else:
+                parts = None
+                break
+        if parts is not None:
+            key = (self.view, tuple(parts))
             view = _getitem_cache.get(key)
-            if view is None:
-                view = self.view.apply_index(index)
-                if len(_getitem_cache) > 8192:
-                    _getitem_cache.clear()
-                _getitem_cache[key] = view
-        except TypeError:       # unhashable index element
-            view = self.view.apply_index(index)
+            if view is not None:
+                return self._with_view(view)
+        view = self.view.apply_index(index)
+        if key is not None:
+            if len(_getitem_cache) > 8192:
+                _getitem_cache.clear()
+            _getitem_cache[key] = view
         return self._with_view(view)
 
     def __setitem__(self, index, value):
@@ -367,7 +383,13 @@ class ndarray:
         if not isinstance(rhs, (ndarray, numbers.Number, np.generic)):
             return NotImplemented
         a, b = (rhs, self) if reverse else (self, rhs)
-        shape = np.broadcast_shapes(_shape_of(a), _shape_of(b))
+        sa, sb = _shape_of(a), _shape_of(b)
+        if sa == sb or sb == ():
+            shape = sa          # fast path: np.broadcast_shapes is ~10 us
+        elif sa == ():
+            shape = sb
+        else:
+            shape = np.broadcast_shapes(sa, sb)
         dt = _result_dtype(op, a, b) if out_dtype is None else out_dtype
         a, b = _bcast(a, shape), _bcast(b, shape)
         if op == "pow" and isinstance(b, (int, np.integer)) \
@@ -633,6 +655,9 @@ class ndarray:
         axes.remove(axis)
         axes.insert(start, axis)
         return self.transpose(axes)
+
+
+deferred.NDARRAY_CLS = ndarray   # fast leaf test in deferred._subst
 
 
 # -- attach the op tables (reference make_method loops, ramba.py:7893-7973) --

@@ -164,13 +164,28 @@ class Runtime:
     # geometry depends on (views, divisions CONTENT, liveness, statements).
     # ------------------------------------------------------------------
 
+    _divb_cache = {}
+
+    @classmethod
+    def _div_bytes(cls, divs):
+        """divisions.tobytes() memoised by array identity (divisions are
+        rebound, never mutated; the ref in the cache keeps the id valid)."""
+        e = cls._divb_cache.get(id(divs))
+        if e is not None and e[0] is divs:
+            return e[1]
+        b = divs.tobytes()
+        if len(cls._divb_cache) > 8192:
+            cls._divb_cache.clear()
+        cls._divb_cache[id(divs)] = (divs, b)
+        return b
+
     def _group_signature(self, group, live):
         parts = [group.shape, self.world,
-                 ("part", group.part_view, group.part_divs.tobytes(),
-                  group.flex)]
+                 ("part", group.part_view,
+                  self._div_bytes(group.part_divs), group.flex)]
         for name, oi in group.arr_vars.items():
             parts.append((name, name in live, oi.bd.shape, oi.bd.dtype,
-                          oi.bd.border, oi.bd.divisions.tobytes(),
+                          oi.bd.border, self._div_bytes(oi.bd.divisions),
                           oi.bd.is_flex, oi.bd.constructed, oi.view,
                           oi.written))
         parts.append(tuple(sorted((n, dt.name)

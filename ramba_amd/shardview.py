@@ -96,12 +96,36 @@ def box_subtract(a, b):
 # View
 # ---------------------------------------------------------------------------
 
-@dataclass(frozen=True)
+_identity_cache = {}
+
+
+@dataclass(frozen=True, eq=False)
 class View:
     shape: tuple      # view shape
     axis_map: tuple   # len(shape); base axis index, or -1 for a new axis
     steps: tuple      # len(shape); 0 = broadcast over the mapped base axis
     offset: tuple     # len = base ndim; base coord of view index 0 / fixed coord
+
+    def __eq__(self, other):
+        # identity fast path: the getitem/identity caches intern Views, so
+        # repeated steps compare the same objects
+        if self is other:
+            return True
+        if other.__class__ is not View:
+            return NotImplemented
+        return (self.shape == other.shape
+                and self.axis_map == other.axis_map
+                and self.steps == other.steps
+                and self.offset == other.offset)
+
+    def __hash__(self):
+        # memoised: Views key the hot frontend caches (getitem, arr_var,
+        # exec-boxes) and the dataclass hash re-hashes 4 tuples per call
+        h = self.__dict__.get("_h")
+        if h is None:
+            h = hash((self.shape, self.axis_map, self.steps, self.offset))
+            object.__setattr__(self, "_h", h)
+        return h
 
     @property
     def ndim(self):
@@ -113,9 +137,15 @@ class View:
 
     @staticmethod
     def identity(shape):
-        nd = len(shape)
-        return View(tuple(int(s) for s in shape), tuple(range(nd)),
-                    (1,) * nd, (0,) * nd)
+        shape = tuple(int(s) for s in shape)
+        v = _identity_cache.get(shape)
+        if v is None:
+            nd = len(shape)
+            v = View(shape, tuple(range(nd)), (1,) * nd, (0,) * nd)
+            if len(_identity_cache) > 4096:
+                _identity_cache.clear()
+            _identity_cache[shape] = v
+        return v
 
     def is_identity_for(self, base_shape):
         return (self.shape == tuple(base_shape)
