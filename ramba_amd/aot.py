@@ -173,8 +173,10 @@ def seed(fuzz_seeds=None):
     Y[1:-1, 1:-1] = (X[:-2, 1:-1] + X[2:, 1:-1] + X[1:-1, :-2]
                      + X[1:-1, 2:] - 4.0 * X[1:-1, 1:-1])
     ra.sync()
-    # configs[4] mixed fp64
+    # configs[4] mixed fp64 (Z materialised first so the pair takes the
+    # staged/tiled path, like the bench loop's steady state)
     Z = ra.zeros((64, 64), dtype=np.float64)
+    ra.sync()
     src = ra.fromfunction(lambda x, y: (x * 64 + y) * 1e-6, (64, 64),
                           dtype=np.float64)
     ss = ra.sin(src)
@@ -199,3 +201,27 @@ def seed(fuzz_seeds=None):
     n = be.compiled
     ra.shutdown()
     return n
+
+
+# -- cross-stage fusion: compile-only hooks (ramba_amd/staged.py) ------------
+
+AotCompileBackend.supports_staged = True
+
+
+def _aot_container_addr(self, bd):
+    return 0
+
+
+def _aot_tiled_kernel(self, desc):
+    key, source, kname, fields = codegen.generate_staged_tiled(desc)
+    self._cc(source)
+    return (key, fields)
+
+
+def _aot_tiled_launch(self, handle, vals, ntiles):
+    pass
+
+
+AotCompileBackend.container_addr = _aot_container_addr
+AotCompileBackend.tiled_kernel = _aot_tiled_kernel
+AotCompileBackend.tiled_launch = _aot_tiled_launch

@@ -1357,3 +1357,208 @@ def pack_axis_reduce_args(fields, out_extents_by_axis, red_extents_by_axis,
             nchunk, clen, ktot = chunk_spec
             out += struct.pack("<qqq", nchunk, clen, ktot)
     return bytes(out)
+
+
+# ---------------------------------------------------------------------------
+# staged/tiled kernel (cross-stage fusion, BASELINE configs[4]): one
+# workgroup computes an index-pure producer stage over a tile+halo
+# FOOTPRINT into LDS (recomputing halo values instead of exchanging
+# them), stores the producer's live outputs for the footprint cells its
+# containers back, then computes the consumer stage over the tile from
+# LDS.  See ramba_amd/staged.py for the orchestration and the
+# sequential-fallback contract.
+# ---------------------------------------------------------------------------
+
+TILE_H = 16
+TILE_W = 64
+
+
+class _StageGen:
+    """Minimal `gen` shim for LaneEmitter: resolves Refs for one stage."""
+
+    def __init__(self, stmts, resolve0, dead):
+        self.resolve0 = resolve0      # fn(base_name) -> C expr (version 0)
+        self.dead = dead
+        self.ssa, self.final_version = ssa_statements(stmts, set())
+        self.wanted = {}
+        for (_, _, e) in self.ssa:
+            collect_nodes(e, self.wanted)
+
+    def lane_value(self, em, versioned, dtype):
+        base, v = versioned.split("@")
+        if int(v) == 0:
+            r = self.resolve0(base, em)
+            if r is None:
+                raise AssertionError(f"unresolvable var {base}")
+            return r
+        return f"r_{base}_{v}{em.tag}"
+
+    def emit_into(self, em):
+        """Emit all statements; returns {base: final C value}."""
+        for (tgt, _, e) in self.ssa:
+            val = em.emit(e)
+            base, v = tgt.split("@")
+            ct = ctype(e.dtype)
+            em.lines.append(f"      {ct} r_{base}_{v}{em.tag} = {val};")
+        out = {}
+        for base, v in self.final_version.items():
+            out[base] = f"r_{base}_{v}{em.tag}"
+        return out
+
+
+def generate_staged_tiled(desc):
+    """desc (structural only; runtime values go through the fields):
+      s1_stmts, s2_stmts : lists of ir.Assign (stage1 names p_-prefixed)
+      staged : [(lds, dtype, live)]  # one per staged gid, emit order;
+               `lds` is also the stage-1 var base name (p_<writer var>)
+      s1_stores : [(var, dtype)]     # non-staged live stage-1 outputs
+      readers : {stage2_var: (lds, dr0, dr1)}
+      s2_ops : [(var, dtype, written)]
+      scalars : {name: np.dtype}     # merged, stage-1 names p_-prefixed
+      E0, E1 : global footprint extents (>= all reader deltas)
+    returns (key, source, kname, fields): fields = ordered packing names.
+    """
+    E0, E1 = desc["E0"], desc["E1"]
+    TH, TW = TILE_H, TILE_W
+    FH, FW = TH + E0, TW + E1
+
+    key = hashlib.sha256(repr((
+        [(st.target, st.expr) for st in desc["s1_stmts"]],
+        [(st.target, st.expr) for st in desc["s2_stmts"]],
+        desc["staged"], desc["s1_stores"], sorted(desc["readers"].items()),
+        desc["s2_ops"], sorted((n, str(d)) for n, d in
+                               desc["scalars"].items()),
+        E0, E1, TH, TW)).encode()).hexdigest()[:24]
+    kname = f"tk_{key}"
+
+    fields = [("n0", "q"), ("n1", "q"), ("gs0", "q"), ("gs1", "q"),
+              ("gb0", "q"), ("gb1", "q"), ("N0", "q"), ("N1", "q")]
+    L = [PREAMBLE]
+    L.append("struct TkArgs {")
+    L.append("  i64 n0, n1;      // consumer exec-box extents")
+    L.append("  i64 gs0, gs1;    // consumer global start (Iota)")
+    L.append("  i64 gb0, gb1;    // base coord of footprint origin at k=0")
+    L.append("  i64 N0, N1;      // producer-array global shape")
+    for (lds, dt, live) in desc["staged"]:
+        if live:
+            for f in (f"{lds}_ptr", f"{lds}_off", f"{lds}_s0", f"{lds}_s1",
+                      f"{lds}_lo0", f"{lds}_hi0", f"{lds}_lo1",
+                      f"{lds}_hi1"):
+                fields.append((f, "Q" if f.endswith("ptr") else "q"))
+            L.append(f"  {ctype(dt)}* __restrict__ {lds}_ptr; "
+                     f"i64 {lds}_off, {lds}_s0, {lds}_s1, "
+                     f"{lds}_lo0, {lds}_hi0, {lds}_lo1, {lds}_hi1;")
+    for (var, dt) in desc["s1_stores"]:
+        for f in (f"{var}_ptr", f"{var}_off", f"{var}_s0", f"{var}_s1",
+                  f"{var}_lo0", f"{var}_hi0", f"{var}_lo1", f"{var}_hi1"):
+            fields.append((f, "Q" if f.endswith("ptr") else "q"))
+        L.append(f"  {ctype(dt)}* __restrict__ {var}_ptr; "
+                 f"i64 {var}_off, {var}_s0, {var}_s1, "
+                 f"{var}_lo0, {var}_hi0, {var}_lo1, {var}_hi1;")
+    for (var, dt, written) in desc["s2_ops"]:
+        for f in (f"{var}_ptr", f"{var}_off", f"{var}_s0", f"{var}_s1"):
+            fields.append((f, "Q" if f.endswith("ptr") else "q"))
+        L.append(f"  {ctype(dt)}* __restrict__ {var}_ptr; "
+                 f"i64 {var}_off, {var}_s0, {var}_s1;")
+    for n in sorted(desc["scalars"]):
+        dt = desc["scalars"][n]
+        if np.dtype(dt).kind == "f":
+            fields.append((n, "d"))
+            L.append(f"  double {n};")
+        else:
+            fields.append((n, "q"))
+            L.append(f"  i64 {n};")
+    L.append("};")
+
+    L.append(f'extern "C" __global__ void __launch_bounds__(256) '
+             f"{kname}(TkArgs a) {{")
+    for (lds, dt, live) in desc["staged"]:
+        L.append(f"  __shared__ {ctype(dt)} lds_{lds}[{FH}][{FW} + 1];")
+    L.append(f"  const i64 tiles1 = (a.n1 + {TW} - 1) / {TW};")
+    L.append(f"  const i64 tiles0 = (a.n0 + {TH} - 1) / {TH};")
+    L.append("  for (i64 tile = blockIdx.x; tile < tiles0 * tiles1; "
+             "tile += gridDim.x) {")
+    L.append("    const i64 k0o = (tile / tiles1) * %d;" % TH)
+    L.append("    const i64 k1o = (tile %% tiles1) * %d;" % TW)
+
+    # ---- stage 1: footprint fill + stores --------------------------------
+    s1gen = _StageGen(desc["s1_stmts"], lambda b, em: None, desc["dead1"])
+    L.append(f"    for (int fi = threadIdx.x; fi < {FH} * {FW}; fi += 256) "
+             "{")
+    L.append(f"      const int f0 = fi / {FW}, f1 = fi % {FW};")
+    L.append("      const i64 B0 = a.gb0 + k0o + f0;")
+    L.append("      const i64 B1 = a.gb1 + k1o + f1;")
+    L.append("      if (B0 < 0 || B0 >= a.N0 || B1 < 0 || B1 >= a.N1) "
+             "continue;")
+    em1 = LaneEmitter(s1gen, "_s1", ["B0", "B1"])
+    finals = s1gen.emit_into(em1)
+    body = list(em1.lines)
+    for (lds, dt, live) in desc["staged"]:
+        body.append(f"      lds_{lds}[f0][f1] = {finals[lds]};")
+        if live:
+            body.append(
+                f"      if (B0 >= a.{lds}_lo0 && B0 <= a.{lds}_hi0 && "
+                f"B1 >= a.{lds}_lo1 && B1 <= a.{lds}_hi1)")
+            body.append(
+                f"        a.{lds}_ptr[a.{lds}_off + B0 * a.{lds}_s0 + "
+                f"B1 * a.{lds}_s1] = {finals[lds]};")
+    for (var, dt) in desc["s1_stores"]:
+        body.append(
+            f"      if (B0 >= a.{var}_lo0 && B0 <= a.{var}_hi0 && "
+            f"B1 >= a.{var}_lo1 && B1 <= a.{var}_hi1)")
+        body.append(
+            f"        a.{var}_ptr[a.{var}_off + B0 * a.{var}_s0 + "
+            f"B1 * a.{var}_s1] = {finals[var]};")
+    L.extend(body)
+    L.append("    }")
+    L.append("    __syncthreads();")
+
+    # ---- stage 2: tile compute from LDS ----------------------------------
+    readers = desc["readers"]
+    s2_op_names = {v for (v, _, _) in desc["s2_ops"]}
+
+    def resolve2(base, em):
+        if base in readers:
+            lds, dr0, dr1 = readers[base]
+            return (f"lds_{lds}[q0{em.tag} + {dr0}]"
+                    f"[q1{em.tag} + {dr1}]")
+        if base in s2_op_names:
+            return (f"a.{base}_ptr[a.{base}_off + k0{em.tag} * a.{base}_s0"
+                    f" + k1{em.tag} * a.{base}_s1]")
+        return None
+
+    s2gen = _StageGen(desc["s2_stmts"], resolve2, desc["dead2"])
+    L.append(f"    for (int ti = threadIdx.x; ti < {TH} * {TW}; ti += 256) "
+             "{")
+    L.append(f"      const int q0_t2 = ti / {TW}, q1_t2 = ti % {TW};")
+    L.append("      const i64 k0_t2 = k0o + q0_t2;")
+    L.append("      const i64 k1_t2 = k1o + q1_t2;")
+    L.append("      if (k0_t2 < a.n0 && k1_t2 < a.n1) {")
+    em2 = LaneEmitter(s2gen, "_t2",
+                      ["(a.gs0 + k0_t2)", "(a.gs1 + k1_t2)"])
+    finals2 = s2gen.emit_into(em2)
+    L.extend(em2.lines)
+    for (var, dt, written) in desc["s2_ops"]:
+        if written:
+            L.append(
+                f"      a.{var}_ptr[a.{var}_off + k0_t2 * a.{var}_s0 + "
+                f"k1_t2 * a.{var}_s1] = {finals2[var]};")
+    L.append("      }")
+    L.append("    }")
+    L.append("    __syncthreads();")
+    L.append("  }")
+    L.append("}")
+    return key, "\n".join(L), kname, fields
+
+
+def pack_tk_args(fields, values):
+    out = bytearray()
+    for (name, kind) in fields:
+        v = values[name]
+        if kind == "Q":
+            out += struct.pack("<Q", int(v))
+        elif kind == "q":
+            out += struct.pack("<q", int(v))
+        else:
+            out += struct.pack("<d", float(v))
+    return bytes(out)

@@ -32,6 +32,11 @@ debug_poison = int(os.environ.get("RAMBA_DEBUG_POISON", "0"))
 # — BASELINE configs[4] "overlapped RCCL"; 0 restores the reference-style
 # serial exchange-then-execute (ramba.py:3547-3693) for A/B measurement
 overlap_exchange = int(os.environ.get("RAMBA_OVERLAP", "1"))
+# cross-stage fusion (BASELINE configs[4] "fused-kernel"): a shifted read
+# of an index-pure in-group-written array seals the group as a producer
+# and the pair fuses into one LDS-tiled kernel on the HIP backend; 0
+# restores the reference's flush-at-alias behaviour (ramba.py:8434-8443)
+stage_fusion = int(os.environ.get("RAMBA_STAGE_FUSION", "1"))
 
 
 def dprint(level, *args):

@@ -110,6 +110,15 @@ class FusedGroup:
         self.read_views = []        # (gid, view) every read
         self.write_views = []       # (gid, view) every write
         self._counter = 0
+        # producer/consumer staging (BASELINE configs[4] cross-stage
+        # fusion): when a shifted read of an in-group-written INDEX-PURE
+        # array arrives, the group is SEALED as this group's `producer`
+        # instead of flushed; the runtime fuses the pair into one
+        # LDS-tiled kernel (or falls back to sequential execution, which
+        # is exactly the reference's flush-at-alias behaviour,
+        # ramba.py:8434-8443)
+        self.producer = None        # sealed stage-1 FusedGroup or None
+        self.staged_gids = set()    # gids served from LDS recompute
 
     # -- naming -------------------------------------------------------------
 
@@ -205,12 +214,16 @@ def current_group():
 
 
 def flush():
-    """do_ops (reference ramba.py:8332): execute the pending fused group."""
+    """do_ops (reference ramba.py:8332): execute the pending fused group
+    (or the sealed producer/consumer pair)."""
     g = _state["group"]
     if g is None:
         return
     _state["group"] = None
-    get_runtime().execute_group(g)
+    if g.producer is not None:
+        get_runtime().execute_staged(g.producer, g)
+    else:
+        get_runtime().execute_group(g)
 
 
 # ---------------------------------------------------------------------------
@@ -307,6 +320,72 @@ def _ensure_group(arr):
     return g
 
 
+def _expr_refs(e, out):
+    if isinstance(e, ir.Ref):
+        out.append(e.name)
+    elif isinstance(e, ir.Bin):
+        _expr_refs(e.a, out)
+        _expr_refs(e.b, out)
+    elif isinstance(e, (ir.Un, ir.Cast)):
+        _expr_refs(e.a, out)
+    elif isinstance(e, ir.Where):
+        _expr_refs(e.c, out)
+        _expr_refs(e.a, out)
+        _expr_refs(e.b, out)
+    return out
+
+
+def group_is_index_pure(g):
+    """True if every statement's leaves are Iota/Const/scalars or refs to
+    vars written EARLIER IN THE GROUP — i.e. the whole group is a pure
+    function of the global index, recomputable on any rank (the halo-
+    recompute precondition for cross-stage fusion)."""
+    written = set()
+    for st in g.statements:
+        for name in _expr_refs(st.expr, []):
+            if name not in written:
+                return False
+        written.add(st.target)
+    return True
+
+
+def _stageable(g, conflicts, write_arr, operands):
+    """May `g` become the sealed producer of a new consumer group?"""
+    from .common import stage_fusion
+    if (not stage_fusion or g.producer is not None or g.reductions
+            or len(g.shape) != 2):
+        return False
+    # a write-after-read hazard on this op goes through the temp dance
+    # (alias check 2), which would lose the seal: don't seal then
+    if any(o.bdarray.gid == write_arr.bdarray.gid
+           and o.view != write_arr.view for o in operands):
+        return False
+    # the consumer's write target must not alias anything the producer
+    # touches (the fused kernel interleaves the stages per tile)
+    if write_arr.bdarray.gid in g.vars_by_gid:
+        return False
+    nd = len(g.shape)
+    # every producer write must be an identity view (base coords == group
+    # iteration coords, so halo recompute uses plain global indices)
+    for (wgid, wview) in g.write_views:
+        bshape = None
+        for name in g.vars_by_gid.get(wgid, ()):
+            bshape = g.arr_vars[name].bd.shape
+            break
+        if bshape is None or not wview.is_identity_for(bshape):
+            return False
+    # conflicting reads must be pure small shifts of the identity layout
+    for (o, wgid, wview) in conflicts:
+        v = o.view
+        if o is write_arr:
+            return False           # consumer writing a producer array
+        if v.ndim != nd or v.axis_map != tuple(range(nd))                 or v.steps != (1,) * nd:
+            return False
+        if any(not (0 <= off <= 16) for off in v.offset):
+            return False
+    return group_is_index_pure(g)
+
+
 def add_op(write_arr, assign_op, rhs_tree, empty_like=None):
     """Record `write_arr <assign_op> rhs_tree` (or a pure read when
     write_arr is None is not allowed -- every statement has a target).
@@ -321,18 +400,30 @@ def add_op(write_arr, assign_op, rhs_tree, empty_like=None):
     g = _state["group"]
 
     # Alias check 1 (ramba.py:8434-8443): an operand reads/writes a shifted
-    # version of an array written earlier in this group -> flush first.
+    # version of an array written earlier in this group.  If the group is
+    # an index-pure elementwise producer and the reads are pure shifts,
+    # SEAL it as the next group's producer (cross-stage fusion); else
+    # flush first (the reference's behaviour).
+    sealed = None
     if g is not None:
         check = operands + ([write_arr] if write_arr is not None else [])
+        conflicts = []
         for o in check:
             for (wgid, wview) in g.write_views:
                 if o.bdarray.gid == wgid and o.view != wview:
-                    dprint(2, "RAW with mismatched views; flushing")
-                    flush()
-                    g = None
-                    break
-            if g is None:
-                break
+                    conflicts.append((o, wgid, wview))
+        if conflicts:
+            if write_arr is not None and _stageable(g, conflicts,
+                                                    write_arr, operands):
+                dprint(2, "RAW with mismatched views; sealing producer")
+                sealed = g
+                sealed_gids = {wgid for (_, wgid, _) in conflicts}
+                _state["group"] = None
+                g = None
+            else:
+                dprint(2, "RAW with mismatched views; flushing")
+                flush()
+                g = None
 
     # Alias check 2 (ramba.py:8445-8465): the write target is a shifted
     # version of something read (in this op or earlier in the group) ->
@@ -354,6 +445,10 @@ def add_op(write_arr, assign_op, rhs_tree, empty_like=None):
             return
 
     g = _ensure_group(arr)
+    if sealed is not None:
+        assert g.producer is None
+        g.producer = sealed
+        g.staged_gids = sealed_gids
 
     # lower compound assignment: W op= rhs  ->  W = W op rhs
     if assign_op != "=":
@@ -422,7 +517,7 @@ def expr_dtype(e):
 # liveness at flush (reference live_gids filter, ramba.py:8123)
 # ---------------------------------------------------------------------------
 
-def compute_live_vars(group):
+def compute_live_vars(group, extra_live_gids=frozenset()):
     """Partition arr_vars into materialised operands and register temps.
 
     A var is LIVE (materialises) iff its backing array still has external
@@ -434,7 +529,7 @@ def compute_live_vars(group):
     live, dead = {}, {}
     for name, oi in group.arr_vars.items():
         alive = (oi.bd.nviews > 0 and oi.bd.gid not in delete_gids) \
-            or oi.bd.constructed
+            or oi.bd.constructed or oi.bd.gid in extra_live_gids
         if alive:
             live[name] = oi
         else:

@@ -896,3 +896,59 @@ def _hb_flat_scatter(self, cont, off0, strides, shape, flat0, buf):
 
 HipBackend.flat_gather = _hb_flat_gather
 HipBackend.flat_scatter = _hb_flat_scatter
+
+
+# -- cross-stage fusion (staged/tiled kernel; ramba_amd/staged.py) -----------
+
+HipBackend.supports_staged = True
+
+
+def _hb_container_addr(self, bd):
+    return self._cont(bd).data_ptr()
+
+
+def _hb_tiled_kernel(self, desc):
+    """Build (or fetch) the staged/tiled kernel; returns a handle token
+    or None (fall back to sequential)."""
+    from . import codegen as cg
+    try:
+        key, source, kname, fields = cg.generate_staged_tiled(desc)
+    except NotImplementedError:
+        return None
+    ck = ("tiled", key)
+    cached = self.kernels.get(ck)
+    if cached is None:
+        if int(os.environ.get("RAMBA_SHOW_CODE", "0")):
+            print(f"=== tiled kernel {key} ===\n{source}\n", flush=True)
+        h = ctypes.c_void_p()
+        self._check(self.lib.rt_kernel_get(
+            key.encode(), source.encode(), kname.encode(),
+            ctypes.byref(h)), "rt_kernel_get(tiled)")
+        cached = (h.value, fields)
+        self.kernels[ck] = cached
+    return cached
+
+
+def _hb_tiled_launch(self, handle, vals, ntiles):
+    from . import codegen as cg
+    h, fields = handle
+    args = cg.pack_tk_args(fields, vals)
+    gx = max(1, min(int(os.environ.get("RAMBA_GRID_CAP", "16384")),
+                    ntiles))
+    if self.time_kernels:
+        ev0 = self.torch.cuda.Event(enable_timing=True)
+        ev1 = self.torch.cuda.Event(enable_timing=True)
+        ev0.record()
+    self._check(self.lib.rt_launch(
+        ctypes.c_void_p(h), gx, 1, 1, 256, self._stream(), args,
+        len(args)), "rt_launch(tiled)")
+    if self.time_kernels:
+        ev1.record()
+        ev1.synchronize()
+        self.kernel_times_ms.append(ev0.elapsed_time(ev1))
+        self.kernel_keys.append("tiled")
+
+
+HipBackend.container_addr = _hb_container_addr
+HipBackend.tiled_kernel = _hb_tiled_kernel
+HipBackend.tiled_launch = _hb_tiled_launch

@@ -576,6 +576,66 @@ class Runtime:
         self._comms_finish(self._comms_begin(msgs, temp_geom))
 
     # ------------------------------------------------------------------
+    # cross-stage fusion (BASELINE configs[4]): a sealed producer group
+    # (index-pure elementwise, identity writes) + its consumer group
+    # (shifted reads of the producer's outputs).  The HIP backend fuses
+    # the pair into ONE LDS-tiled kernel — halo values are RECOMPUTED
+    # from index arithmetic, so the halo exchange for staged arrays
+    # disappears entirely.  Every other case (and every non-HIP backend)
+    # falls back to sequential pair execution, which is byte-identical
+    # to the reference's flush-at-alias behaviour (ramba.py:8434-8443).
+    # ------------------------------------------------------------------
+
+    def execute_staged(self, g1, g2):
+        t0 = time.perf_counter()
+        try:
+            if not self._try_execute_tiled(g1, g2):
+                self._execute_pair_sequential(g1, g2)
+        finally:
+            add_time("run_deferred_ops", time.perf_counter() - t0)
+
+    def _execute_pair_sequential(self, g1, g2):
+        # arrays the consumer reads from HBM pin the producer's stores
+        # even if their last python ref died while the pair was pending
+        extra = {oi.bd.gid for oi in g2.arr_vars.values()}
+        live, dead = deferred.compute_live_vars(g1, extra_live_gids=extra)
+        # consumer-read gids must not be freed by the producer's deletes
+        keep = [bd for bd in g1.delete_bds if bd.gid in extra]
+        g1.delete_bds = [bd for bd in g1.delete_bds if bd.gid not in extra]
+        g2.delete_bds.extend(keep)
+        self._execute_group_with(g1, live, dead)
+        self._execute_group(g2)
+
+    def _execute_group_with(self, group, live, dead):
+        """_execute_group with a precomputed liveness split (recipe cache
+        keyed on the extra-live-adjusted liveness)."""
+        cache = getattr(self, "_recipe_cache", None)
+        if cache is None:
+            cache = self._recipe_cache = {}
+        try:
+            sig = self._group_signature(group, live)
+        except TypeError:
+            sig = None
+        recipe = cache.get(sig) if sig is not None else None
+        if recipe is not None:
+            self._run_recipe(recipe, group, live)
+            return
+        recipe = self._build_and_run(group, live, dead)
+        if sig is not None and recipe is not None:
+            if len(cache) > 128:
+                cache.clear()
+            cache[sig] = recipe
+
+    def _try_execute_tiled(self, g1, g2):
+        """Fused LDS-tiled execution of the pair; False -> caller falls
+        back to sequential."""
+        backend = self.backend
+        if not getattr(backend, "supports_staged", False):
+            return False
+        from . import staged
+        return staged.try_execute_tiled(self, g1, g2)
+
+    # ------------------------------------------------------------------
     def finish_reduction(self, pend):
         val = pend.partial
         if self.world > 1:

@@ -20,6 +20,8 @@ pytestmark = pytest.mark.gpu
 HERE = os.path.dirname(os.path.abspath(__file__))
 ROOT = os.path.dirname(HERE)
 
+_PORT_COUNTER = [0]
+
 WORKER = r"""
 import os, sys
 sys.path.insert(0, {root!r})
@@ -62,7 +64,8 @@ needs_multi_gpu = pytest.mark.skipif(
 def run_spmd_gpu(body_src, world=2, tol=None, backend="gloo"):
     body = textwrap.indent(textwrap.dedent(body_src).strip(), "    ")
     script = WORKER.format(root=ROOT, body=body, tol=tol)
-    port = str(29700 + (hash((body, world, backend)) % 200))
+    _PORT_COUNTER[0] += 1
+    port = str(26000 + (os.getpid() * 7 + _PORT_COUNTER[0] * 13) % 3000)
     ndev = max(1, _device_count())
     procs = []
     for r in range(world):

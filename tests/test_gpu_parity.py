@@ -517,3 +517,142 @@ def test_reshape_gpu(ra_gpu):
         return _np.concatenate([s1.asarray(), r2.asarray()[::997],
                                 r3.asarray().reshape(-1)[::997]])
     run_both(impl, ra_gpu, tol=1e-9)
+
+
+class TestStagedFusion:
+    """Cross-stage fusion (ramba_amd/staged.py): producer recomputed into
+    LDS, consumer stencil reads LDS — parity vs plain NumPy, plus a
+    white-box check that the tiled path (not the sequential fallback)
+    actually ran."""
+
+    def _spy(self):
+        import ramba_amd.staged as st
+        calls = []
+        orig = st._execute_tiled
+
+        def wrap(*a, **k):
+            r = orig(*a, **k)
+            calls.append(r)
+            return r
+        st._execute_tiled = wrap
+        return calls, lambda: setattr(st, "_execute_tiled", orig)
+
+    def test_mixed_pipeline_staged_live(self, ra_gpu):
+        S = 512
+        calls, restore = self._spy()
+        try:
+            A = ra_gpu.zeros((S, S), dtype=np.float64)
+            ra_gpu.sync()
+            src = ra_gpu.fromfunction(
+                lambda x, y: (x * S + y) * 1e-6, (S, S), dtype=np.float64)
+            ssin = ra_gpu.sin(src)
+            A[1:-1, 1:-1] = (ssin[:-2, 1:-1] + ssin[2:, 1:-1]
+                             + ssin[1:-1, :-2] + ssin[1:-1, 2:]
+                             - 4.0 * ssin[1:-1, 1:-1])
+            s = float(A.sum())
+        finally:
+            restore()
+        assert calls and calls[-1] is True, f"tiled path not taken: {calls}"
+        a = np.zeros((S, S))
+        sr = np.fromfunction(lambda x, y: (x * S + y) * 1e-6, (S, S))
+        ss = np.sin(sr)
+        a[1:-1, 1:-1] = (ss[:-2, 1:-1] + ss[2:, 1:-1] + ss[1:-1, :-2]
+                         + ss[1:-1, 2:] - 4.0 * ss[1:-1, 1:-1])
+        np.testing.assert_allclose(A.asarray(), a, rtol=1e-12, atol=1e-12)
+        np.testing.assert_allclose(ssin.asarray(), ss, rtol=1e-12,
+                                   atol=1e-12)     # live staged var stored
+        np.testing.assert_allclose(src.asarray(), sr, rtol=1e-12,
+                                   atol=1e-12)     # non-staged live store
+        assert abs(s - a.sum()) < 1e-7 * abs(a.sum())
+
+    def test_mixed_pipeline_staged_dead_intermediates(self, ra_gpu):
+        S = 300
+        calls, restore = self._spy()
+        try:
+            A = ra_gpu.zeros((S, S), dtype=np.float64)
+            ra_gpu.sync()
+            src = ra_gpu.fromfunction(
+                lambda x, y: (x + 2.0 * y) * 1e-3, (S, S), dtype=np.float64)
+            sc = ra_gpu.cos(src)
+            A[1:-1, 1:-1] = sc[:-2, 1:-1] + sc[2:, 1:-1] - 2.0 * sc[1:-1, 1:-1]
+            del src, sc
+            s = float(A.sum())
+        finally:
+            restore()
+        assert calls and calls[-1] is True, f"tiled path not taken: {calls}"
+        a = np.zeros((S, S))
+        sr = np.fromfunction(lambda x, y: (x + 2.0 * y) * 1e-3, (S, S))
+        sc2 = np.cos(sr)
+        a[1:-1, 1:-1] = sc2[:-2, 1:-1] + sc2[2:, 1:-1] - 2 * sc2[1:-1, 1:-1]
+        np.testing.assert_allclose(A.asarray(), a, rtol=1e-12, atol=1e-12)
+        assert abs(s - a.sum()) < 1e-9 * max(1.0, abs(a.sum()))
+
+    def test_staged_fp32_asymmetric(self, ra_gpu):
+        """fp32, asymmetric offsets (radius-2 one side), odd sizes."""
+        S0, S1 = 257, 131
+        calls, restore = self._spy()
+        try:
+            B = ra_gpu.zeros((S0, S1), dtype=np.float32)
+            ra_gpu.sync()
+            f = ra_gpu.fromfunction(lambda x, y: x * 0.5 + y * 0.25,
+                                    (S0, S1), dtype=np.float32)
+            g = ra_gpu.sqrt(f)
+            B[3:-1, :-2] = g[:-4, 1:-1] + 2.0 * g[4:, 2:] - g[2:-2, :-2]
+            out = B.asarray()
+        finally:
+            restore()
+        assert calls and calls[-1] is True, f"tiled path not taken: {calls}"
+        b = np.zeros((S0, S1), dtype=np.float32)
+        fn = np.fromfunction(lambda x, y: x * 0.5 + y * 0.25, (S0, S1),
+                             dtype=np.float32).astype(np.float32)
+        gn = np.sqrt(fn)
+        b[3:-1, :-2] = gn[:-4, 1:-1] + 2.0 * gn[4:, 2:] - gn[2:-2, :-2]
+        np.testing.assert_allclose(out, b, rtol=1e-5, atol=1e-5)
+
+    def test_staged_consumer_reads_other_arrays(self, ra_gpu):
+        """Consumer mixes LDS-staged reads with normal HBM operands."""
+        S = 200
+        calls, restore = self._spy()
+        try:
+            W = ra_gpu.fromfunction(lambda x, y: (x % 7) * 0.1 + y * 0.01,
+                                    (S, S), dtype=np.float64)
+            A = ra_gpu.zeros((S, S), dtype=np.float64)
+            ra_gpu.sync()
+            p = ra_gpu.fromfunction(lambda x, y: x * 1.5 + y, (S, S),
+                                    dtype=np.float64)
+            q = ra_gpu.sin(p * 0.01)
+            A[1:-1, 1:-1] = (q[:-2, 1:-1] + q[2:, 1:-1]) * W[1:-1, 1:-1]
+            out = A.asarray()
+        finally:
+            restore()
+        assert calls and calls[-1] is True, f"tiled path not taken: {calls}"
+        w = np.fromfunction(lambda x, y: (x % 7) * 0.1 + y * 0.01, (S, S))
+        a = np.zeros((S, S))
+        pn = np.fromfunction(lambda x, y: x * 1.5 + y, (S, S))
+        qn = np.sin(pn * 0.01)
+        a[1:-1, 1:-1] = (qn[:-2, 1:-1] + qn[2:, 1:-1]) * w[1:-1, 1:-1]
+        np.testing.assert_allclose(out, a, rtol=1e-12, atol=1e-12)
+
+    def test_stage_fusion_off_env_matches(self, ra_gpu):
+        """RAMBA_STAGE_FUSION=0 path (sequential) must agree; toggled via
+        the module flag (env is read at import)."""
+        from ramba_amd import common
+        S = 128
+        res = {}
+        for mode in (1, 0):
+            old = common.stage_fusion
+            common.stage_fusion = mode
+            try:
+                A = ra_gpu.zeros((S, S), dtype=np.float64)
+                ra_gpu.sync()
+                src = ra_gpu.fromfunction(
+                    lambda x, y: (x * S + y) * 1e-5, (S, S),
+                    dtype=np.float64)
+                ssin = ra_gpu.sin(src)
+                A[1:-1, 1:-1] = (ssin[:-2, 1:-1] + ssin[2:, 1:-1]
+                                 + ssin[1:-1, :-2] + ssin[1:-1, 2:]
+                                 - 4.0 * ssin[1:-1, 1:-1])
+                res[mode] = A.asarray()
+            finally:
+                common.stage_fusion = old
+        np.testing.assert_array_equal(res[0], res[1])
