@@ -126,12 +126,17 @@ def main():
         wl_desc = "1e9-elem fp64 sum(), 8 B/elem read + one 8 B allreduce"
     elif args.workload == "mixed":
         # configs[4]: iota fill -> sin -> 5-pt stencil -> sum, 8192^2 fp64.
-        # Three fused kernels per step: {iota+sin store}, {stencil},
-        # {read+reduce} + one 8 B allreduce.  32 B/elem if stages are not
-        # cross-fused.
+        # With cross-stage fusion (default, ramba_amd/staged.py) the sin
+        # producer is recomputed into LDS inside the stencil kernel and
+        # the dead intermediates never materialise: ONE tiled kernel
+        # (A store, 8 B) + the fused sum read (8 B) = 16 B/elem
+        # algorithmic — SURVEY §8d cfg5's "less if stencil fuses with sin
+        # producer".  RAMBA_STAGE_FUSION=0 restores the 3-kernel
+        # reference-order pipeline at 32 B/elem.
+        from ramba_amd.common import stage_fusion
         S = 8192
         N = S * S
-        alg_bytes_per_elem = 32
+        alg_bytes_per_elem = 16 if stage_fusion else 32
         A = ra.zeros((S, S), dtype=np.float64)
         ra.sync()
 
@@ -142,6 +147,11 @@ def main():
             A[1:-1, 1:-1] = (ssin[:-2, 1:-1] + ssin[2:, 1:-1]
                              + ssin[1:-1, :-2] + ssin[1:-1, 2:]
                              - 4.0 * ssin[1:-1, 1:-1])
+            # the intermediates are dead from here on: dropping the refs
+            # before the flush lets the engine demote them to registers/
+            # LDS (the reference's live_gids rule, ramba.py:8123, applied
+            # through the fused pair)
+            del src, ssin
             # force the reduction like the reference does (its sum is
             # eager: internal_reduction2b gathers immediately).  Leaving
             # the scalar pending would also let the LAST step's group
@@ -150,8 +160,10 @@ def main():
             return float(A.sum())
         metric = ("GElem/s, 8192^2 fp64 iota→sin→stencil→sum pipeline "
                   "(BASELINE configs[4])")
-        wl_desc = ("8192^2 fp64: fused iota+sin fill, 5-pt stencil, global "
-                   "sum + RCCL allreduce; 32 B/elem algorithmic")
+        wl_desc = ("8192^2 fp64: iota+sin fused INTO the 5-pt stencil "
+                   "kernel (LDS-staged producer), global sum + RCCL "
+                   f"allreduce; {16 if stage_fusion else 32} B/elem "
+                   "algorithmic")
     else:  # stencil
         S = args.stencil_n
         N = S * S

@@ -1406,6 +1406,18 @@ class _StageGen:
         return out
 
 
+def staged_tiled_key(desc):
+    """Structural cache key WITHOUT source generation (the per-step hot
+    path looks the kernel up by this before building anything)."""
+    return hashlib.sha256(repr((
+        [(st.target, st.expr) for st in desc["s1_stmts"]],
+        [(st.target, st.expr) for st in desc["s2_stmts"]],
+        desc["staged"], desc["s1_stores"], sorted(desc["readers"].items()),
+        desc["s2_ops"], sorted((n, str(d)) for n, d in
+                               desc["scalars"].items()),
+        desc["E0"], desc["E1"], TILE_H, TILE_W)).encode()).hexdigest()[:24]
+
+
 def generate_staged_tiled(desc):
     """desc (structural only; runtime values go through the fields):
       s1_stmts, s2_stmts : lists of ir.Assign (stage1 names p_-prefixed)
@@ -1422,13 +1434,7 @@ def generate_staged_tiled(desc):
     TH, TW = TILE_H, TILE_W
     FH, FW = TH + E0, TW + E1
 
-    key = hashlib.sha256(repr((
-        [(st.target, st.expr) for st in desc["s1_stmts"]],
-        [(st.target, st.expr) for st in desc["s2_stmts"]],
-        desc["staged"], desc["s1_stores"], sorted(desc["readers"].items()),
-        desc["s2_ops"], sorted((n, str(d)) for n, d in
-                               desc["scalars"].items()),
-        E0, E1, TH, TW)).encode()).hexdigest()[:24]
+    key = staged_tiled_key(desc)
     kname = f"tk_{key}"
 
     fields = [("n0", "q"), ("n1", "q"), ("gs0", "q"), ("gs1", "q"),

@@ -911,13 +911,15 @@ def _hb_tiled_kernel(self, desc):
     """Build (or fetch) the staged/tiled kernel; returns a handle token
     or None (fall back to sequential)."""
     from . import codegen as cg
-    try:
-        key, source, kname, fields = cg.generate_staged_tiled(desc)
-    except NotImplementedError:
-        return None
+    key = cg.staged_tiled_key(desc)
     ck = ("tiled", key)
     cached = self.kernels.get(ck)
     if cached is None:
+        try:
+            key2, source, kname, fields = cg.generate_staged_tiled(desc)
+        except NotImplementedError:
+            return None
+        assert key2 == key
         if int(os.environ.get("RAMBA_SHOW_CODE", "0")):
             print(f"=== tiled kernel {key} ===\n{source}\n", flush=True)
         h = ctypes.c_void_p()
