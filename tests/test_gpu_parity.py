@@ -526,16 +526,19 @@ class TestStagedFusion:
     actually ran."""
 
     def _spy(self):
-        import ramba_amd.staged as st
+        import ramba_amd.staged_exec as se
         calls = []
-        orig = st._execute_tiled
+        orig = se.run_recipe
 
         def wrap(*a, **k):
             r = orig(*a, **k)
             calls.append(r)
             return r
-        st._execute_tiled = wrap
-        return calls, lambda: setattr(st, "_execute_tiled", orig)
+        se.run_recipe = wrap
+        import ramba_amd.staged as st
+        st.staged_exec.run_recipe = wrap
+        return calls, lambda: (setattr(se, "run_recipe", orig),
+                               setattr(st.staged_exec, "run_recipe", orig))
 
     def test_mixed_pipeline_staged_live(self, ra_gpu):
         S = 512
