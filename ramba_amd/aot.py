@@ -67,6 +67,13 @@ class AotCompileBackend:
         self._cc(gk.source)
         if gk.finish_source:
             self._cc(gk.finish_source)
+        # also precompile the LDS load-tiled stencil variant when the
+        # plan qualifies (the HIP backend prefers it at runtime)
+        if not plan.reductions and len(plan.itershape) == 2                 and plan.itershape[0] * plan.itershape[1] >= (1 << 16):
+            fams = codegen.find_stencil_families(plan)
+            if fams:
+                _, src, _, _ = codegen.generate_load_tiled(plan, fams)
+                self._cc(src)
         return [np.asarray(ir.reduction_init(s.kind, s.dtype),
                            dtype=s.dtype)[()] for s in plan.reductions]
 
@@ -172,6 +179,14 @@ def seed(fuzz_seeds=None):
     Y = ra.zeros((128, 130), dtype=np.float32)
     Y[1:-1, 1:-1] = (X[:-2, 1:-1] + X[2:, 1:-1] + X[1:-1, :-2]
                      + X[1:-1, 2:] - 4.0 * X[1:-1, 1:-1])
+    ra.sync()
+    # configs[3] at load-tiled scale (>=2^16 elements) so the LDS
+    # stencil kernel lands in the cache (key is size-independent)
+    X2 = ra.fromfunction(lambda x, y: x + y, (512, 512), dtype=np.float32)
+    Y2 = ra.zeros((512, 512), dtype=np.float32)
+    ra.sync()
+    Y2[1:-1, 1:-1] = (X2[:-2, 1:-1] + X2[2:, 1:-1] + X2[1:-1, :-2]
+                      + X2[1:-1, 2:] - 4.0 * X2[1:-1, 1:-1])
     ra.sync()
     # configs[4] mixed fp64 (Z materialised first so the pair takes the
     # staged/tiled path, like the bench loop's steady state)

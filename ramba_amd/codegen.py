@@ -286,6 +286,18 @@ class LaneEmitter:
         g = self.gen
         if e.op in ("sin", "cos"):
             other = ir.Un("cos" if e.op == "sin" else "sin", e.a, e.dtype)
+            if ctype(e.dtype) == "double" and other not in g.wanted:
+                # SOLO fp64 sin/cos: rt_sincos (Cody-Waite fast path) and
+                # discard the other result — measurably cheaper than the
+                # generic ocml sin/cos for |x|<1e6 (the mixed pipeline's
+                # producer recompute is transcendental-bound)
+                x = self.emit(e.a)
+                self.n += 1
+                s, c = f"sc_s{self.n}{self.tag}", f"sc_c{self.n}{self.tag}"
+                self.lines.append(
+                    f"      double {s}, {c}; "
+                    f"rt_sincos((double)({x}), &{s}, &{c});")
+                return s if e.op == "sin" else c
             if other in g.wanted and other not in self.memo:
                 x = self.emit(e.a)
                 ct = ctype(e.dtype)
@@ -1568,3 +1580,168 @@ def pack_tk_args(fields, values):
         else:
             out += struct.pack("<d", float(v))
     return bytes(out)
+
+
+# ---------------------------------------------------------------------------
+# load-tiled stencil kernel: operands that read the
+# SAME array at several small shifts ("stencil family") are staged
+# through LDS — one cooperative tile+halo load replaces the per-element
+# shifted reads, killing the 1.3-1.6x L2/HBM over-fetch the PMC showed
+# at cache-resident sizes (profiles/r02_pmc_sten_*.csv).
+# ---------------------------------------------------------------------------
+
+LT_MAXD = 8
+
+
+def find_stencil_families(plan):
+    """Group container read operands by (buffer, strides); return
+    (families, members) where families = [(anchor_off, s0, dtype,
+    [(name, dr0, dr1)])] for groups of >=3 pure-shift readers with inner
+    stride 1, or None if the plan does not qualify."""
+    nd = len(plan.itershape)
+    if nd != 2 or plan.reductions:
+        return None
+    written = {st.target for st in plan.statements}
+    groups = {}
+    for op in plan.operands:
+        if op.kind != "container" or op.name in written:
+            continue
+        if len(op.strides) != 2 or op.strides[1] != 1 \
+                or op.strides[0] <= 0:
+            continue
+        groups.setdefault((id(op.bd), op.strides), []).append(op)
+    fams = []
+    for (key, ops) in groups.items():
+        if len(ops) < 3:
+            continue
+        s0 = ops[0].strides[0]
+        base = min(o.offset0 for o in ops)
+        mem = []
+        ok = True
+        for o in ops:
+            d = o.offset0 - base
+            dr0 = int(round(d / s0))
+            dr1 = d - dr0 * s0
+            if not (0 <= dr0 <= LT_MAXD and abs(dr1) <= LT_MAXD):
+                ok = False
+                break
+            mem.append((o.name, dr0, dr1))
+        if not ok:
+            continue
+        dr1_min = min(m[2] for m in mem)
+        anchor = base + dr1_min
+        mem = [(n, a, b - dr1_min) for (n, a, b) in mem]
+        fams.append((anchor, s0, str(ops[0].dtype), mem, ops[0].name))
+    return fams or None
+
+
+def generate_load_tiled(plan, fams):
+    """Tiled kernel: cooperative LDS load of each family's tile+halo,
+    then the plan's statements with family reads served from LDS."""
+    TH, TW = TILE_H, TILE_W
+    nd = 2
+    fam_members = {}
+    for fi, (anchor, s0, dt, mem, rep) in enumerate(fams):
+        for (n, dr0, dr1) in mem:
+            fam_members[n] = (fi, dr0, dr1)
+    fam_ext = []
+    for (anchor, s0, dt, mem, rep) in fams:
+        E0 = max(m[1] for m in mem)
+        E1 = max(m[2] for m in mem)
+        fam_ext.append((E0, E1))
+
+    other_ops = [op for op in plan.operands if op.name not in fam_members]
+    written = {st.target for st in plan.statements}
+
+    key = hashlib.sha256(repr((
+        "loadtiled", [(st.target, st.expr) for st in plan.statements],
+        [(fi, dt, sorted(mem)) for fi, (a, s, dt, mem, rep)
+         in enumerate(fams)], fam_ext,
+        [(o.name, str(o.dtype), o.name in written, o.strides[1] == 0)
+         for o in other_ops],
+        sorted((n, str(dt)) for n, (v, dt) in plan.scalars.items()),
+        sorted((n, str(dt)) for n, dt in plan.dead_vars.items()),
+        TH, TW)).encode()).hexdigest()[:24]
+    kname = f"lt_{key}"
+
+    fields = [("n0", "q"), ("n1", "q"), ("gs0", "q"), ("gs1", "q")]
+    L = [PREAMBLE]
+    L.append("struct LtArgs {")
+    L.append("  i64 n0, n1, gs0, gs1;")
+    for fi, (anchor, s0, dt, mem, rep) in enumerate(fams):
+        L.append(f"  {ctype(dt)}* __restrict__ fam{fi}_ptr; "
+                 f"i64 fam{fi}_off, fam{fi}_s0;")
+        fields += [(f"fam{fi}_ptr", "Q"), (f"fam{fi}_off", "q"),
+                   (f"fam{fi}_s0", "q")]
+    for o in other_ops:
+        L.append(f"  {ctype(o.dtype)}* __restrict__ {o.name}_ptr; "
+                 f"i64 {o.name}_off, {o.name}_s0, {o.name}_s1;")
+        fields += [(f"{o.name}_ptr", "Q"), (f"{o.name}_off", "q"),
+                   (f"{o.name}_s0", "q"), (f"{o.name}_s1", "q")]
+    for n in sorted(plan.scalars):
+        dt = plan.scalars[n][1]
+        if np.dtype(dt).kind == "f":
+            fields.append((n, "d"))
+            L.append(f"  double {n};")
+        else:
+            fields.append((n, "q"))
+            L.append(f"  i64 {n};")
+    L.append("};")
+
+    L.append(f'extern "C" __global__ void __launch_bounds__(256) '
+             f"{kname}(LtArgs a) {{")
+    for fi, (E0, E1) in enumerate(fam_ext):
+        dt = fams[fi][2]
+        L.append(f"  __shared__ {ctype(dt)} lds_f{fi}"
+                 f"[{TH + E0}][{TW + E1} + 1];")
+    L.append(f"  const i64 tiles1 = (a.n1 + {TW} - 1) / {TW};")
+    L.append(f"  const i64 tiles0 = (a.n0 + {TH} - 1) / {TH};")
+    L.append("  for (i64 tile = blockIdx.x; tile < tiles0 * tiles1; "
+             "tile += gridDim.x) {")
+    L.append(f"    const i64 k0o = (tile / tiles1) * {TH};")
+    L.append(f"    const i64 k1o = (tile % tiles1) * {TW};")
+    for fi, (E0, E1) in enumerate(fam_ext):
+        FH, FW = TH + E0, TW + E1
+        L.append(f"    {{ i64 fh_ = a.n0 - k0o + {E0}; "
+                 f"const int fh = (int)(fh_ < {FH} ? fh_ : {FH});")
+        L.append(f"      i64 fw_ = a.n1 - k1o + {E1}; "
+                 f"const int fw = (int)(fw_ < {FW} ? fw_ : {FW});")
+        L.append(f"      for (int fi_ = threadIdx.x; fi_ < fh * fw; "
+                 f"fi_ += 256) {{")
+        L.append(f"        const int f0 = fi_ / fw, f1 = fi_ % fw;")
+        L.append(f"        lds_f{fi}[f0][f1] = a.fam{fi}_ptr[a.fam{fi}_off"
+                 f" + (k0o + f0) * a.fam{fi}_s0 + (k1o + f1)];")
+        L.append("      } }")
+    L.append("    __syncthreads();")
+
+    def resolve0(base, em):
+        if base in fam_members:
+            fi, dr0, dr1 = fam_members[base]
+            return f"lds_f{fi}[q0{em.tag} + {dr0}][q1{em.tag} + {dr1}]"
+        for o in other_ops:
+            if o.name == base:
+                return (f"a.{base}_ptr[a.{base}_off + k0{em.tag} * "
+                        f"a.{base}_s0 + k1{em.tag} * a.{base}_s1]")
+        return None
+
+    gen2 = _StageGen(plan.statements, resolve0, plan.dead_vars)
+    L.append(f"    for (int ti = threadIdx.x; ti < {TH} * {TW}; ti += 256)"
+             " {")
+    L.append(f"      const int q0_t = ti / {TW}, q1_t = ti % {TW};")
+    L.append("      const i64 k0_t = k0o + q0_t;")
+    L.append("      const i64 k1_t = k1o + q1_t;")
+    L.append("      if (k0_t < a.n0 && k1_t < a.n1) {")
+    em = LaneEmitter(gen2, "_t", ["(a.gs0 + k0_t)", "(a.gs1 + k1_t)"])
+    finals = gen2.emit_into(em)
+    L.extend(em.lines)
+    for o in other_ops:
+        if o.name in written:
+            L.append(f"      a.{o.name}_ptr[a.{o.name}_off + k0_t * "
+                     f"a.{o.name}_s0 + k1_t * a.{o.name}_s1] = "
+                     f"{finals[o.name]};")
+    L.append("      }")
+    L.append("    }")
+    L.append("    __syncthreads();")
+    L.append("  }")
+    L.append("}")
+    return key, "\n".join(L), kname, fields

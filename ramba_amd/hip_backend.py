@@ -380,9 +380,14 @@ class HipBackend:
         if recipe is not None and recipe.backend_kernel is not None:
             gk = recipe.backend_kernel
         else:
-            gk = self._get_kernel(plan)
+            gk = self._maybe_load_tiled(plan) or self._get_kernel(plan)
             if recipe is not None:
                 recipe.backend_kernel = gk
+        if isinstance(gk, tuple) and gk[0] == "lt":
+            r = self._launch_load_tiled(gk, plan)
+            if ntiming:
+                add_time("hb_submit", _time.perf_counter() - _t0)
+            return r
         nd = gk.nd
         shape = plan.itershape
         V = gk.vec
@@ -954,3 +959,86 @@ def _hb_tiled_launch(self, handle, vals, ntiles):
 HipBackend.container_addr = _hb_container_addr
 HipBackend.tiled_kernel = _hb_tiled_kernel
 HipBackend.tiled_launch = _hb_tiled_launch
+
+
+# -- load-tiled stencil dispatch -------------------------
+
+
+def _hb_maybe_load_tiled(self, plan):
+    """If the plan is a qualifying 2-D stencil (>=3 same-array shifted
+    readers), build the LDS load-tiled kernel instead of the vectorized
+    elementwise one.  Returns ("lt", handle, fields, fam_meta, others)
+    or None."""
+    if os.environ.get("RAMBA_STENCIL_LDS", "1") == "0":
+        return None
+    if plan.reductions or len(plan.itershape) != 2:
+        return None
+    if plan.itershape[0] * plan.itershape[1] < (1 << 16):
+        return None         # rim slabs / tiny boxes: vectorized path wins
+    from . import codegen as cg
+    fams = cg.find_stencil_families(plan)
+    if not fams:
+        return None
+    key, source, kname, fields = cg.generate_load_tiled(plan, fams)
+    ck = ("lt", key)
+    cached = self.kernels.get(ck)
+    if cached is None:
+        if int(os.environ.get("RAMBA_SHOW_CODE", "0")):
+            print(f"=== load-tiled kernel {key} ===\n{source}\n",
+                  flush=True)
+        h = ctypes.c_void_p()
+        self._check(self.lib.rt_kernel_get(
+            key.encode(), source.encode(), kname.encode(),
+            ctypes.byref(h)), "rt_kernel_get(load_tiled)")
+        cached = (h.value, fields)
+        self.kernels[ck] = cached
+    fam_meta = [(rep, anchor, s0) for (anchor, s0, dt, mem, rep) in fams]
+    member_names = {n for (a, s, dt, mem, rep) in fams for (n, _, _) in mem}
+    others = [op.name for op in plan.operands
+              if op.name not in member_names]
+    return ("lt", cached[0], cached[1], fam_meta, others)
+
+
+def _hb_launch_load_tiled(self, gk, plan):
+    from . import codegen as cg
+    _, handle, fields, fam_meta, others = gk
+    opmap = {o.name: o for o in plan.operands}
+    vals = {"n0": plan.itershape[0], "n1": plan.itershape[1],
+            "gs0": plan.global_start[0], "gs1": plan.global_start[1]}
+    for fi, (rep, anchor, s0) in enumerate(fam_meta):
+        op = opmap[rep]
+        vals[f"fam{fi}_ptr"] = self._cont(op.bd).data_ptr() \
+            if op.kind == "container" else self.temps[op.name].data_ptr()
+        vals[f"fam{fi}_off"] = anchor
+        vals[f"fam{fi}_s0"] = s0
+    for name in others:
+        op = opmap[name]
+        vals[f"{name}_ptr"] = self._cont(op.bd).data_ptr() \
+            if op.kind == "container" else self.temps[name].data_ptr()
+        vals[f"{name}_off"] = op.offset0
+        vals[f"{name}_s0"] = op.strides[0]
+        vals[f"{name}_s1"] = op.strides[1]
+    for n, (v, dt) in plan.scalars.items():
+        vals[n] = v
+    args = cg.pack_tk_args(fields, vals)
+    ntiles = ((plan.itershape[0] + cg.TILE_H - 1) // cg.TILE_H) \
+        * ((plan.itershape[1] + cg.TILE_W - 1) // cg.TILE_W)
+    gx = max(1, min(int(os.environ.get("RAMBA_GRID_CAP", "16384")),
+                    ntiles))
+    if self.time_kernels:
+        ev0 = self.torch.cuda.Event(enable_timing=True)
+        ev1 = self.torch.cuda.Event(enable_timing=True)
+        ev0.record()
+    self._check(self.lib.rt_launch(
+        ctypes.c_void_p(handle), gx, 1, 1, 256, self._stream(), args,
+        len(args)), "rt_launch(load_tiled)")
+    if self.time_kernels:
+        ev1.record()
+        ev1.synchronize()
+        self.kernel_times_ms.append(ev0.elapsed_time(ev1))
+        self.kernel_keys.append("lt")
+    return []
+
+
+HipBackend._maybe_load_tiled = _hb_maybe_load_tiled
+HipBackend._launch_load_tiled = _hb_launch_load_tiled
