@@ -195,9 +195,12 @@ def main():
     # warmup steps actually done.
     warm_requested = args.warmup
     warm_done = 0
+    # profiling runs shrink the floor (RAMBA_WARMUP_FLOOR=0) so PMC
+    # databases stay small; the default 6 s stays for timing runs
+    wfloor = float(os.environ.get("RAMBA_WARMUP_FLOOR", "6.0"))
     tw = time.perf_counter()
     if world == 1:
-        while warm_done < args.warmup or time.perf_counter() - tw < 6.0:
+        while warm_done < args.warmup or time.perf_counter() - tw < wfloor:
             keep = step()
             warm_done += 1
             if warm_done > args.warmup + 100_000:
@@ -213,7 +216,7 @@ def main():
             warm_done += 1
         while True:
             flag = torch.tensor(
-                [1.0 if time.perf_counter() - tw >= 6.0 else 0.0],
+                [1.0 if time.perf_counter() - tw >= wfloor else 0.0],
                 dtype=torch.float64, device="cuda")
             dist.all_reduce(flag, op=dist.ReduceOp.MIN)
             if float(flag.cpu()[0]) >= 1.0 or warm_done > 100_000:
