@@ -72,7 +72,7 @@ class AotCompileBackend:
         if not plan.reductions and len(plan.itershape) == 2                 and plan.itershape[0] * plan.itershape[1] >= (1 << 16):
             fams = codegen.find_stencil_families(plan)
             if fams:
-                _, src, _, _ = codegen.generate_load_tiled(plan, fams)
+                _, src, _, _, _ = codegen.generate_load_tiled(plan, fams)
                 self._cc(src)
         return [np.asarray(ir.reduction_init(s.kind, s.dtype),
                            dtype=s.dtype)[()] for s in plan.reductions]

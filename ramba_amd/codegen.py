@@ -1635,33 +1635,48 @@ def find_stencil_families(plan):
     return fams or None
 
 
+LT_TH = 32            # tile rows
+LT_TXCH = 64          # column chunks per tile (one per lane of a wave?)
+
+
 def generate_load_tiled(plan, fams):
-    """Tiled kernel: cooperative LDS load of each family's tile+halo,
-    then the plan's statements with family reads served from LDS."""
-    TH, TW = TILE_H, TILE_W
+    """v2: 16 B/lane vectorised tiled stencil.  One workgroup = 256
+    threads as (64 col-chunks x 4 row-groups); each thread loads/stores
+    VEC consecutive elements (16 B) per access, so the memory
+    instruction count matches the CDNA4 full-rate idiom, and the tile is
+    32 x 64*VEC outputs so the halo re-read is ~(34/32)x(~1.01) ~ 1.07x
+    of the input instead of v1's 1.16x."""
     nd = 2
+    # one vector width for the whole kernel: 16 B / largest itemsize
+    items = [np.dtype(dt).itemsize for (a, s, dt, mem, rep) in fams]
+    for op in plan.operands:
+        if any(op.name == n for (a, s, dt, mem, rep) in fams
+               for (n, _, _) in mem):
+            continue
+        items.append(np.dtype(op.dtype).itemsize)
+    V = max(1, 16 // max(items))
+    TH, NCH = LT_TH, LT_TXCH
+    CW = NCH * V                     # tile output columns
+
     fam_members = {}
+    fam_ext = []
     for fi, (anchor, s0, dt, mem, rep) in enumerate(fams):
         for (n, dr0, dr1) in mem:
             fam_members[n] = (fi, dr0, dr1)
-    fam_ext = []
-    for (anchor, s0, dt, mem, rep) in fams:
-        E0 = max(m[1] for m in mem)
-        E1 = max(m[2] for m in mem)
-        fam_ext.append((E0, E1))
+        fam_ext.append((max(m[1] for m in mem), max(m[2] for m in mem)))
 
     other_ops = [op for op in plan.operands if op.name not in fam_members]
     written = {st.target for st in plan.statements}
 
     key = hashlib.sha256(repr((
-        "loadtiled", [(st.target, st.expr) for st in plan.statements],
+        "loadtiled2", [(st.target, st.expr) for st in plan.statements],
         [(fi, dt, sorted(mem)) for fi, (a, s, dt, mem, rep)
          in enumerate(fams)], fam_ext,
         [(o.name, str(o.dtype), o.name in written, o.strides[1] == 0)
          for o in other_ops],
         sorted((n, str(dt)) for n, (v, dt) in plan.scalars.items()),
         sorted((n, str(dt)) for n, dt in plan.dead_vars.items()),
-        TH, TW)).encode()).hexdigest()[:24]
+        TH, NCH, V)).encode()).hexdigest()[:24]
     kname = f"lt_{key}"
 
     fields = [("n0", "q"), ("n1", "q"), ("gs0", "q"), ("gs1", "q")]
@@ -1688,60 +1703,133 @@ def generate_load_tiled(plan, fams):
             L.append(f"  i64 {n};")
     L.append("};")
 
+    # unaligned-tolerant vector structs (global loads/stores of V elems)
     L.append(f'extern "C" __global__ void __launch_bounds__(256) '
              f"{kname}(LtArgs a) {{")
     for fi, (E0, E1) in enumerate(fam_ext):
         dt = fams[fi][2]
+        # pitch: multiple of V so vector LDS rows stay aligned
+        FW = CW + E1
+        pitch = -(-FW // V) * V + V
         L.append(f"  __shared__ {ctype(dt)} lds_f{fi}"
-                 f"[{TH + E0}][{TW + E1} + 1];")
-    L.append(f"  const i64 tiles1 = (a.n1 + {TW} - 1) / {TW};")
+                 f"[{TH + E0}][{pitch}];")
+    L.append(f"  const i64 tiles1 = (a.n1 + {CW} - 1) / {CW};")
     L.append(f"  const i64 tiles0 = (a.n0 + {TH} - 1) / {TH};")
+    L.append("  const int tx = threadIdx.x & 63;")
+    L.append("  const int ty = threadIdx.x >> 6;")
     L.append("  for (i64 tile = blockIdx.x; tile < tiles0 * tiles1; "
              "tile += gridDim.x) {")
     L.append(f"    const i64 k0o = (tile / tiles1) * {TH};")
-    L.append(f"    const i64 k1o = (tile % tiles1) * {TW};")
+    L.append(f"    const i64 k1o = (tile % tiles1) * {CW};")
+    # ---- cooperative fill, vectorised -----------------------------------
     for fi, (E0, E1) in enumerate(fam_ext):
-        FH, FW = TH + E0, TW + E1
+        dt = fams[fi][2]
+        ct = ctype(dt)
+        FH, FW = TH + E0, CW + E1
+        vct = {("double", 2): "d2_t", ("float", 4): "f4_t",
+               ("long long", 2): "l2_t", ("i64", 2): "l2_t",
+               ("int", 4): "i4_t", ("short", 4): "s4_t",  # wait: 16B/maxitem
+               }.get((ct, V))
         L.append(f"    {{ i64 fh_ = a.n0 - k0o + {E0}; "
                  f"const int fh = (int)(fh_ < {FH} ? fh_ : {FH});")
         L.append(f"      i64 fw_ = a.n1 - k1o + {E1}; "
                  f"const int fw = (int)(fw_ < {FW} ? fw_ : {FW});")
-        L.append(f"      for (int fi_ = threadIdx.x; fi_ < fh * fw; "
-                 f"fi_ += 256) {{")
-        L.append(f"        const int f0 = fi_ / fw, f1 = fi_ % fw;")
-        L.append(f"        lds_f{fi}[f0][f1] = a.fam{fi}_ptr[a.fam{fi}_off"
-                 f" + (k0o + f0) * a.fam{fi}_s0 + (k1o + f1)];")
+        L.append(f"      const int nch = (fw + {V} - 1) / {V};")
+        L.append(f"      for (int r = ty; r < fh; r += 4) {{")
+        L.append(f"        const i64 rb = a.fam{fi}_off + (k0o + r) * "
+                 f"a.fam{fi}_s0 + k1o;")
+        L.append(f"        for (int c = tx; c < nch; c += 64) {{")
+        L.append(f"          const int col = c * {V};")
+        if vct:
+            L.append(f"          if (col + {V} <= fw) {{")
+            L.append(f"            {vct} v; __builtin_memcpy(&v, "
+                     f"&a.fam{fi}_ptr[rb + col], sizeof(v));")
+            L.append(f"            __builtin_memcpy(&lds_f{fi}[r][col], "
+                     f"&v, sizeof(v));")
+            L.append("          } else {")
+            L.append(f"            for (int j = 0; j < {V}; ++j) "
+                     f"if (col + j < fw) lds_f{fi}[r][col + j] = "
+                     f"a.fam{fi}_ptr[rb + col + j];")
+            L.append("          }")
+        else:
+            L.append(f"          for (int j = 0; j < {V}; ++j) "
+                     f"if (col + j < fw) lds_f{fi}[r][col + j] = "
+                     f"a.fam{fi}_ptr[rb + col + j];")
+        L.append("        }")
         L.append("      } }")
     L.append("    __syncthreads();")
 
-    def resolve0(base, em):
-        if base in fam_members:
-            fi, dr0, dr1 = fam_members[base]
-            return f"lds_f{fi}[q0{em.tag} + {dr0}][q1{em.tag} + {dr1}]"
-        for o in other_ops:
-            if o.name == base:
-                return (f"a.{base}_ptr[a.{base}_off + k0{em.tag} * "
-                        f"a.{base}_s0 + k1{em.tag} * a.{base}_s1]")
-        return None
+    # ---- stage 2: V outputs per thread ----------------------------------
+    def mk_resolver(jexpr):
+        def resolve0(base, em):
+            if base in fam_members:
+                fi, dr0, dr1 = fam_members[base]
+                return (f"lds_f{fi}[r2 + {dr0}]"
+                        f"[tx * {V} + {jexpr} + {dr1}]")
+            for o in other_ops:
+                if o.name == base:
+                    return (f"a.{base}_ptr[a.{base}_off + k0_t * "
+                            f"a.{base}_s0 + (k1_t + {jexpr}) * "
+                            f"a.{base}_s1]")
+            return None
+        return resolve0
 
-    gen2 = _StageGen(plan.statements, resolve0, plan.dead_vars)
-    L.append(f"    for (int ti = threadIdx.x; ti < {TH} * {TW}; ti += 256)"
-             " {")
-    L.append(f"      const int q0_t = ti / {TW}, q1_t = ti % {TW};")
-    L.append("      const i64 k0_t = k0o + q0_t;")
-    L.append("      const i64 k1_t = k1o + q1_t;")
-    L.append("      if (k0_t < a.n0 && k1_t < a.n1) {")
-    em = LaneEmitter(gen2, "_t", ["(a.gs0 + k0_t)", "(a.gs1 + k1_t)"])
-    finals = gen2.emit_into(em)
-    L.extend(em.lines)
+    L.append(f"    for (int r2 = ty; r2 < {TH}; r2 += 4) {{")
+    L.append("      const i64 k0_t = k0o + r2;")
+    L.append("      if (k0_t >= a.n0) break;")
+    L.append(f"      const i64 k1_t = k1o + tx * {V};")
+    L.append(f"      const bool full = (k1_t + {V} <= a.n1);")
+    # full path: unrolled j, vector stores
+    L.append("      if (full) {")
+    store_vecs = {}
+    for j in range(V):
+        genj = _StageGen(plan.statements, mk_resolver(str(j)),
+                         plan.dead_vars)
+        em = LaneEmitter(genj, f"_j{j}",
+                         ["(a.gs0 + k0_t)", f"(a.gs1 + k1_t + {j})"])
+        finals = genj.emit_into(em)
+        L.extend(em.lines)
+        for o in other_ops:
+            if o.name in written:
+                store_vecs.setdefault(o.name, []).append(finals[o.name])
+    for o in other_ops:
+        if o.name not in written:
+            continue
+        ct = ctype(o.dtype)
+        vct = {("double", 2): "d2_t", ("float", 4): "f4_t",
+               ("long long", 2): "l2_t", ("int", 4): "i4_t",
+               ("short", 4): "s4_t", ("signed char", 4): "c4_t",
+               ("unsigned char", 4): "b4_t"}.get((ct, V))
+        vals = store_vecs[o.name]
+        if vct and o.strides[1] == 1:
+            L.append(f"      {{ {vct} sv; " + " ".join(
+                f"sv[{j}] = {vals[j]};" for j in range(V)) + "")
+            L.append(f"        __builtin_memcpy(&a.{o.name}_ptr["
+                     f"a.{o.name}_off + k0_t * a.{o.name}_s0 + k1_t], "
+                     f"&sv, sizeof(sv)); }}")
+        else:
+            for j in range(V):
+                L.append(f"      a.{o.name}_ptr[a.{o.name}_off + k0_t * "
+                         f"a.{o.name}_s0 + (k1_t + {j}) * a.{o.name}_s1]"
+                         f" = {vals[j]};")
+    L.append("      } else {")
+    # edge path: scalar with per-element guard
+    L.append(f"        for (int j = 0; j < {V}; ++j) {{")
+    L.append("          if (k1_t + j >= a.n1) break;")
+    gene = _StageGen(plan.statements, mk_resolver("j"), plan.dead_vars)
+    eme = LaneEmitter(gene, "_je",
+                      ["(a.gs0 + k0_t)", "(a.gs1 + k1_t + j)"])
+    finals_e = gene.emit_into(eme)
+    L.extend(eme.lines)
     for o in other_ops:
         if o.name in written:
-            L.append(f"      a.{o.name}_ptr[a.{o.name}_off + k0_t * "
-                     f"a.{o.name}_s0 + k1_t * a.{o.name}_s1] = "
-                     f"{finals[o.name]};")
+            L.append(f"          a.{o.name}_ptr[a.{o.name}_off + k0_t * "
+                     f"a.{o.name}_s0 + (k1_t + j) * a.{o.name}_s1] = "
+                     f"{finals_e[o.name]};")
+    L.append("        }")
     L.append("      }")
     L.append("    }")
     L.append("    __syncthreads();")
     L.append("  }")
     L.append("}")
-    return key, "\n".join(L), kname, fields
+    return key, "\n".join(L), kname, fields, (TH, CW)

@@ -979,7 +979,7 @@ def _hb_maybe_load_tiled(self, plan):
     fams = cg.find_stencil_families(plan)
     if not fams:
         return None
-    key, source, kname, fields = cg.generate_load_tiled(plan, fams)
+    key, source, kname, fields, tile = cg.generate_load_tiled(plan, fams)
     ck = ("lt", key)
     cached = self.kernels.get(ck)
     if cached is None:
@@ -996,12 +996,12 @@ def _hb_maybe_load_tiled(self, plan):
     member_names = {n for (a, s, dt, mem, rep) in fams for (n, _, _) in mem}
     others = [op.name for op in plan.operands
               if op.name not in member_names]
-    return ("lt", cached[0], cached[1], fam_meta, others)
+    return ("lt", cached[0], cached[1], fam_meta, others, tile)
 
 
 def _hb_launch_load_tiled(self, gk, plan):
     from . import codegen as cg
-    _, handle, fields, fam_meta, others = gk
+    _, handle, fields, fam_meta, others, tile = gk
     opmap = {o.name: o for o in plan.operands}
     vals = {"n0": plan.itershape[0], "n1": plan.itershape[1],
             "gs0": plan.global_start[0], "gs1": plan.global_start[1]}
@@ -1021,8 +1021,9 @@ def _hb_launch_load_tiled(self, gk, plan):
     for n, (v, dt) in plan.scalars.items():
         vals[n] = v
     args = cg.pack_tk_args(fields, vals)
-    ntiles = ((plan.itershape[0] + cg.TILE_H - 1) // cg.TILE_H) \
-        * ((plan.itershape[1] + cg.TILE_W - 1) // cg.TILE_W)
+    th, cw = tile
+    ntiles = ((plan.itershape[0] + th - 1) // th) \
+        * ((plan.itershape[1] + cw - 1) // cw)
     gx = max(1, min(int(os.environ.get("RAMBA_GRID_CAP", "16384")),
                     ntiles))
     if self.time_kernels:
