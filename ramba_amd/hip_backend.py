@@ -969,7 +969,11 @@ def _hb_maybe_load_tiled(self, plan):
     readers), build the LDS load-tiled kernel instead of the vectorized
     elementwise one.  Returns ("lt", handle, fields, fam_meta, others)
     or None."""
-    if os.environ.get("RAMBA_STENCIL_LDS", "1") == "0":
+    # measured on MI355X (profiles/README r02): the LDS kernel cuts HBM
+    # FETCH 1.56x -> 1.17x of algorithmic but loses wall time to the
+    # vectorized cross-row-sharing kernel (30000^2: 2.28 vs 1.60 ms), so
+    # the fast path stays default and LDS mode is opt-in evidence
+    if os.environ.get("RAMBA_STENCIL_LDS", "0") == "0":
         return None
     if plan.reductions or len(plan.itershape) != 2:
         return None
