@@ -264,6 +264,7 @@ def main():
         for i in range(eb.shape[1]):
             local_elems *= int(eb[1, i] - eb[0, i] + 1)
     kms = None
+    kbreak = None
     kt = backend.kernel_times_ms
     if kt:
         # a step may launch several kernels (the mixed pipeline): sum the
@@ -271,6 +272,9 @@ def main():
         per_step = len(kt) // 3 if len(kt) % 3 == 0 else len(kt)
         sums = [sum(kt[i:i + per_step]) for i in range(0, len(kt), per_step)]
         kms = min(sums)
+        keys = backend.kernel_keys[:per_step]
+        kbreak = [{"kernel": k, "ms": round(t, 4)}
+                  for k, t in zip(keys, kt[:per_step])]
     roofline = None
     if kms:
         achieved = alg_bytes_per_elem * local_elems / (kms / 1e3)
@@ -310,7 +314,15 @@ def main():
             "traffic": float(traffic) if traffic else None,
             "traffic_source": traffic_src if traffic else None,
             "kernel_ms": kms,
+            "kernels": kbreak,
         }
+        if args.workload == "mixed" and alg_bytes_per_elem == 16:
+            # transparency: cross-stage fusion halves the algorithmic
+            # bytes ("less if stencil fuses with sin
+            # producer"), so `frac` above is the TRUE-traffic rate; the
+            # round-1-comparable figure (the unfused 32 B/elem workload
+            # unit over the fused execution time) is reported alongside
+            roofline["equiv_unfused_frac"] =                 32 * local_elems / (kms / 1e3) / HBM_PEAK_BYTES
 
     # ---- cpu_baseline leg (rank 0, N=1 only): oracle/fused_cpu.c ----------
     cpu_baseline = None

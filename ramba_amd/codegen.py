@@ -22,6 +22,7 @@ scalars) are kernel arguments, so iterating workloads reuse one kernel.
 """
 
 import hashlib
+import os
 import struct
 
 import numpy as np
@@ -1655,7 +1656,8 @@ def generate_load_tiled(plan, fams):
             continue
         items.append(np.dtype(op.dtype).itemsize)
     V = max(1, 16 // max(items))
-    TH, NCH = LT_TH, LT_TXCH
+    TH = int(os.environ.get("RAMBA_LT_TH", str(LT_TH)))
+    NCH = int(os.environ.get("RAMBA_LT_NCH", str(LT_TXCH)))
     CW = NCH * V                     # tile output columns
 
     fam_members = {}

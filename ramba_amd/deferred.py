@@ -122,9 +122,16 @@ class FusedGroup:
 
     # -- naming -------------------------------------------------------------
 
+    _NAME_TABLES = {}
+
     def fresh(self, prefix):
         self._counter += 1
-        return f"{prefix}{self._counter:04d}"
+        tab = FusedGroup._NAME_TABLES.get(prefix)
+        if tab is None:
+            tab = FusedGroup._NAME_TABLES[prefix] = [
+                f"{prefix}{i:04d}" for i in range(256)]
+        c = self._counter
+        return tab[c] if c < 256 else f"{prefix}{c:04d}"
 
     def exec_boxes(self):
         if self._exec_boxes is None:
