@@ -1636,7 +1636,7 @@ def find_stencil_families(plan):
     return fams or None
 
 
-LT_TH = 32            # tile rows
+LT_TH = 16            # tile rows (TH=32's 36 KB LDS halves occupancy: 2.24 vs 1.58 ms at 30000^2 — r02 sweep)
 LT_TXCH = 64          # column chunks per tile (one per lane of a wave?)
 
 
