@@ -969,11 +969,13 @@ def _hb_maybe_load_tiled(self, plan):
     readers), build the LDS load-tiled kernel instead of the vectorized
     elementwise one.  Returns ("lt", handle, fields, fam_meta, others)
     or None."""
-    # measured on MI355X (profiles/README r02): the LDS kernel cuts HBM
-    # FETCH 1.56x -> 1.17x of algorithmic but loses wall time to the
-    # vectorized cross-row-sharing kernel (30000^2: 2.28 vs 1.60 ms), so
-    # the fast path stays default and LDS mode is opt-in evidence
-    if os.environ.get("RAMBA_STENCIL_LDS", "0") == "0":
+    # measured on MI355X (profiles/README r02): at 16x256 tiles the LDS
+    # kernel cuts HBM FETCH 1.56x -> 1.21-1.27x of algorithmic AND beats
+    # the vectorized cross-row-sharing kernel at PRK scale (30000^2:
+    # 1.562 vs 1.599 ms; 4096^2 par).  Default ON; =0 selects the
+    # vectorized kernel for A/B.  (The first 32x256 tiling lost to
+    # occupancy: 36 KB LDS -> 2.24 ms.)
+    if os.environ.get("RAMBA_STENCIL_LDS", "1") == "0":
         return None
     if plan.reductions or len(plan.itershape) != 2:
         return None
