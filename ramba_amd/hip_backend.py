@@ -392,7 +392,7 @@ class HipBackend:
         shape = plan.itershape
         V = gk.vec
         nx = shape[nd - 1]
-        cap = int(os.environ.get("RAMBA_GRID_CAP", "16384"))
+        cap = int(os.environ.get("RAMBA_GRID_CAP", "32768"))
         gx = max(1, min(cap, (nx + 256 * V - 1) // (256 * V)))
         yb = getattr(gk, "yblock", 1)
         gy = max(1, min(8192, (shape[nd - 2] + yb - 1) // yb)) \
@@ -940,7 +940,7 @@ def _hb_tiled_launch(self, handle, vals, ntiles):
     from . import codegen as cg
     h, fields = handle
     args = cg.pack_tk_args(fields, vals)
-    gx = max(1, min(int(os.environ.get("RAMBA_GRID_CAP", "16384")),
+    gx = max(1, min(int(os.environ.get("RAMBA_GRID_CAP", "32768")),
                     ntiles))
     if self.time_kernels:
         ev0 = self.torch.cuda.Event(enable_timing=True)
@@ -1030,7 +1030,7 @@ def _hb_launch_load_tiled(self, gk, plan):
     th, cw = tile
     ntiles = ((plan.itershape[0] + th - 1) // th) \
         * ((plan.itershape[1] + cw - 1) // cw)
-    gx = max(1, min(int(os.environ.get("RAMBA_GRID_CAP", "16384")),
+    gx = max(1, min(int(os.environ.get("RAMBA_GRID_CAP", "32768")),
                     ntiles))
     if self.time_kernels:
         ev0 = self.torch.cuda.Event(enable_timing=True)
