@@ -1641,14 +1641,13 @@ LT_TXCH = 64          # column chunks per tile (one per lane of a wave?)
 
 
 def generate_load_tiled(plan, fams):
-    """v2: 16 B/lane vectorised tiled stencil.  One workgroup = 256
-    threads as (64 col-chunks x 4 row-groups); each thread loads/stores
-    VEC consecutive elements (16 B) per access, so the memory
-    instruction count matches the CDNA4 full-rate idiom, and the tile is
-    32 x 64*VEC outputs so the halo re-read is ~(34/32)x(~1.01) ~ 1.07x
-    of the input instead of v1's 1.16x."""
+    """v3: 16 B/lane vectorised, LDS-ring rolling stencil.  Workgroups
+    own (column-strip, row-segment) pairs of SEG consecutive tiles and
+    march down the strip carrying the E0 overlap rows in an LDS ring
+    (slot = input_row % (TH+E0)) — the row-halo re-read amortises from
+    (TH+E0)/TH per tile to ~1+E0/(SEG*TH) per segment.  16x256 tiles,
+    18 KB LDS (f32), 8 blocks/CU."""
     nd = 2
-    # one vector width for the whole kernel: 16 B / largest itemsize
     items = [np.dtype(dt).itemsize for (a, s, dt, mem, rep) in fams]
     for op in plan.operands:
         if any(op.name == n for (a, s, dt, mem, rep) in fams
@@ -1658,6 +1657,7 @@ def generate_load_tiled(plan, fams):
     V = max(1, 16 // max(items))
     TH = int(os.environ.get("RAMBA_LT_TH", str(LT_TH)))
     NCH = int(os.environ.get("RAMBA_LT_NCH", str(LT_TXCH)))
+    SEG = int(os.environ.get("RAMBA_LT_SEG", "8"))
     CW = NCH * V                     # tile output columns
 
     fam_members = {}
@@ -1666,19 +1666,21 @@ def generate_load_tiled(plan, fams):
         for (n, dr0, dr1) in mem:
             fam_members[n] = (fi, dr0, dr1)
         fam_ext.append((max(m[1] for m in mem), max(m[2] for m in mem)))
+    E0g = max(e[0] for e in fam_ext)     # shared ring depth
+    RING = TH + E0g
 
     other_ops = [op for op in plan.operands if op.name not in fam_members]
     written = {st.target for st in plan.statements}
 
     key = hashlib.sha256(repr((
-        "loadtiled2", [(st.target, st.expr) for st in plan.statements],
+        "loadtiled3", [(st.target, st.expr) for st in plan.statements],
         [(fi, dt, sorted(mem)) for fi, (a, s, dt, mem, rep)
          in enumerate(fams)], fam_ext,
         [(o.name, str(o.dtype), o.name in written, o.strides[1] == 0)
          for o in other_ops],
         sorted((n, str(dt)) for n, (v, dt) in plan.scalars.items()),
         sorted((n, str(dt)) for n, dt in plan.dead_vars.items()),
-        TH, NCH, V)).encode()).hexdigest()[:24]
+        TH, NCH, V, SEG)).encode()).hexdigest()[:24]
     kname = f"lt_{key}"
 
     fields = [("n0", "q"), ("n1", "q"), ("gs0", "q"), ("gs1", "q")]
@@ -1705,86 +1707,105 @@ def generate_load_tiled(plan, fams):
             L.append(f"  i64 {n};")
     L.append("};")
 
-    # unaligned-tolerant vector structs (global loads/stores of V elems)
     L.append(f'extern "C" __global__ void __launch_bounds__(256) '
              f"{kname}(LtArgs a) {{")
+    pitches = []
     for fi, (E0, E1) in enumerate(fam_ext):
         dt = fams[fi][2]
-        # pitch: multiple of V so vector LDS rows stay aligned
         FW = CW + E1
         pitch = -(-FW // V) * V + V
+        pitches.append(pitch)
         L.append(f"  __shared__ {ctype(dt)} lds_f{fi}"
-                 f"[{TH + E0}][{pitch}];")
-    L.append(f"  const i64 tiles1 = (a.n1 + {CW} - 1) / {CW};")
+                 f"[{RING}][{pitch}];")
+    L.append(f"  const i64 strips = (a.n1 + {CW} - 1) / {CW};")
     L.append(f"  const i64 tiles0 = (a.n0 + {TH} - 1) / {TH};")
+    L.append(f"  const i64 nseg = (tiles0 + {SEG} - 1) / {SEG};")
     L.append("  const int tx = threadIdx.x & 63;")
     L.append("  const int ty = threadIdx.x >> 6;")
-    L.append("  for (i64 tile = blockIdx.x; tile < tiles0 * tiles1; "
-             "tile += gridDim.x) {")
-    L.append(f"    const i64 k0o = (tile / tiles1) * {TH};")
-    L.append(f"    const i64 k1o = (tile % tiles1) * {CW};")
-    # ---- cooperative fill, vectorised -----------------------------------
+    L.append("  for (i64 work = blockIdx.x; work < strips * nseg; "
+             "work += gridDim.x) {")
+    L.append("    const i64 strip = work % strips;")
+    L.append("    const i64 seg = work / strips;")
+    L.append(f"    const i64 k1o = strip * {CW};")
+    L.append(f"    const i64 t0 = seg * {SEG};")
+    L.append(f"    i64 tmax_ = tiles0 - t0; "
+             f"const int tmax = (int)(tmax_ < {SEG} ? tmax_ : {SEG});")
+    L.append("    int base = 0;   // ring slot of input row k0o")
+    L.append("    for (int t = 0; t < tmax; ++t) {")
+    L.append(f"      const i64 k0o = (t0 + t) * {TH};")
+
+    # ---- fill: first tile loads RING rows, later tiles TH new rows ------
+    # new input rows for tile t: t==0 -> [k0o, k0o+RING), else
+    # [k0o+E0g, k0o+TH+E0g); clamped to < n0 + E0g
     for fi, (E0, E1) in enumerate(fam_ext):
         dt = fams[fi][2]
         ct = ctype(dt)
-        FH, FW = TH + E0, CW + E1
+        FW = CW + E1
         vct = {("double", 2): "d2_t", ("float", 4): "f4_t",
-               ("long long", 2): "l2_t", ("i64", 2): "l2_t",
-               ("int", 4): "i4_t", ("short", 4): "s4_t",  # wait: 16B/maxitem
-               }.get((ct, V))
-        L.append(f"    {{ i64 fh_ = a.n0 - k0o + {E0}; "
-                 f"const int fh = (int)(fh_ < {FH} ? fh_ : {FH});")
-        L.append(f"      i64 fw_ = a.n1 - k1o + {E1}; "
+               ("long long", 2): "l2_t", ("int", 4): "i4_t",
+               ("short", 4): "s4_t"}.get((ct, V))
+        L.append(f"      {{ const int rlo = t == 0 ? 0 : {E0g};")
+        L.append(f"        i64 rhi_ = a.n0 + {E0} - k0o; "
+                 f"const int rhi = (int)(rhi_ < {RING} ? rhi_ : {RING});")
+        L.append(f"        i64 fw_ = a.n1 - k1o + {E1}; "
                  f"const int fw = (int)(fw_ < {FW} ? fw_ : {FW});")
-        L.append(f"      const int nch = (fw + {V} - 1) / {V};")
-        L.append(f"      for (int r = ty; r < fh; r += 4) {{")
-        L.append(f"        const i64 rb = a.fam{fi}_off + (k0o + r) * "
+        L.append(f"        const int nch = (fw + {V} - 1) / {V};")
+        L.append("        for (int r = rlo + ty; r < rhi; r += 4) {")
+        L.append(f"          int slot = base + r; "
+                 f"if (slot >= {RING}) slot -= {RING};")
+        L.append(f"          const i64 rb = a.fam{fi}_off + (k0o + r) * "
                  f"a.fam{fi}_s0 + k1o;")
-        L.append(f"        for (int c = tx; c < nch; c += 64) {{")
-        L.append(f"          const int col = c * {V};")
+        L.append("          for (int c = tx; c < nch; c += 64) {")
+        L.append(f"            const int col = c * {V};")
         if vct:
-            L.append(f"          if (col + {V} <= fw) {{")
-            L.append(f"            {vct} v; __builtin_memcpy(&v, "
+            L.append(f"            if (col + {V} <= fw) {{")
+            L.append(f"              {vct} v; __builtin_memcpy(&v, "
                      f"&a.fam{fi}_ptr[rb + col], sizeof(v));")
-            L.append(f"            __builtin_memcpy(&lds_f{fi}[r][col], "
-                     f"&v, sizeof(v));")
-            L.append("          } else {")
-            L.append(f"            for (int j = 0; j < {V}; ++j) "
-                     f"if (col + j < fw) lds_f{fi}[r][col + j] = "
+            L.append(f"              __builtin_memcpy(&lds_f{fi}[slot]"
+                     f"[col], &v, sizeof(v));")
+            L.append("            } else {")
+            L.append(f"              for (int j = 0; j < {V}; ++j) "
+                     f"if (col + j < fw) lds_f{fi}[slot][col + j] = "
                      f"a.fam{fi}_ptr[rb + col + j];")
-            L.append("          }")
+            L.append("            }")
         else:
-            L.append(f"          for (int j = 0; j < {V}; ++j) "
-                     f"if (col + j < fw) lds_f{fi}[r][col + j] = "
+            L.append(f"            for (int j = 0; j < {V}; ++j) "
+                     f"if (col + j < fw) lds_f{fi}[slot][col + j] = "
                      f"a.fam{fi}_ptr[rb + col + j];")
-        L.append("        }")
-        L.append("      } }")
-    L.append("    __syncthreads();")
+        L.append("          }")
+        L.append("        } }")
+    L.append("      __syncthreads();")
 
-    # ---- stage 2: V outputs per thread ----------------------------------
+    # ---- stage 2 ---------------------------------------------------------
     def mk_resolver(jexpr):
-        def resolve0(base, em):
-            if base in fam_members:
-                fi, dr0, dr1 = fam_members[base]
-                return (f"lds_f{fi}[r2 + {dr0}]"
+        def resolve0(base_name, em):
+            if base_name in fam_members:
+                fi, dr0, dr1 = fam_members[base_name]
+                return (f"lds_f{fi}[sl{dr0}{em.tag}]"
                         f"[tx * {V} + {jexpr} + {dr1}]")
             for o in other_ops:
-                if o.name == base:
-                    return (f"a.{base}_ptr[a.{base}_off + k0_t * "
-                            f"a.{base}_s0 + (k1_t + {jexpr}) * "
-                            f"a.{base}_s1]")
+                if o.name == base_name:
+                    return (f"a.{base_name}_ptr[a.{base_name}_off + k0_t * "
+                            f"a.{base_name}_s0 + (k1_t + {jexpr}) * "
+                            f"a.{base_name}_s1]")
             return None
         return resolve0
 
-    L.append(f"    for (int r2 = ty; r2 < {TH}; r2 += 4) {{")
-    L.append("      const i64 k0_t = k0o + r2;")
-    L.append("      if (k0_t >= a.n0) break;")
-    L.append(f"      const i64 k1_t = k1o + tx * {V};")
-    L.append(f"      const bool full = (k1_t + {V} <= a.n1);")
-    # full path: unrolled j, vector stores
-    L.append("      if (full) {")
+    dr0s = sorted({dr0 for (fi, dr0, dr1) in fam_members.values()})
+    L.append(f"      for (int r2 = ty; r2 < {TH}; r2 += 4) {{")
+    L.append("        const i64 k0_t = k0o + r2;")
+    L.append("        if (k0_t >= a.n0) break;")
+    L.append(f"        const i64 k1_t = k1o + tx * {V};")
+    for dr0 in dr0s:
+        L.append(f"        int sl{dr0}_jf = base + r2 + {dr0}; "
+                 f"if (sl{dr0}_jf >= {RING}) sl{dr0}_jf -= {RING};")
+    L.append(f"        const bool full = (k1_t + {V} <= a.n1);")
+    L.append("        if (full) {")
+    # per-j emissions share the slot vars: alias per tag
     store_vecs = {}
     for j in range(V):
+        for dr0 in dr0s:
+            L.append(f"        const int sl{dr0}_j{j} = sl{dr0}_jf;")
         genj = _StageGen(plan.statements, mk_resolver(str(j)),
                          plan.dead_vars)
         em = LaneEmitter(genj, f"_j{j}",
@@ -1804,20 +1825,21 @@ def generate_load_tiled(plan, fams):
                ("unsigned char", 4): "b4_t"}.get((ct, V))
         vals = store_vecs[o.name]
         if vct and o.strides[1] == 1:
-            L.append(f"      {{ {vct} sv; " + " ".join(
-                f"sv[{j}] = {vals[j]};" for j in range(V)) + "")
-            L.append(f"        __builtin_memcpy(&a.{o.name}_ptr["
+            L.append(f"        {{ {vct} sv; " + " ".join(
+                f"sv[{j}] = {vals[j]};" for j in range(V)))
+            L.append(f"          __builtin_memcpy(&a.{o.name}_ptr["
                      f"a.{o.name}_off + k0_t * a.{o.name}_s0 + k1_t], "
                      f"&sv, sizeof(sv)); }}")
         else:
             for j in range(V):
-                L.append(f"      a.{o.name}_ptr[a.{o.name}_off + k0_t * "
-                         f"a.{o.name}_s0 + (k1_t + {j}) * a.{o.name}_s1]"
-                         f" = {vals[j]};")
-    L.append("      } else {")
-    # edge path: scalar with per-element guard
-    L.append(f"        for (int j = 0; j < {V}; ++j) {{")
-    L.append("          if (k1_t + j >= a.n1) break;")
+                L.append(f"        a.{o.name}_ptr[a.{o.name}_off + k0_t *"
+                         f" a.{o.name}_s0 + (k1_t + {j}) * "
+                         f"a.{o.name}_s1] = {vals[j]};")
+    L.append("        } else {")
+    L.append(f"          for (int j = 0; j < {V}; ++j) {{")
+    L.append("            if (k1_t + j >= a.n1) break;")
+    for dr0 in dr0s:
+        L.append(f"            const int sl{dr0}_je = sl{dr0}_jf;")
     gene = _StageGen(plan.statements, mk_resolver("j"), plan.dead_vars)
     eme = LaneEmitter(gene, "_je",
                       ["(a.gs0 + k0_t)", "(a.gs1 + k1_t + j)"])
@@ -1825,13 +1847,15 @@ def generate_load_tiled(plan, fams):
     L.extend(eme.lines)
     for o in other_ops:
         if o.name in written:
-            L.append(f"          a.{o.name}_ptr[a.{o.name}_off + k0_t * "
-                     f"a.{o.name}_s0 + (k1_t + j) * a.{o.name}_s1] = "
+            L.append(f"            a.{o.name}_ptr[a.{o.name}_off + k0_t *"
+                     f" a.{o.name}_s0 + (k1_t + j) * a.{o.name}_s1] = "
                      f"{finals_e[o.name]};")
+    L.append("          }")
     L.append("        }")
     L.append("      }")
+    L.append("      __syncthreads();")
+    L.append(f"      base += {TH}; if (base >= {RING}) base -= {RING};")
     L.append("    }")
-    L.append("    __syncthreads();")
     L.append("  }")
     L.append("}")
-    return key, "\n".join(L), kname, fields, (TH, CW)
+    return key, "\n".join(L), kname, fields, (TH, CW, SEG)

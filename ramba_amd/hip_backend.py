@@ -1027,8 +1027,9 @@ def _hb_launch_load_tiled(self, gk, plan):
     for n, (v, dt) in plan.scalars.items():
         vals[n] = v
     args = cg.pack_tk_args(fields, vals)
-    th, cw = tile
-    ntiles = ((plan.itershape[0] + th - 1) // th) \
+    th, cw, seg = tile
+    tiles0 = (plan.itershape[0] + th - 1) // th
+    ntiles = ((tiles0 + seg - 1) // seg) \
         * ((plan.itershape[1] + cw - 1) // cw)
     gx = max(1, min(int(os.environ.get("RAMBA_GRID_CAP", "32768")),
                     ntiles))
