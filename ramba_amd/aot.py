@@ -180,14 +180,19 @@ def seed(fuzz_seeds=None):
     Y[1:-1, 1:-1] = (X[:-2, 1:-1] + X[2:, 1:-1] + X[1:-1, :-2]
                      + X[1:-1, 2:] - 4.0 * X[1:-1, 1:-1])
     ra.sync()
-    # configs[3] at load-tiled scale (>=2^16 elements) so the LDS
-    # stencil kernel lands in the cache (key is size-independent)
-    X2 = ra.fromfunction(lambda x, y: x + y, (512, 512), dtype=np.float32)
-    Y2 = ra.zeros((512, 512), dtype=np.float32)
-    ra.sync()
-    Y2[1:-1, 1:-1] = (X2[:-2, 1:-1] + X2[2:, 1:-1] + X2[1:-1, :-2]
-                      + X2[1:-1, 2:] - 4.0 * X2[1:-1, 1:-1])
-    ra.sync()
+    # configs[3] at the bench geometries so the adaptive LDS stencil
+    # kernels land in the cache (tile geometry keys on the iteration
+    # shape class; no data is allocated on the aot backend)
+    for SS in (512, 4096, 30000):
+        X2 = ra.fromfunction(lambda x, y: x + y, (SS, SS),
+                             dtype=np.float32)
+        Y2 = ra.zeros((SS, SS), dtype=np.float32)
+        ra.sync()
+        Y2[1:-1, 1:-1] = (X2[:-2, 1:-1] + X2[2:, 1:-1] + X2[1:-1, :-2]
+                          + X2[1:-1, 2:] - 4.0 * X2[1:-1, 1:-1])
+        ra.sync()
+        del X2, Y2
+        ra.sync()
     # configs[4] mixed fp64 (Z materialised first so the pair takes the
     # staged/tiled path, like the bench loop's steady state)
     Z = ra.zeros((64, 64), dtype=np.float64)

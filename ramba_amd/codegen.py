@@ -1655,10 +1655,23 @@ def generate_load_tiled(plan, fams):
             continue
         items.append(np.dtype(op.dtype).itemsize)
     V = max(1, 16 // max(items))
-    TH = int(os.environ.get("RAMBA_LT_TH", str(LT_TH)))
-    NCH = int(os.environ.get("RAMBA_LT_NCH", str(LT_TXCH)))
-    SEG = int(os.environ.get("RAMBA_LT_SEG", "8"))
+    # adaptive tile geometry (same-box sweeps, profiles/README r02):
+    # wider strips cut the per-strip boundary-line over-fetch
+    # (~strips*128B per array row), shorter tiles keep LDS/occupancy;
+    # SEG shrinks when the (strips x segments) grid would starve 256 CUs
+    n0g, n1g = plan.itershape
+    nch_d = 128 if n1g >= 8192 else 64
+    th_d = 8 if nch_d == 128 else LT_TH
+    TH = int(os.environ.get("RAMBA_LT_TH", str(th_d)))
+    NCH = int(os.environ.get("RAMBA_LT_NCH", str(nch_d)))
     CW = NCH * V                     # tile output columns
+    strips_d = -(-n1g // CW)
+    tiles0_d = -(-n0g // TH)
+    seg_d = 1
+    for cand in (2, 4, 8, 16):
+        if -(-tiles0_d // cand) * strips_d >= 6144:
+            seg_d = cand
+    SEG = int(os.environ.get("RAMBA_LT_SEG", str(seg_d)))
 
     fam_members = {}
     fam_ext = []
