@@ -1668,7 +1668,7 @@ def generate_load_tiled(plan, fams):
     strips_d = -(-n1g // CW)
     tiles0_d = -(-n0g // TH)
     seg_d = 1
-    for cand in (2, 4, 8, 16):
+    for cand in (2, 4, 8):
         if -(-tiles0_d // cand) * strips_d >= 6144:
             seg_d = cand
     SEG = int(os.environ.get("RAMBA_LT_SEG", str(seg_d)))
