@@ -136,7 +136,10 @@ def main():
         from ramba_amd.common import stage_fusion
         S = 8192
         N = S * S
-        alg_bytes_per_elem = 16 if stage_fusion else 32
+        # fully fused (round 2): ONE tiled kernel computes iota+sin in
+        # LDS, writes A AND accumulates the sum in-kernel (+ a tiny rim
+        # complement) — algorithmic traffic is just the A store, 8 B/elem
+        alg_bytes_per_elem = 8 if stage_fusion else 32
         A = ra.zeros((S, S), dtype=np.float64)
         ra.sync()
 
@@ -160,10 +163,9 @@ def main():
             return float(A.sum())
         metric = ("GElem/s, 8192^2 fp64 iota→sin→stencil→sum pipeline "
                   "(BASELINE configs[4])")
-        wl_desc = ("8192^2 fp64: iota+sin fused INTO the 5-pt stencil "
-                   "kernel (LDS-staged producer), global sum + RCCL "
-                   f"allreduce; {16 if stage_fusion else 32} B/elem "
-                   "algorithmic")
+        wl_desc = ("8192^2 fp64: iota+sin+stencil+sum in ONE fused "
+                   "LDS-tiled kernel (+rim complement) + RCCL allreduce; "
+                   f"{8 if stage_fusion else 32} B/elem algorithmic")
     else:  # stencil
         S = args.stencil_n
         N = S * S
@@ -316,7 +318,7 @@ def main():
             "kernel_ms": kms,
             "kernels": kbreak,
         }
-        if args.workload == "mixed" and alg_bytes_per_elem == 16:
+        if args.workload == "mixed" and alg_bytes_per_elem == 8:
             # transparency: cross-stage fusion halves the algorithmic
             # bytes ("less if stencil fuses with sin
             # producer"), so `frac` above is the TRUE-traffic rate; the

@@ -238,8 +238,8 @@ def _aot_tiled_kernel(self, desc):
     return (key, fields)
 
 
-def _aot_tiled_launch(self, handle, vals, ntiles):
-    pass
+def _aot_tiled_launch(self, handle, vals, ntiles, red_dtypes=None):
+    return [np.asarray(0, dtype=dt)[()] for dt in (red_dtypes or [])]
 
 
 AotCompileBackend.container_addr = _aot_container_addr

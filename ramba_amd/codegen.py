@@ -1428,6 +1428,7 @@ def staged_tiled_key(desc):
         desc["staged"], desc["s1_stores"], sorted(desc["readers"].items()),
         desc["s2_ops"], sorted((n, str(d)) for n, d in
                                desc["scalars"].items()),
+        desc.get("tk_reds", []),
         desc["E0"], desc["E1"], TILE_H, TILE_W)).encode()).hexdigest()[:24]
 
 
@@ -1487,10 +1488,16 @@ def generate_staged_tiled(desc):
         else:
             fields.append((n, "q"))
             L.append(f"  i64 {n};")
+    tk_reds = desc.get("tk_reds", [])
+    for ri, (wvar, dt) in enumerate(tk_reds):
+        L.append(f"  {ctype(dt)}* __restrict__ red{ri}_ptr;")
+        fields.append((f"red{ri}_ptr", "Q"))
     L.append("};")
 
     L.append(f'extern "C" __global__ void __launch_bounds__(256) '
              f"{kname}(TkArgs a) {{")
+    for ri, (wvar, dt) in enumerate(tk_reds):
+        L.append(f"  {ctype(dt)} red{ri}_acc = ({ctype(dt)})0;")
     for (lds, dt, live) in desc["staged"]:
         L.append(f"  __shared__ {ctype(dt)} lds_{lds}[{FH}][{FW} + 1];")
     L.append(f"  const i64 tiles1 = (a.n1 + {TW} - 1) / {TW};")
@@ -1562,10 +1569,34 @@ def generate_staged_tiled(desc):
             L.append(
                 f"      a.{var}_ptr[a.{var}_off + k0_t2 * a.{var}_s0 + "
                 f"k1_t2 * a.{var}_s1] = {finals2[var]};")
+    for ri, (wvar, dt) in enumerate(tk_reds):
+        L.append(f"      red{ri}_acc += ({ctype(dt)})({finals2[wvar]});")
     L.append("      }")
     L.append("    }")
     L.append("    __syncthreads();")
     L.append("  }")
+    if tk_reds:
+        # per-block tree reduce of the fused-sum accumulators (the same
+        # wave shfl + cross-wave LDS idiom as the fused reduce kernels)
+        L.append("  {")
+        L.append("    const int lane = threadIdx.x & 63;")
+        L.append("    const int wid = threadIdx.x >> 6;")
+        for ri, (wvar, dt) in enumerate(tk_reds):
+            ct = ctype(dt)
+            L.append("    for (int o = 32; o > 0; o >>= 1)")
+            L.append(f"      red{ri}_acc += "
+                     f"__shfl_down(red{ri}_acc, o, 64);")
+            L.append(f"    __shared__ {ct} lds_red{ri}[4];")
+            L.append(f"    if (lane == 0) lds_red{ri}[wid] = red{ri}_acc;")
+        L.append("    __syncthreads();")
+        L.append("    if (threadIdx.x == 0) {")
+        for ri, (wvar, dt) in enumerate(tk_reds):
+            ct = ctype(dt)
+            L.append(f"      {ct} t_red{ri} = lds_red{ri}[0] + "
+                     f"lds_red{ri}[1] + lds_red{ri}[2] + lds_red{ri}[3];")
+            L.append(f"      a.red{ri}_ptr[blockIdx.x] = t_red{ri};")
+        L.append("    }")
+        L.append("  }")
     L.append("}")
     return key, "\n".join(L), kname, fields
 

@@ -119,6 +119,10 @@ class FusedGroup:
         # ramba.py:8434-8443)
         self.producer = None        # sealed stage-1 FusedGroup or None
         self.staged_gids = set()    # gids served from LDS recompute
+        # sum(A) immediately after the pair, where the consumer wrote a
+        # sub-box of A: fused into the tiled kernel (interior partials)
+        # + a tiny complement reduce (runtime/staged_exec)
+        self.staged_reductions = [] # [(src ndarray, written_view, pend)]
 
     # -- naming -------------------------------------------------------------
 
@@ -491,6 +495,28 @@ def add_reduction(src_tree, kind, dtype):
     operands = _tree_ndarrays(src_tree, [])
     assert operands, "reduction of non-array"
     arr = operands[0]
+    # sum(A) arriving while a sealed producer/consumer pair is pending,
+    # with A's sub-box written by the consumer: fuse the reduction into
+    # the pair (interior accumulated in the tiled kernel + complement
+    # boxes reduced separately) instead of flushing on the shape
+    # mismatch.  The caller flushes right after, so no further op can
+    # slip between the pair and the reduction.
+    g = _state["group"]
+    if (g is not None and g.producer is not None and kind == "sum"
+            and _is_ndarray(src_tree)
+            and src_tree.view.is_identity_for(src_tree.bdarray.shape)
+            and np.dtype(dtype) == src_tree.bdarray.dtype):
+        gid = src_tree.bdarray.gid
+        wvs = [wv for (wg, wv) in g.write_views if wg == gid]
+        nd = len(src_tree.bdarray.shape)
+        if (len(wvs) == 1 and wvs[0].ndim == nd
+                and wvs[0].axis_map == tuple(range(nd))
+                and wvs[0].steps == (1,) * nd
+                and not any(rg == gid and rv != wvs[0]
+                            for (rg, rv) in g.read_views)):
+            pend = PendingReduction(kind, np.dtype(dtype))
+            g.staged_reductions.append((src_tree, wvs[0], pend))
+            return pend
     g = _ensure_group(arr)
 
     reads = []

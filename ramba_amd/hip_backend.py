@@ -936,12 +936,18 @@ def _hb_tiled_kernel(self, desc):
     return cached
 
 
-def _hb_tiled_launch(self, handle, vals, ntiles):
+def _hb_tiled_launch(self, handle, vals, ntiles, red_dtypes=None):
     from . import codegen as cg
     h, fields = handle
-    args = cg.pack_tk_args(fields, vals)
     gx = max(1, min(int(os.environ.get("RAMBA_GRID_CAP", "32768")),
                     ntiles))
+    red_dtypes = red_dtypes or []
+    parts = []
+    for ri, dt in enumerate(red_dtypes):
+        t = self.torch.empty(gx, dtype=self._tdt(dt), device="cuda")
+        parts.append(t)
+        vals[f"red{ri}_ptr"] = t.data_ptr()
+    args = cg.pack_tk_args(fields, vals)
     if self.time_kernels:
         ev0 = self.torch.cuda.Event(enable_timing=True)
         ev1 = self.torch.cuda.Event(enable_timing=True)
@@ -954,6 +960,17 @@ def _hb_tiled_launch(self, handle, vals, ntiles):
         ev1.synchronize()
         self.kernel_times_ms.append(ev0.elapsed_time(ev1))
         self.kernel_keys.append("tiled")
+    out = []
+    for t, dt in zip(parts, red_dtypes):
+        total = self.torch.empty(1, dtype=t.dtype, device="cuda")
+        rc = self.lib.rt_cumsum(
+            self._stream(), None, 0, 0, 0, None, 0,
+            ctypes.c_void_p(t.data_ptr()), gx,
+            ctypes.c_void_p(total.data_ptr()), 0.0, 0,
+            _CS_DT[str(np.dtype(dt))], 2)
+        self._check(rc, "rt_cumsum(tk partials)")
+        out.append(np.asarray(total.cpu().numpy()[0], dtype=dt)[()])
+    return out
 
 
 HipBackend.container_addr = _hb_container_addr

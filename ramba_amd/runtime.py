@@ -605,6 +605,13 @@ class Runtime:
         g2.delete_bds.extend(keep)
         self._execute_group_with(g1, live, dead)
         self._execute_group(g2)
+        if g2.staged_reductions:
+            # the reduction that was fused onto the pair: the sequential
+            # order recomputes it over the full local core
+            from . import staged_exec
+            staged_exec.finish_staged_reductions(rt=self, g2=g2,
+                                                 interior=None,
+                                                 written_images={})
 
     def _execute_group_with(self, group, live, dead):
         """_execute_group with a precomputed liveness split (recipe cache

@@ -48,6 +48,10 @@ def try_execute_tiled(rt, g1, g2):
         return False
     if g2.reductions:
         return False
+    for (src, wv, pend) in g2.staged_reductions:
+        if pend.kind != "sum" or str(pend.dtype) not in (
+                "float64", "float32", "int64", "int32"):
+            return False
     if not deferred.group_is_index_pure(g1):
         return False
     readers = classify_readers(g2, g2.staged_gids)
@@ -73,7 +77,10 @@ def try_execute_tiled(rt, g1, g2):
     cache = rt.__dict__.setdefault("_staged_cache", {})
     try:
         sig = (rt._group_signature(g1, live1),
-               rt._group_signature(g2, live2))
+               rt._group_signature(g2, live2),
+               tuple((p.kind, str(p.dtype), src.bdarray.gid in
+                      {oi.bd.gid for oi in live2.values() if oi.written})
+                     for (src, wv, p) in g2.staged_reductions))
     except TypeError:
         sig = None
     ent = cache.get(sig) if sig is not None else None
@@ -182,13 +189,28 @@ def _validate_and_build(rt, g1, g2, live1, dead1, live2, dead2, readers):
               for n in s2_names]
     scalars = {("p_" + n): dt for n, (v, dt) in g1.scalars.items()}
     scalars.update({n: dt for n, (v, dt) in g2.scalars.items()})
+    # fused reductions: accumulate the consumer-written value of the
+    # reduction source's written var
+    tk_reds = []
+    red_ok = True
+    for (src, wv, pend) in g2.staged_reductions:
+        wvar = None
+        for name, oi in live2.items():
+            if oi.bd.gid == src.bdarray.gid and oi.written:
+                wvar = name
+        if wvar is None:
+            red_ok = False
+            break
+        tk_reds.append((wvar, str(pend.dtype)))
+    if not red_ok:
+        return None
     desc = {
         "s1_stmts": s1_stmts, "s2_stmts": g2.statements,
         "staged": staged_list, "s1_stores": s1_stores,
         "readers": reader_map, "s2_ops": s2_ops, "scalars": scalars,
         "dead1": {("p_" + n): oi.dtype for n, oi in dead1.items()},
         "dead2": {n: oi.dtype for n, oi in dead2.items()},
-        "E0": E[0], "E1": E[1],
+        "E0": E[0], "E1": E[1], "tk_reds": tk_reds,
     }
     handle = backend.tiled_kernel(desc)
     if handle is None:
@@ -202,4 +224,5 @@ def _validate_and_build(rt, g1, g2, live1, dead1, live2, dead2, readers):
     rec.F_lo = F_lo
     rec.E = E
     rec.lds_of_writer = lds_of_writer
+    rec.nred = len(tk_reds)
     return rec

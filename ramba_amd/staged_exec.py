@@ -12,9 +12,10 @@ from .shardview import box_shape, box_subtract
 class StagedRecipe:
     __slots__ = ("handle", "writer_names", "store_names", "s2_names",
                  "F_lo", "E", "hbm_read_names", "residual_units",
-                 "lds_of_writer")
+                 "lds_of_writer", "nred")
 
     def __init__(self):
+        self.nred = 0
         self.handle = None
         self.writer_names = []      # g1 var names of staged writers
         self.store_names = []       # g1 var names of non-staged stores
@@ -125,7 +126,9 @@ def build_vals_and_launch(rt, rec, g1, g2, live1, live2, ib2):
         vals[n] = v
     ntiles = ((n0 + codegen.TILE_H - 1) // codegen.TILE_H) \
         * ((n1 + codegen.TILE_W - 1) // codegen.TILE_W)
-    backend.tiled_launch(rec.handle, vals, ntiles)
+    return backend.tiled_launch(rec.handle, vals, ntiles,
+                                [dt for (_, _, p) in g2.staged_reductions
+                                 for dt in [p.dtype]])
 
 
 def run_residual(rt, rec, g1, live1, dead1, ib1, ib2):
@@ -170,12 +173,94 @@ def run_recipe(rt, rec, g1, g2, live1, dead1, live2, hbm_read_gids):
     adopt_and_alloc(rt, g2, live2)
     ib1 = g1.exec_boxes()[rt.rank]
     ib2 = g2.exec_boxes()[rt.rank]
+    interior = None
     if ib2 is not None:
-        build_vals_and_launch(rt, rec, g1, g2, live1, live2, ib2)
+        interior = build_vals_and_launch(rt, rec, g1, g2, live1, live2,
+                                         ib2)
     run_residual(rt, rec, g1, live1, dead1, ib1, ib2)
+    if g2.staged_reductions:
+        wimgs = {}
+        if ib2 is not None:
+            for (src, wv, pend) in g2.staged_reductions:
+                wimgs[src.bdarray.gid] = wv.image_box(ib2)
+        finish_staged_reductions(rt, g2, interior, wimgs)
     backend.free_temps()
     for bd in g1.delete_bds + g2.delete_bds:
         if bd.constructed:
             backend.free_container(bd)
             bd.constructed = False
     return True
+
+
+# -- staged reductions (sum(A) fused into the pair) --------------------------
+
+
+def reduce_boxes_partial(rt, bd, boxes, kind, dtype):
+    """Partial reduction of identity-view `bd` over base-space `boxes`
+    on this rank, via the standard fused-reduce kernel (one launch per
+    box; plan cached by the backend's structural key)."""
+    from . import ir
+    from .runtime import KernelPlan, OperandPlan
+    backend = rt.backend
+    acc = np.asarray(ir.reduction_init(kind, dtype), dtype=dtype)[()]
+    d, _, cs, pads = rt.shard_geometry(bd)
+    if d is None:
+        return acc
+    comb, _ = ir.REDUCTIONS[kind]
+    for box in boxes:
+        shape = box_shape(box)
+        if 0 in shape:
+            continue
+        p = KernelPlan()
+        p.itershape = shape
+        p.global_start = tuple(int(x) for x in box[0])
+        p.scalars = {}
+        src = ir.Ref("vR", bd.dtype)
+        if np.dtype(bd.dtype) != np.dtype(dtype):
+            src = ir.Cast(src, np.dtype(dtype))
+        spec = ir.ReductionSpec("accR", kind, np.dtype(dtype), 0)
+        p.statements = [ir.Assign(
+            "accR", ir.Bin(comb, ir.Ref("accR", np.dtype(dtype)), src,
+                           np.dtype(dtype)))]
+        p.reductions = [spec]
+        p.dead_vars = {}
+        nd = len(bd.shape)
+        off0 = sum((int(box[0, i]) - int(d[0, i]) + pads[i]) * cs[i]
+                   for i in range(nd))
+        p.operands = [OperandPlan("vR", "container", bd, None, off0,
+                                  tuple(cs), bd.dtype)]
+        vals = backend.launch(p, None)
+        v = vals[0]
+        acc = np.asarray(ir.BINOPS[comb](acc, v) if comb in ir.BINOPS
+                         else acc + v, dtype=dtype)[()]
+    return acc
+
+
+def finish_staged_reductions(rt, g2, interior, written_images):
+    """Set each staged-reduction pend.partial = interior partial (from
+    the tiled kernel, or identity) + the complement boxes of this rank's
+    core."""
+    from . import ir
+    from .shardview import box_subtract as _bsub
+    for i, (src, wv, pend) in enumerate(g2.staged_reductions):
+        bd = src.bdarray
+        core = rt.core_box(bd, rt.rank)
+        if core is None:
+            pend.partial = np.asarray(
+                ir.reduction_init(pend.kind, pend.dtype),
+                dtype=pend.dtype)[()]
+            continue
+        wimg = written_images.get(bd.gid)
+        if wimg is None:
+            comp = [core]
+        else:
+            comp = _bsub(core, wimg)
+        cval = reduce_boxes_partial(rt, bd, comp, pend.kind, pend.dtype)
+        ival = interior[i] if interior is not None else None
+        if ival is None:
+            pend.partial = cval
+        else:
+            comb, _ = ir.REDUCTIONS[pend.kind]
+            pend.partial = np.asarray(
+                ir.BINOPS[comb](np.asarray(ival, dtype=pend.dtype)[()],
+                                cval), dtype=pend.dtype)[()]

@@ -455,3 +455,26 @@ def test_overlap_split_stencil_spmd(world, overlap):
         outs.append(out.decode())
         ok = ok and p.returncode == 0
     assert ok, "\n==== rank outputs ====\n" + "\n----\n".join(outs)
+
+
+@pytest.mark.parametrize("world", [2, 4])
+def test_staged_sum_fusion_spmd(world):
+    """The fused sum(A)-after-stencil pair at world>1 (sequential
+    fallback on the oracle backend; partial = full-core reduce +
+    allreduce)."""
+    run_spmd("""
+        S = 128
+        A = np_.zeros((S, S), dtype=np.float64)
+        out = []
+        for it in range(3):
+            src = np_.fromfunction(
+                lambda x, y: (x * S + y + it) * 1e-4, (S, S),
+                dtype=np.float64)
+            ssin = np_.sin(src)
+            A[1:-1, 1:-1] = (ssin[:-2, 1:-1] + ssin[2:, 1:-1]
+                             + ssin[1:-1, :-2] + ssin[1:-1, 2:]
+                             - 4.0 * ssin[1:-1, 1:-1])
+            del src, ssin
+            out.append(float(A.sum()))
+        return np.asarray(out)
+    """, world=world, tol=1e-9)

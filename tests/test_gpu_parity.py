@@ -709,3 +709,60 @@ class TestLoadTiledStencil:
                              + A[1:-1, 2:]) * W[1:-1, 1:-1] + 0.125
             return B
         run_both(impl, ra_gpu, tol=1e-12)
+
+
+class TestStagedReduction:
+    """sum(A) fused into the staged pair: interior accumulated in the
+    tiled kernel + rim complement reduce (no full re-read of A)."""
+
+    def test_fused_sum_parity(self, ra_gpu):
+        import ramba_amd.deferred as D
+        S = 600
+        A = ra_gpu.zeros((S, S), dtype=np.float64)
+        ra_gpu.sync()
+        tots = []
+        seen_fused = []
+        for it in range(3):
+            src = ra_gpu.fromfunction(
+                lambda x, y: (x * S + y + it) * 1e-5, (S, S),
+                dtype=np.float64)
+            ssin = ra_gpu.sin(src)
+            A[1:-1, 1:-1] = (ssin[:-2, 1:-1] + ssin[2:, 1:-1]
+                             + ssin[1:-1, :-2] + ssin[1:-1, 2:]
+                             - 4.0 * ssin[1:-1, 1:-1])
+            del src, ssin
+            g = D.current_group()
+            tots.append(float(A.sum()))
+            seen_fused.append(g is not None
+                              and len(g.staged_reductions) == 1)
+        assert all(seen_fused), seen_fused
+        a = np.zeros((S, S))
+        for it in range(3):
+            sr = np.fromfunction(lambda x, y: (x * S + y + it) * 1e-5,
+                                 (S, S))
+            ss = np.sin(sr)
+            a[1:-1, 1:-1] = (ss[:-2, 1:-1] + ss[2:, 1:-1] + ss[1:-1, :-2]
+                             + ss[1:-1, 2:] - 4.0 * ss[1:-1, 1:-1])
+            ref = a.sum()
+            assert abs(tots[it] - ref) < 1e-9 * max(1.0, abs(ref)), \
+                (it, tots[it], ref)
+        # and A itself is intact
+        np.testing.assert_allclose(A.asarray(), a, rtol=1e-12, atol=1e-12)
+
+    def test_fused_sum_with_live_intermediate(self, ra_gpu):
+        """ssin kept alive: tiled kernel stores it AND accumulates."""
+        S = 300
+        A = ra_gpu.zeros((S, S), dtype=np.float64)
+        ra_gpu.sync()
+        src = ra_gpu.fromfunction(lambda x, y: (x + 3.0 * y) * 1e-3,
+                                  (S, S), dtype=np.float64)
+        ssin = ra_gpu.cos(src)
+        A[1:-1, 1:-1] = ssin[:-2, 1:-1] - ssin[2:, 1:-1]
+        s = float(A.sum())
+        a = np.zeros((S, S))
+        sr = np.fromfunction(lambda x, y: (x + 3.0 * y) * 1e-3, (S, S))
+        ss = np.cos(sr)
+        a[1:-1, 1:-1] = ss[:-2, 1:-1] - ss[2:, 1:-1]
+        assert abs(s - a.sum()) < 1e-9 * max(1.0, abs(a.sum()))
+        np.testing.assert_allclose(ssin.asarray(), ss, rtol=1e-12,
+                                   atol=1e-12)
