@@ -1492,6 +1492,21 @@ def generate_staged_tiled(desc):
     for ri, (wvar, dt) in enumerate(tk_reds):
         L.append(f"  {ctype(dt)}* __restrict__ red{ri}_ptr;")
         fields.append((f"red{ri}_ptr", "Q"))
+        # base-coordinate addressing of the reduction SOURCE array + up
+        # to 4 rim boxes (core minus written image): the kernel folds
+        # the unwritten rim cells into the same partials, so the fused
+        # sum needs no extra launches at all
+        L.append(f"  {ctype(dt)}* __restrict__ rsrc{ri}_ptr; "
+                 f"i64 rsrc{ri}_off, rsrc{ri}_s0, rsrc{ri}_s1, "
+                 f"rim{ri}_n;")
+        fields += [(f"rsrc{ri}_ptr", "Q"), (f"rsrc{ri}_off", "q"),
+                   (f"rsrc{ri}_s0", "q"), (f"rsrc{ri}_s1", "q"),
+                   (f"rim{ri}_n", "q")]
+        for k in range(4):
+            L.append(f"  i64 rim{ri}_{k}_lo0, rim{ri}_{k}_hi0, "
+                     f"rim{ri}_{k}_lo1, rim{ri}_{k}_hi1;")
+            fields += [(f"rim{ri}_{k}_lo0", "q"), (f"rim{ri}_{k}_hi0", "q"),
+                       (f"rim{ri}_{k}_lo1", "q"), (f"rim{ri}_{k}_hi1", "q")]
     L.append("};")
 
     L.append(f'extern "C" __global__ void __launch_bounds__(256) '
@@ -1576,6 +1591,26 @@ def generate_staged_tiled(desc):
     L.append("    __syncthreads();")
     L.append("  }")
     if tk_reds:
+        # rim cells (core minus written image) folded into the same
+        # accumulators: grid-strided over the <=4 rim boxes
+        for ri, (wvar, dt) in enumerate(tk_reds):
+            ct = ctype(dt)
+            for k in range(4):
+                L.append(f"  if ({k} < a.rim{ri}_n) {{")
+                L.append(f"    const i64 h0 = a.rim{ri}_{k}_hi0 - "
+                         f"a.rim{ri}_{k}_lo0 + 1;")
+                L.append(f"    const i64 h1 = a.rim{ri}_{k}_hi1 - "
+                         f"a.rim{ri}_{k}_lo1 + 1;")
+                L.append("    for (i64 i = (i64)blockIdx.x * 256 + "
+                         "threadIdx.x; i < h0 * h1; "
+                         "i += (i64)gridDim.x * 256) {")
+                L.append(f"      const i64 r = a.rim{ri}_{k}_lo0 + i / h1;")
+                L.append(f"      const i64 c = a.rim{ri}_{k}_lo1 + i % h1;")
+                L.append(f"      red{ri}_acc += a.rsrc{ri}_ptr["
+                         f"a.rsrc{ri}_off + r * a.rsrc{ri}_s0 + "
+                         f"c * a.rsrc{ri}_s1];")
+                L.append("    }")
+                L.append("  }")
         # per-block tree reduce of the fused-sum accumulators (the same
         # wave shfl + cross-wave LDS idiom as the fused reduce kernels)
         L.append("  {")

@@ -124,6 +124,33 @@ def build_vals_and_launch(rt, rec, g1, g2, live1, live2, ib2):
         vals["p_" + n] = v
     for n, (v, dt) in g2.scalars.items():
         vals[n] = v
+    # fused reductions: base-coordinate source addressing + rim boxes
+    # (core minus written image) folded in-kernel
+    for ri, (src, wv, pend) in enumerate(g2.staged_reductions):
+        bd = src.bdarray
+        d, _, cs, pads = rt.shard_geometry(bd)
+        assert d is not None    # written on this rank => shard exists
+        vals[f"rsrc{ri}_ptr"] = backend.container_addr(bd)
+        vals[f"rsrc{ri}_off"] = sum((pads[i] - int(d[0, i])) * cs[i]
+                                    for i in range(2))
+        vals[f"rsrc{ri}_s0"], vals[f"rsrc{ri}_s1"] = cs[0], cs[1]
+        core = rt.core_box(bd, rt.rank)
+        wimg = wv.image_box(ib2)
+        rims = box_subtract(core, wimg) if wimg is not None else [core]
+        assert len(rims) <= 4
+        vals[f"rim{ri}_n"] = len(rims)
+        for k in range(4):
+            if k < len(rims):
+                b = rims[k]
+                vals[f"rim{ri}_{k}_lo0"] = int(b[0, 0])
+                vals[f"rim{ri}_{k}_hi0"] = int(b[1, 0])
+                vals[f"rim{ri}_{k}_lo1"] = int(b[0, 1])
+                vals[f"rim{ri}_{k}_hi1"] = int(b[1, 1])
+            else:
+                vals[f"rim{ri}_{k}_lo0"] = 1
+                vals[f"rim{ri}_{k}_hi0"] = 0
+                vals[f"rim{ri}_{k}_lo1"] = 1
+                vals[f"rim{ri}_{k}_hi1"] = 0
     ntiles = ((n0 + codegen.TILE_H - 1) // codegen.TILE_H) \
         * ((n1 + codegen.TILE_W - 1) // codegen.TILE_W)
     return backend.tiled_launch(rec.handle, vals, ntiles,
@@ -179,11 +206,14 @@ def run_recipe(rt, rec, g1, g2, live1, dead1, live2, hbm_read_gids):
                                          ib2)
     run_residual(rt, rec, g1, live1, dead1, ib1, ib2)
     if g2.staged_reductions:
-        wimgs = {}
         if ib2 is not None:
-            for (src, wv, pend) in g2.staged_reductions:
-                wimgs[src.bdarray.gid] = wv.image_box(ib2)
-        finish_staged_reductions(rt, g2, interior, wimgs)
+            # the tiled kernel already folded the rim: partial = interior
+            for i, (src, wv, pend) in enumerate(g2.staged_reductions):
+                pend.partial = np.asarray(interior[i],
+                                          dtype=pend.dtype)[()]
+        else:
+            # no local tiles: reduce this rank's whole core (if any)
+            finish_staged_reductions(rt, g2, None, {})
     backend.free_temps()
     for bd in g1.delete_bds + g2.delete_bds:
         if bd.constructed:
