@@ -158,6 +158,28 @@ def test_axis_cumsum_and_mask_2rank_gpu():
     """, tol=1e-12)
 
 
+def test_mixed_fused_sum_2rank_gpu():
+    """configs[4] at world 2 with DEAD intermediates: the staged/tiled
+    kernel runs on both ranks (rank-uniform decision), each folding its
+    interior+rim partial, combined by allreduce."""
+    run_spmd_gpu("""
+        S = 512
+        A = np_.zeros((S, S), dtype=np.float64)
+        out = []
+        for it in range(3):
+            src = np_.fromfunction(
+                lambda x, y: (x * S + y + it) * 1e-6, (S, S),
+                dtype=np.float64)
+            ssin = np_.sin(src)
+            A[1:-1, 1:-1] = (ssin[:-2, 1:-1] + ssin[2:, 1:-1]
+                             + ssin[1:-1, :-2] + ssin[1:-1, 2:]
+                             - 4.0 * ssin[1:-1, 1:-1])
+            del src, ssin
+            out.append(float(A.sum()))
+        return np.asarray(out)
+    """, tol=1e-9)
+
+
 def test_mixed_pipeline_2rank_gpu():
     """The full configs[4] shape at world 2: fused fill, stencil with halo
     exchange, fused reduction + allreduce, eager force each step."""
