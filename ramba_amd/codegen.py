@@ -1429,7 +1429,10 @@ def staged_tiled_key(desc):
         desc["s2_ops"], sorted((n, str(d)) for n, d in
                                desc["scalars"].items()),
         desc.get("tk_reds", []),
-        desc["E0"], desc["E1"], TILE_H, TILE_W)).encode()).hexdigest()[:24]
+        desc["E0"], desc["E1"],
+        os.environ.get("RAMBA_TK_TH", str(TILE_H)),
+        os.environ.get("RAMBA_TK_CW", "128"),
+        os.environ.get("RAMBA_TK_SEG", "4"))).encode()).hexdigest()[:24]
 
 
 def generate_staged_tiled(desc):
@@ -1445,8 +1448,11 @@ def generate_staged_tiled(desc):
     returns (key, source, kname, fields): fields = ordered packing names.
     """
     E0, E1 = desc["E0"], desc["E1"]
-    TH, TW = TILE_H, TILE_W
-    FH, FW = TH + E0, TW + E1
+    TH = int(os.environ.get("RAMBA_TK_TH", str(TILE_H)))
+    TW = int(os.environ.get("RAMBA_TK_CW", "128"))
+    SEG = int(os.environ.get("RAMBA_TK_SEG", "4"))
+    RING = TH + E0
+    FH, FW = RING, TW + E1
 
     key = staged_tiled_key(desc)
     kname = f"tk_{key}"
@@ -1513,68 +1519,89 @@ def generate_staged_tiled(desc):
              f"{kname}(TkArgs a) {{")
     for ri, (wvar, dt) in enumerate(tk_reds):
         L.append(f"  {ctype(dt)} red{ri}_acc = ({ctype(dt)})0;")
+    # rolling LDS ring down column strips (same idea as the load-tiled
+    # stencil): the producer recomputes only TH NEW rows per tile, the
+    # E0 overlap rows stay in the ring — recompute amortises from
+    # (TH+E0)/TH to ~1 + E0/(SEG*TH)
     for (lds, dt, live) in desc["staged"]:
-        L.append(f"  __shared__ {ctype(dt)} lds_{lds}[{FH}][{FW} + 1];")
-    L.append(f"  const i64 tiles1 = (a.n1 + {TW} - 1) / {TW};")
+        L.append(f"  __shared__ {ctype(dt)} lds_{lds}[{RING}]"
+                 f"[{FW} + 1];")
+    L.append(f"  const i64 strips = (a.n1 + {TW} - 1) / {TW};")
     L.append(f"  const i64 tiles0 = (a.n0 + {TH} - 1) / {TH};")
-    L.append("  for (i64 tile = blockIdx.x; tile < tiles0 * tiles1; "
-             "tile += gridDim.x) {")
-    L.append("    const i64 k0o = (tile / tiles1) * %d;" % TH)
-    L.append("    const i64 k1o = (tile %% tiles1) * %d;" % TW)
+    L.append(f"  const i64 nseg = (tiles0 + {SEG} - 1) / {SEG};")
+    L.append("  for (i64 work = blockIdx.x; work < strips * nseg; "
+             "work += gridDim.x) {")
+    L.append("    const i64 strip = work % strips;")
+    L.append("    const i64 seg = work / strips;")
+    L.append(f"    const i64 k1o = strip * {TW};")
+    L.append(f"    const i64 t0 = seg * {SEG};")
+    L.append(f"    i64 tmax_ = tiles0 - t0; "
+             f"const int tmax = (int)(tmax_ < {SEG} ? tmax_ : {SEG});")
+    L.append("    int base = 0;")
+    L.append("    for (int t = 0; t < tmax; ++t) {")
+    L.append(f"      const i64 k0o = (t0 + t) * {TH};")
 
-    # ---- stage 1: footprint fill + stores --------------------------------
-    s1gen = _StageGen(desc["s1_stmts"], lambda b, em: None, desc["dead1"])
-    L.append(f"    for (int fi = threadIdx.x; fi < {FH} * {FW}; fi += 256) "
-             "{")
-    L.append(f"      const int f0 = fi / {FW}, f1 = fi % {FW};")
-    L.append("      const i64 B0 = a.gb0 + k0o + f0;")
-    L.append("      const i64 B1 = a.gb1 + k1o + f1;")
-    L.append("      if (B0 < 0 || B0 >= a.N0 || B1 < 0 || B1 >= a.N1) "
+    # ---- stage 1: compute NEW footprint rows into the ring ---------------
+    s1gen = _StageGen(desc["s1_stmts"], lambda b_, em: None, desc["dead1"])
+    L.append(f"      const int rlo = t == 0 ? 0 : {E0};")
+    L.append(f"      for (int fi = threadIdx.x; fi < ({RING} - rlo) * {FW};"
+             " fi += 256) {")
+    L.append(f"        const int r = rlo + fi / {FW};")
+    L.append(f"        const int f1 = fi % {FW};")
+    L.append(f"        int slot = base + r; "
+             f"if (slot >= {RING}) slot -= {RING};")
+    L.append("        const i64 B0 = a.gb0 + k0o + r;")
+    L.append("        const i64 B1 = a.gb1 + k1o + f1;")
+    L.append("        if (B0 < 0 || B0 >= a.N0 || B1 < 0 || B1 >= a.N1) "
              "continue;")
     em1 = LaneEmitter(s1gen, "_s1", ["B0", "B1"])
     finals = s1gen.emit_into(em1)
     body = list(em1.lines)
     for (lds, dt, live) in desc["staged"]:
-        body.append(f"      lds_{lds}[f0][f1] = {finals[lds]};")
+        body.append(f"        lds_{lds}[slot][f1] = {finals[lds]};")
         if live:
             body.append(
-                f"      if (B0 >= a.{lds}_lo0 && B0 <= a.{lds}_hi0 && "
+                f"        if (B0 >= a.{lds}_lo0 && B0 <= a.{lds}_hi0 && "
                 f"B1 >= a.{lds}_lo1 && B1 <= a.{lds}_hi1)")
             body.append(
-                f"        a.{lds}_ptr[a.{lds}_off + B0 * a.{lds}_s0 + "
+                f"          a.{lds}_ptr[a.{lds}_off + B0 * a.{lds}_s0 + "
                 f"B1 * a.{lds}_s1] = {finals[lds]};")
     for (var, dt) in desc["s1_stores"]:
         body.append(
-            f"      if (B0 >= a.{var}_lo0 && B0 <= a.{var}_hi0 && "
+            f"        if (B0 >= a.{var}_lo0 && B0 <= a.{var}_hi0 && "
             f"B1 >= a.{var}_lo1 && B1 <= a.{var}_hi1)")
         body.append(
-            f"        a.{var}_ptr[a.{var}_off + B0 * a.{var}_s0 + "
+            f"          a.{var}_ptr[a.{var}_off + B0 * a.{var}_s0 + "
             f"B1 * a.{var}_s1] = {finals[var]};")
     L.extend(body)
-    L.append("    }")
-    L.append("    __syncthreads();")
+    L.append("      }")
+    L.append("      __syncthreads();")
 
-    # ---- stage 2: tile compute from LDS ----------------------------------
+    # ---- stage 2 ---------------------------------------------------------
     readers = desc["readers"]
     s2_op_names = {v for (v, _, _) in desc["s2_ops"]}
+    dr0s = sorted({dr0 for (_, dr0, _) in readers.values()})
 
-    def resolve2(base, em):
-        if base in readers:
-            lds, dr0, dr1 = readers[base]
-            return (f"lds_{lds}[q0{em.tag} + {dr0}]"
+    def resolve2(base_name, em):
+        if base_name in readers:
+            lds, dr0, dr1 = readers[base_name]
+            return (f"lds_{lds}[sl{dr0}{em.tag}]"
                     f"[q1{em.tag} + {dr1}]")
-        if base in s2_op_names:
-            return (f"a.{base}_ptr[a.{base}_off + k0{em.tag} * a.{base}_s0"
-                    f" + k1{em.tag} * a.{base}_s1]")
+        if base_name in s2_op_names:
+            return (f"a.{base_name}_ptr[a.{base_name}_off + k0{em.tag} * "
+                    f"a.{base_name}_s0 + k1{em.tag} * a.{base_name}_s1]")
         return None
 
     s2gen = _StageGen(desc["s2_stmts"], resolve2, desc["dead2"])
-    L.append(f"    for (int ti = threadIdx.x; ti < {TH} * {TW}; ti += 256) "
-             "{")
-    L.append(f"      const int q0_t2 = ti / {TW}, q1_t2 = ti % {TW};")
-    L.append("      const i64 k0_t2 = k0o + q0_t2;")
-    L.append("      const i64 k1_t2 = k1o + q1_t2;")
-    L.append("      if (k0_t2 < a.n0 && k1_t2 < a.n1) {")
+    L.append(f"      for (int ti = threadIdx.x; ti < {TH} * {TW}; "
+             "ti += 256) {")
+    L.append(f"        const int q0_t2 = ti / {TW}, q1_t2 = ti % {TW};")
+    L.append("        const i64 k0_t2 = k0o + q0_t2;")
+    L.append("        const i64 k1_t2 = k1o + q1_t2;")
+    L.append("        if (k0_t2 < a.n0 && k1_t2 < a.n1) {")
+    for d in dr0s:
+        L.append(f"        int sl{d}_t2 = base + q0_t2 + {d}; "
+                 f"if (sl{d}_t2 >= {RING}) sl{d}_t2 -= {RING};")
     em2 = LaneEmitter(s2gen, "_t2",
                       ["(a.gs0 + k0_t2)", "(a.gs1 + k1_t2)"])
     finals2 = s2gen.emit_into(em2)
@@ -1586,9 +1613,11 @@ def generate_staged_tiled(desc):
                 f"k1_t2 * a.{var}_s1] = {finals2[var]};")
     for ri, (wvar, dt) in enumerate(tk_reds):
         L.append(f"      red{ri}_acc += ({ctype(dt)})({finals2[wvar]});")
+    L.append("        }")
     L.append("      }")
+    L.append("      __syncthreads();")
+    L.append(f"      base += {TH}; if (base >= {RING}) base -= {RING};")
     L.append("    }")
-    L.append("    __syncthreads();")
     L.append("  }")
     if tk_reds:
         # rim cells (core minus written image) folded into the same
@@ -1611,8 +1640,6 @@ def generate_staged_tiled(desc):
                          f"c * a.rsrc{ri}_s1];")
                 L.append("    }")
                 L.append("  }")
-        # per-block tree reduce of the fused-sum accumulators (the same
-        # wave shfl + cross-wave LDS idiom as the fused reduce kernels)
         L.append("  {")
         L.append("    const int lane = threadIdx.x & 63;")
         L.append("    const int wid = threadIdx.x >> 6;")

@@ -151,8 +151,12 @@ def build_vals_and_launch(rt, rec, g1, g2, live1, live2, ib2):
                 vals[f"rim{ri}_{k}_hi0"] = 0
                 vals[f"rim{ri}_{k}_lo1"] = 1
                 vals[f"rim{ri}_{k}_hi1"] = 0
-    ntiles = ((n0 + codegen.TILE_H - 1) // codegen.TILE_H) \
-        * ((n1 + codegen.TILE_W - 1) // codegen.TILE_W)
+    import os
+    th = int(os.environ.get("RAMBA_TK_TH", str(codegen.TILE_H)))
+    cw = int(os.environ.get("RAMBA_TK_CW", "128"))
+    seg = int(os.environ.get("RAMBA_TK_SEG", "4"))
+    tiles0 = (n0 + th - 1) // th
+    ntiles = ((tiles0 + seg - 1) // seg) * ((n1 + cw - 1) // cw)
     return backend.tiled_launch(rec.handle, vals, ntiles,
                                 [dt for (_, _, p) in g2.staged_reductions
                                  for dt in [p.dtype]])
