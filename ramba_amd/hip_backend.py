@@ -1019,12 +1019,12 @@ def _hb_maybe_load_tiled(self, plan):
     member_names = {n for (a, s, dt, mem, rep) in fams for (n, _, _) in mem}
     others = [op.name for op in plan.operands
               if op.name not in member_names]
-    return ("lt", cached[0], cached[1], fam_meta, others, tile)
+    return ("lt", cached[0], cached[1], fam_meta, others, tile, key)
 
 
 def _hb_launch_load_tiled(self, gk, plan):
     from . import codegen as cg
-    _, handle, fields, fam_meta, others, tile = gk
+    _, handle, fields, fam_meta, others, tile, ltkey = gk
     opmap = {o.name: o for o in plan.operands}
     vals = {"n0": plan.itershape[0], "n1": plan.itershape[1],
             "gs0": plan.global_start[0], "gs1": plan.global_start[1]}
@@ -1061,7 +1061,7 @@ def _hb_launch_load_tiled(self, gk, plan):
         ev1.record()
         ev1.synchronize()
         self.kernel_times_ms.append(ev0.elapsed_time(ev1))
-        self.kernel_keys.append("lt")
+        self.kernel_keys.append(ltkey)
     return []
 
 
