@@ -931,14 +931,14 @@ def _hb_tiled_kernel(self, desc):
         self._check(self.lib.rt_kernel_get(
             key.encode(), source.encode(), kname.encode(),
             ctypes.byref(h)), "rt_kernel_get(tiled)")
-        cached = (h.value, fields)
+        cached = (h.value, fields, key)
         self.kernels[ck] = cached
     return cached
 
 
 def _hb_tiled_launch(self, handle, vals, ntiles, red_dtypes=None):
     from . import codegen as cg
-    h, fields = handle
+    h, fields, tkkey = handle
     gx = max(1, min(int(os.environ.get("RAMBA_GRID_CAP", "32768")),
                     ntiles))
     red_dtypes = red_dtypes or []
@@ -959,7 +959,7 @@ def _hb_tiled_launch(self, handle, vals, ntiles, red_dtypes=None):
         ev1.record()
         ev1.synchronize()
         self.kernel_times_ms.append(ev0.elapsed_time(ev1))
-        self.kernel_keys.append("tiled")
+        self.kernel_keys.append(tkkey)
     out = []
     for t, dt in zip(parts, red_dtypes):
         total = self.torch.empty(1, dtype=t.dtype, device="cuda")
