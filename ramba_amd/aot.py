@@ -212,12 +212,16 @@ def seed(fuzz_seeds=None):
             os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
             "tests"))
         from fuzz_programs import build_program
+        from fuzz_staged import build_staged_program
         with np.errstate(all="ignore"):
             for sd in range(fuzz_seeds):
                 impl, _ = build_program(sd, mode="numpy")
                 impl(ra)
                 impl, _ = build_program(sd, mode="oracle")
                 impl(ra)
+            # staged-fusion fuzz kernels (tiled producer/consumer pairs)
+            for sd in range(min(fuzz_seeds, 200)):
+                build_staged_program(sd)(ra)
     n = be.compiled
     ra.shutdown()
     return n

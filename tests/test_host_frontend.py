@@ -1151,3 +1151,11 @@ def test_multi_axis_reductions(ra):
             np.asarray(a.sum(axis=(-1, 0))).reshape(-1)])
     from conftest import run_both
     run_both(impl, ra, tol=1e-12)
+
+
+def test_fuzz_staged_cpu(ra):
+    """Targeted staged-fusion fuzzer (tests/fuzz_staged.py), sequential
+    fallback semantics on the oracle backend."""
+    from fuzz_staged import check_staged_seed
+    for seed in range(100):
+        check_staged_seed(ra, seed)
