@@ -1215,12 +1215,70 @@ def meshgrid(*xi, copy=True, sparse=False, indexing="ij"):
     return out
 
 
+def pad(arr, pad_width, mode="constant", **kwargs):
+    """reference pad (ramba.py:9400; worker 9280): constant mode — a
+    fresh constant-filled array with the source written into its core
+    through the fused engine (the reference instead grows the edge
+    shards' distributions and np.pads per worker)."""
+    if isinstance(arr, np.ndarray):
+        arr = fromarray(arr)
+    nd = arr.ndim
+    if isinstance(pad_width, (int, np.integer)):
+        pad_width = ((int(pad_width),) * 2,) * nd
+    else:
+        pad_width = tuple(pad_width)
+        if pad_width and not isinstance(pad_width[0], (tuple, list)):
+            pad_width = (tuple(int(x) for x in pad_width),) * nd
+        else:
+            pad_width = tuple(tuple(int(x) for x in p) for p in pad_width)
+    assert len(pad_width) == nd
+    if mode != "constant":
+        raise NotImplementedError(
+            f"pad mode {mode!r} (constant only; the reference defers to "
+            "per-worker np.pad for other modes)")
+    cval = kwargs.get("constant_values", 0)
+    newshape = tuple(s + lo + hi
+                     for s, (lo, hi) in zip(arr.shape, pad_width))
+    out = full(newshape, cval, dtype=arr.dtype)
+    out[tuple(slice(lo, lo + s)
+              for s, (lo, hi) in zip(arr.shape, pad_width))] = arr
+    return out
+
+
+class _MGrid:
+    """reference mgrid (MgridGen, ramba.py:9001-9017): dense index grids;
+    one slice -> a 1-D array, k slices -> a stacked (k, n1..nk) array
+    (the reference materialises np.mgrid per shard, :3909-3927)."""
+
+    def __getitem__(self, index):
+        if not isinstance(index, tuple):
+            index = (index,)
+        axes = []
+        for sl in index:
+            assert isinstance(sl, slice), "mgrid takes slices"
+            step = sl.step if sl.step is not None else 1
+            start = sl.start if sl.start is not None else 0
+            if isinstance(step, complex):
+                axes.append(np.linspace(start, sl.stop, int(abs(step))))
+            else:
+                axes.append(np.arange(start, sl.stop, step))
+        dt = np.result_type(*axes)
+        axes = [a.astype(dt) for a in axes]
+        if len(axes) == 1:
+            return fromarray(axes[0])
+        return meshgrid(*axes, indexing="ij")
+
+
+mgrid = _MGrid()
+
+
 ndarray._ARRAY_FUNC.update({
     "matmul": lambda a, b, **kw: matmul(a, b),
     "dot": lambda a, b, **kw: matmul(a, b),
     "expand_dims": lambda a, axis: expand_dims(a, axis),
     "triu": lambda m, k=0: triu(m, k),
     "tril": lambda m, k=0: tril(m, k),
+    "pad": lambda a, w, mode="constant", **kw: pad(a, w, mode=mode, **kw),
 })
 
 

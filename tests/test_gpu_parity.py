@@ -793,3 +793,15 @@ def test_fuzz_staged_gpu(ra_gpu):
         st.staged_exec.run_recipe = orig
     assert sum(1 for h in hits if h) >= 150, \
         f"tiled path ran only {sum(1 for h in hits if h)}/200"
+
+
+def test_pad_mgrid_gpu(ra_gpu):
+    """pad + mgrid on the HIP path (reference ramba.py:9400/9017)."""
+    a = ra_gpu.fromfunction(lambda x, y: x * 10 + y, (70, 90))
+    n = np.fromfunction(lambda x, y: x * 10 + y, (70, 90))
+    assert np.array_equal(ra_gpu.pad(a, 3).asarray(), np.pad(n, 3))
+    assert np.array_equal(
+        ra_gpu.pad(a, ((1, 0), (2, 5)), constant_values=-2.0).asarray(),
+        np.pad(n, ((1, 0), (2, 5)), constant_values=-2.0))
+    assert np.array_equal(ra_gpu.mgrid[0:40, 2:31].asarray(),
+                          np.mgrid[0:40, 2:31])

@@ -1159,3 +1159,21 @@ def test_fuzz_staged_cpu(ra):
     from fuzz_staged import check_staged_seed
     for seed in range(100):
         check_staged_seed(ra, seed)
+
+
+def test_pad_and_mgrid(ra):
+    """reference pad (ramba.py:9400) and mgrid (ramba.py:9017) parity."""
+    def impl(np_):
+        a = np_.fromfunction(lambda x, y: x * 10 + y, (7, 9))
+        outs = [np_.pad(a, 2).reshape(-1) if np_ is np
+                else np_.pad(a, 2).reshape(-1)]
+        outs.append(np_.pad(a, ((0, 2), (3, 1))).reshape(-1))
+        outs.append(np_.pad(a, 1, constant_values=7.5).reshape(-1))
+        outs.append((np_.mgrid[0:5] * 1.0))
+        outs.append(np_.mgrid[0:4, 1:7].reshape(-1) * 1.0)
+        outs.append(np_.mgrid[0:1:5j, 0:2:4j].reshape(-1))
+        import numpy as _np
+        return _np.concatenate([_np.asarray(o, dtype=_np.float64)
+                                for o in outs])
+    from conftest import run_both
+    run_both(impl, ra, tol=1e-12)
