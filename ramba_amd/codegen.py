@@ -1121,40 +1121,48 @@ def generate(plan):
 
 
 def pack_args(gk, plan, ptr_of):
-    """Pack the Args struct per gk.fields.  ptr_of(name) -> device address
-    of an operand buffer; special names '__partials__'."""
-    out = bytearray()
+    """Pack the Args struct per gk.fields (one struct.pack call; the
+    format string is cached on the kernel).  ptr_of(name) -> device
+    address of an operand buffer; special names '__partials__'."""
     opmap = {o.name: o for o in plan.operands}
     lead = 0
     if gk.anchor is not None and gk.vec > 1:
         lead = (-opmap[gk.anchor].offset0) % gk.vec
+    fmt = getattr(gk, "_pack_fmt", None)
+    if fmt is None:
+        codes = {"iter_n": "q", "iter_gs": "q", "ptr": "Q", "off": "q",
+                 "stride": "q", "scalar_f": "d", "scalar_i": "q",
+                 "lead": "q", "partials": "Q", "npartials": "q"}
+        fmt = "<" + "".join(codes[k] for (k, _) in gk.fields)
+        gk._pack_fmt = fmt
+    vals = []
+    ap = vals.append
+    scalars = plan.scalars
     for kind, payload in gk.fields:
         if kind == "iter_n":
-            out += struct.pack("<q", plan.itershape[payload])
+            ap(plan.itershape[payload])
         elif kind == "iter_gs":
-            out += struct.pack("<q", plan.global_start[payload])
+            ap(plan.global_start[payload])
         elif kind == "ptr":
-            out += struct.pack("<Q", ptr_of(payload))
+            ap(ptr_of(payload))
         elif kind == "off":
-            out += struct.pack("<q", opmap[payload].offset0)
+            ap(opmap[payload].offset0)
         elif kind == "stride":
             name, d = payload
-            out += struct.pack("<q", opmap[name].strides[d])
+            ap(opmap[name].strides[d])
         elif kind == "scalar_f":
-            v, dt = plan.scalars[payload]
-            out += struct.pack("<d", float(v))
+            ap(float(scalars[payload][0]))
         elif kind == "scalar_i":
-            v, dt = plan.scalars[payload]
-            out += struct.pack("<q", int(v))
+            ap(int(scalars[payload][0]))
         elif kind == "lead":
-            out += struct.pack("<q", lead)
+            ap(lead)
         elif kind == "partials":
-            out += struct.pack("<Q", ptr_of("__partials__"))
+            ap(ptr_of("__partials__"))
         elif kind == "npartials":
-            out += struct.pack("<q", ptr_of("__npartials__"))
+            ap(ptr_of("__npartials__"))
         else:
             raise AssertionError(kind)
-    return bytes(out)
+    return struct.pack(fmt, *vals)
 
 
 def pack_finish_args(partials_ptr, npartials, out_ptrs):
@@ -1663,17 +1671,21 @@ def generate_staged_tiled(desc):
     return key, "\n".join(L), kname, fields
 
 
+_TK_FMT_CACHE = {}
+
+
 def pack_tk_args(fields, values):
-    out = bytearray()
-    for (name, kind) in fields:
-        v = values[name]
-        if kind == "Q":
-            out += struct.pack("<Q", int(v))
-        elif kind == "q":
-            out += struct.pack("<q", int(v))
-        else:
-            out += struct.pack("<d", float(v))
-    return bytes(out)
+    fid = id(fields)
+    ent = _TK_FMT_CACHE.get(fid)
+    if ent is None or ent[0] is not fields:
+        fmt = "<" + "".join(k for (_, k) in fields)
+        if len(_TK_FMT_CACHE) > 512:
+            _TK_FMT_CACHE.clear()
+        _TK_FMT_CACHE[fid] = ent = (fields, fmt)
+    fmt = ent[1]
+    return struct.pack(fmt, *[
+        float(values[n]) if k == "d" else int(values[n])
+        for (n, k) in fields])
 
 
 # ---------------------------------------------------------------------------
