@@ -609,9 +609,7 @@ class Runtime:
             # the reduction that was fused onto the pair: the sequential
             # order recomputes it over the full local core
             from . import staged_exec
-            staged_exec.finish_staged_reductions(rt=self, g2=g2,
-                                                 interior=None,
-                                                 written_images={})
+            staged_exec.finish_staged_reductions(self, g2)
 
     def _execute_group_with(self, group, live, dead):
         """_execute_group with a precomputed liveness split (recipe cache

@@ -11,8 +11,7 @@ from .shardview import box_shape, box_subtract
 
 class StagedRecipe:
     __slots__ = ("handle", "writer_names", "store_names", "s2_names",
-                 "F_lo", "E", "hbm_read_names", "residual_units",
-                 "lds_of_writer", "nred")
+                 "F_lo", "E", "residual_units", "lds_of_writer", "nred")
 
     def __init__(self):
         self.nred = 0
@@ -22,7 +21,6 @@ class StagedRecipe:
         self.s2_names = []          # g2 var names of HBM operands
         self.F_lo = (0, 0)
         self.E = (0, 0)
-        self.hbm_read_names = []    # g2 var names pinning g1 stores
         self.residual_units = None  # built lazily on first run
         self.lds_of_writer = {}
 
@@ -217,7 +215,7 @@ def run_recipe(rt, rec, g1, g2, live1, dead1, live2, hbm_read_gids):
                                           dtype=pend.dtype)[()]
         else:
             # no local tiles: reduce this rank's whole core (if any)
-            finish_staged_reductions(rt, g2, None, {})
+            finish_staged_reductions(rt, g2)
     backend.free_temps()
     for bd in g1.delete_bds + g2.delete_bds:
         if bd.constructed:
@@ -270,13 +268,12 @@ def reduce_boxes_partial(rt, bd, boxes, kind, dtype):
     return acc
 
 
-def finish_staged_reductions(rt, g2, interior, written_images):
-    """Set each staged-reduction pend.partial = interior partial (from
-    the tiled kernel, or identity) + the complement boxes of this rank's
-    core."""
+def finish_staged_reductions(rt, g2):
+    """Fallback/no-local-tiles path: each staged-reduction partial =
+    the plain reduce over this rank's whole core (the tiled path instead
+    folds interior+rim in-kernel and sets the partial directly)."""
     from . import ir
-    from .shardview import box_subtract as _bsub
-    for i, (src, wv, pend) in enumerate(g2.staged_reductions):
+    for (src, wv, pend) in g2.staged_reductions:
         bd = src.bdarray
         core = rt.core_box(bd, rt.rank)
         if core is None:
@@ -284,17 +281,5 @@ def finish_staged_reductions(rt, g2, interior, written_images):
                 ir.reduction_init(pend.kind, pend.dtype),
                 dtype=pend.dtype)[()]
             continue
-        wimg = written_images.get(bd.gid)
-        if wimg is None:
-            comp = [core]
-        else:
-            comp = _bsub(core, wimg)
-        cval = reduce_boxes_partial(rt, bd, comp, pend.kind, pend.dtype)
-        ival = interior[i] if interior is not None else None
-        if ival is None:
-            pend.partial = cval
-        else:
-            comb, _ = ir.REDUCTIONS[pend.kind]
-            pend.partial = np.asarray(
-                ir.BINOPS[comb](np.asarray(ival, dtype=pend.dtype)[()],
-                                cval), dtype=pend.dtype)[()]
+        pend.partial = reduce_boxes_partial(rt, bd, [core], pend.kind,
+                                            pend.dtype)
