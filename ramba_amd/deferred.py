@@ -154,9 +154,11 @@ class FusedGroup:
     # -- operand registration ------------------------------------------------
 
     def arr_var(self, arr, written):
-        """Same (gid, view) -> same var (reference add_gid, ramba.py:8078)."""
+        """Same (gid, view) -> same var (reference add_gid, ramba.py:8078).
+        Views are interned, so identity keys the map (a content-equal but
+        distinct View object only costs an extra var name — harmless)."""
         gid = arr.bdarray.gid
-        key = (gid, arr.view)
+        key = (gid, id(arr.view))
         name = self._var_by_key.get(key)
         if name is not None:
             if written:
@@ -192,16 +194,19 @@ _eb_cache = {}
 
 
 def exec_boxes_cached(view, divisions):
-    """exec_boxes with a content-keyed cache: _ensure_group recomputes the
-    per-rank preimages for EVERY op of an iterating workload otherwise
-    (~5-10 us x ops x steps of pure frontend overhead)."""
-    key = (view, divisions.shape, divisions.tobytes())
-    r = _eb_cache.get(key)
-    if r is None:
-        r = exec_boxes(view, divisions)
-        if len(_eb_cache) > 4096:
-            _eb_cache.clear()
-        _eb_cache[key] = r
+    """exec_boxes with an IDENTITY-keyed cache (Views are interned by the
+    getitem/identity caches and divisions arrays are shared/rebound, so
+    object identity is the fast, safe key; the stored refs pin the ids):
+    _ensure_group would otherwise recompute the per-rank preimages for
+    EVERY op of an iterating workload."""
+    k = (id(view), id(divisions))
+    e = _eb_cache.get(k)
+    if e is not None and e[0] is view and e[1] is divisions:
+        return e[2]
+    r = exec_boxes(view, divisions)
+    if len(_eb_cache) > 4096:
+        _eb_cache.clear()
+    _eb_cache[k] = (view, divisions, r)
     return r
 
 

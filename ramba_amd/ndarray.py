@@ -223,15 +223,16 @@ class ndarray:
                 parts = None
                 break
         if parts is not None:
-            key = (self.view, tuple(parts))
-            view = _getitem_cache.get(key)
-            if view is not None:
-                return self._with_view(view)
+            key = (id(self.view), tuple(parts))
+            ent = _getitem_cache.get(key)
+            if ent is not None and ent[0] is self.view:
+                return self._with_view(ent[1])
         view = self.view.apply_index(index)
         if key is not None:
             if len(_getitem_cache) > 8192:
                 _getitem_cache.clear()
-            _getitem_cache[key] = view
+            # the stored base-view ref pins the id against reuse
+            _getitem_cache[key] = (self.view, view)
         return self._with_view(view)
 
     def __setitem__(self, index, value):
