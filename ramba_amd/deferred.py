@@ -155,10 +155,13 @@ class FusedGroup:
 
     def arr_var(self, arr, written):
         """Same (gid, view) -> same var (reference add_gid, ramba.py:8078).
-        Views are interned, so identity keys the map (a content-equal but
-        distinct View object only costs an extra var name — harmless)."""
+        The key MUST be view CONTENT, not identity: a write and a read of
+        the same (gid, view) arriving as distinct-but-equal View objects
+        must merge into ONE var, or the kernel's SSA forwarding loses the
+        in-group RAW (caught by fuzz seed 39 on the HIP path — the CPU
+        oracle's sequential memory interpretation hides it)."""
         gid = arr.bdarray.gid
-        key = (gid, id(arr.view))
+        key = (gid, arr.view)
         name = self._var_by_key.get(key)
         if name is not None:
             if written:
