@@ -242,7 +242,8 @@ def _aot_tiled_kernel(self, desc):
     return (key, fields)
 
 
-def _aot_tiled_launch(self, handle, vals, ntiles, red_dtypes=None):
+def _aot_tiled_launch(self, handle, vals, ntiles, red_dtypes=None,
+                      rec=None):
     return [np.asarray(0, dtype=dt)[()] for dt in (red_dtypes or [])]
 
 

@@ -936,17 +936,27 @@ def _hb_tiled_kernel(self, desc):
     return cached
 
 
-def _hb_tiled_launch(self, handle, vals, ntiles, red_dtypes=None):
+def _hb_tiled_launch(self, handle, vals, ntiles, red_dtypes=None,
+                     rec=None):
     from . import codegen as cg
     h, fields, tkkey = handle
     gx = max(1, min(int(os.environ.get("RAMBA_GRID_CAP", "32768")),
                     ntiles))
     red_dtypes = red_dtypes or []
-    parts = []
-    for ri, dt in enumerate(red_dtypes):
-        t = self.torch.empty(gx, dtype=self._tdt(dt), device="cuda")
-        parts.append(t)
-        vals[f"red{ri}_ptr"] = t.data_ptr()
+    if rec is not None and rec.partials is not None \
+            and len(rec.partials) == len(red_dtypes) \
+            and all(t.shape[0] == gx for t in rec.partials):
+        parts = rec.partials
+        for ri, t in enumerate(parts):
+            vals[f"red{ri}_ptr"] = t.data_ptr()
+    else:
+        parts = []
+        for ri, dt in enumerate(red_dtypes):
+            t = self.torch.empty(gx, dtype=self._tdt(dt), device="cuda")
+            parts.append(t)
+            vals[f"red{ri}_ptr"] = t.data_ptr()
+        if rec is not None:
+            rec.partials = parts
     args = cg.pack_tk_args(fields, vals)
     if self.time_kernels:
         ev0 = self.torch.cuda.Event(enable_timing=True)

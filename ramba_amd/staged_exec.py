@@ -11,10 +11,14 @@ from .shardview import box_shape, box_subtract
 
 class StagedRecipe:
     __slots__ = ("handle", "writer_names", "store_names", "s2_names",
-                 "F_lo", "E", "residual_units", "lds_of_writer", "nred")
+                 "F_lo", "E", "residual_units", "lds_of_writer", "nred",
+                 "vals", "vals_key", "partials")
 
     def __init__(self):
         self.nred = 0
+        self.vals = None        # cached arg values (ptr+scalar validated)
+        self.vals_key = None
+        self.partials = None    # persistent reduction partials tensors
         self.handle = None
         self.writer_names = []      # g1 var names of staged writers
         self.store_names = []       # g1 var names of non-staged stores
@@ -75,9 +79,35 @@ def _store_vals(vals, prefix, rt, bd):
     return d
 
 
+def _vals_probe(rt, rec, g1, g2, live1, live2):
+    """Cheap validity key for the cached vals: every container address
+    that appears in them + every scalar value (iterating workloads keep
+    both stable; the staged fuzzer's per-iteration scalars invalidate)."""
+    backend = rt.backend
+    ptrs = []
+    for wname in rec.writer_names:
+        if wname in live1:
+            ptrs.append(backend.container_addr(g1.arr_vars[wname].bd))
+    for name in rec.store_names:
+        ptrs.append(backend.container_addr(live1[name].bd))
+    for name in rec.s2_names:
+        ptrs.append(backend.container_addr(live2[name].bd))
+    for (src, wv, pend) in g2.staged_reductions:
+        ptrs.append(backend.container_addr(src.bdarray))
+    scal = tuple(v for (v, dt) in g1.scalars.values())         + tuple(v for (v, dt) in g2.scalars.values())
+    return (tuple(ptrs), scal)
+
+
 def build_vals_and_launch(rt, rec, g1, g2, live1, live2, ib2):
     backend = rt.backend
     from . import codegen
+    probe = None
+    try:
+        probe = _vals_probe(rt, rec, g1, g2, live1, live2)
+    except Exception:  # noqa: BLE001  (missing shard etc: rebuild)
+        probe = None
+    if probe is not None and rec.vals is not None             and rec.vals_key == probe:
+        return _tk_launch(rt, rec, g2, rec.vals, ib2)
     vals = {}
     n0, n1 = box_shape(ib2)
     vals["n0"], vals["n1"] = n0, n1
@@ -149,15 +179,22 @@ def build_vals_and_launch(rt, rec, g1, g2, live1, live2, ib2):
                 vals[f"rim{ri}_{k}_hi0"] = 0
                 vals[f"rim{ri}_{k}_lo1"] = 1
                 vals[f"rim{ri}_{k}_hi1"] = 0
+    rec.vals, rec.vals_key = vals, probe
+    return _tk_launch(rt, rec, g2, vals, ib2)
+
+
+def _tk_launch(rt, rec, g2, vals, ib2):
     import os
+    from . import codegen
+    n0, n1 = box_shape(ib2)
     th = int(os.environ.get("RAMBA_TK_TH", str(codegen.TILE_H)))
     cw = int(os.environ.get("RAMBA_TK_CW", "128"))
     seg = int(os.environ.get("RAMBA_TK_SEG", "4"))
     tiles0 = (n0 + th - 1) // th
     ntiles = ((tiles0 + seg - 1) // seg) * ((n1 + cw - 1) // cw)
-    return backend.tiled_launch(rec.handle, vals, ntiles,
-                                [dt for (_, _, p) in g2.staged_reductions
-                                 for dt in [p.dtype]])
+    return rt.backend.tiled_launch(
+        rec.handle, vals, ntiles,
+        [p.dtype for (_, _, p) in g2.staged_reductions], rec=rec)
 
 
 def run_residual(rt, rec, g1, live1, dead1, ib1, ib2):
