@@ -69,7 +69,8 @@ class AotCompileBackend:
             self._cc(gk.finish_source)
         # also precompile the LDS load-tiled stencil variant when the
         # plan qualifies (the HIP backend prefers it at runtime)
-        if not plan.reductions and len(plan.itershape) == 2                 and plan.itershape[0] * plan.itershape[1] >= (1 << 16):
+        if (not plan.reductions and len(plan.itershape) == 2
+                and plan.itershape[0] * plan.itershape[1] >= (1 << 16)):
             fams = codegen.find_stencil_families(plan)
             if fams:
                 _, src, _, _, _ = codegen.generate_load_tiled(plan, fams)

@@ -94,7 +94,8 @@ def _vals_probe(rt, rec, g1, g2, live1, live2):
         ptrs.append(backend.container_addr(live2[name].bd))
     for (src, wv, pend) in g2.staged_reductions:
         ptrs.append(backend.container_addr(src.bdarray))
-    scal = tuple(v for (v, dt) in g1.scalars.values())         + tuple(v for (v, dt) in g2.scalars.values())
+    scal = (tuple(v for (v, dt) in g1.scalars.values())
+            + tuple(v for (v, dt) in g2.scalars.values()))
     return (tuple(ptrs), scal)
 
 
@@ -106,7 +107,8 @@ def build_vals_and_launch(rt, rec, g1, g2, live1, live2, ib2):
         probe = _vals_probe(rt, rec, g1, g2, live1, live2)
     except Exception:  # noqa: BLE001  (missing shard etc: rebuild)
         probe = None
-    if probe is not None and rec.vals is not None             and rec.vals_key == probe:
+    if (probe is not None and rec.vals is not None
+            and rec.vals_key == probe):
         return _tk_launch(rt, rec, g2, rec.vals, ib2)
     vals = {}
     n0, n1 = box_shape(ib2)
