@@ -398,7 +398,8 @@ def _stageable(g, conflicts, write_arr, operands):
         v = o.view
         if o is write_arr:
             return False           # consumer writing a producer array
-        if v.ndim != nd or v.axis_map != tuple(range(nd))                 or v.steps != (1,) * nd:
+        if (v.ndim != nd or v.axis_map != tuple(range(nd))
+                or v.steps != (1,) * nd):
             return False
         if any(not (0 <= off <= 16) for off in v.offset):
             return False
