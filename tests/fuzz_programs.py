@@ -188,8 +188,18 @@ def build_program(seed, mode="numpy"):
                 op = str(rng.choice(ops))
                 r = _apply_bin(np_, val, v2, op)
                 ex = exact and ex2 and not (op == "div1" and div_taints)
-                pool.append((r, "f" if op == "div1" or "f" in (kind, k2)
-                             else kind, ex))
+                # result-kind: int only for int op int (seed 598751: the
+                # old label kept 'i' for int x float32('g'), letting a
+                # float64 value reach the integer-only op pool — both
+                # engines then raise identically, a generator-validity
+                # bug, not a parity gap)
+                if kind == "i" and k2 == "i":
+                    rk = "f" if op == "div1" else "i"
+                elif "f" in (kind, k2) or op == "div1":
+                    rk = "f"
+                else:
+                    rk = "g"
+                pool.append((r, rk, ex))
             elif action == "view":
                 if val.ndim == 1:
                     sl = _rand_slice(rng, val.shape[0])
